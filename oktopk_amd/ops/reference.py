@@ -1,0 +1,148 @@
+"""Pure-PyTorch reference implementations of every engine op.
+
+These are the numerics oracle for the HIP/CDNA4 kernels (tests compare the
+kernels against these in fp32) and the CPU execution path (gloo CI).  Each op
+mirrors a torch/numpy call-site of the reference implementation — citations on
+each function point into /root/reference.
+
+Contract shared with the HIP backend (oktopk_amd/ops/csrc):
+  * gradients / dense tensors: 1-D contiguous fp32 (or bf16 where noted)
+  * index vectors: int32, ascending order when produced by compact_gt
+"""
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+__all__ = [
+    "kth_abs_value",
+    "compact_gt",
+    "count_gt",
+    "scatter_add_",
+    "zero_at_",
+    "fill_sparse_scaled_",
+    "isin_sorted",
+    "ef_restore_snapshot_",
+    "fused_sgd_",
+    "fused_adam_",
+    "l2norm",
+]
+
+
+def kth_abs_value(t: torch.Tensor, k: int) -> float:
+    """|t|'s k-th largest value (the exact top-k threshold).
+
+    Reference: torch.topk in ratio2threshold, VGG/compression.py:370-381."""
+    k = max(1, min(int(k), t.numel()))
+    vals = torch.topk(t.abs().reshape(-1), k, sorted=True).values
+    return float(vals[-1].item())
+
+
+def compact_gt(t: torch.Tensor, tau: float) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Indices (int32, ascending) and values where |t| > tau.
+
+    Reference: compressbythreshold, VGG/compression.py:122-132."""
+    flat = t.reshape(-1)
+    idx = (flat.abs() > tau).nonzero(as_tuple=False).reshape(-1)
+    return idx.to(torch.int32), flat[idx]
+
+
+def count_gt(t: torch.Tensor, tau: float) -> int:
+    return int((t.reshape(-1).abs() > tau).sum().item())
+
+
+def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
+    """dest[idx] += val (duplicate indices accumulate).
+
+    Reference: reduced_t[indexes] += values, VGG/allreducer.py:779,794."""
+    dest.reshape(-1).index_add_(0, idx.long(), val.to(dest.dtype))
+    return dest
+
+
+def zero_at_(t: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    """t[idx] = 0. Reference: update_residuals, VGG/compression.py:467-471."""
+    t.reshape(-1)[idx.long()] = 0
+    return t
+
+
+def fill_sparse_scaled_(
+    out: torch.Tensor, idx: torch.Tensor, val: torch.Tensor, scale: float
+) -> torch.Tensor:
+    """out.zero_(); out[idx] = val * scale.
+
+    Reference: result.fill_(0); result[idx] = values/P, VGG/allreducer.py:838-841."""
+    flat = out.reshape(-1)
+    flat.zero_()
+    flat[idx.long()] = val.to(flat.dtype) * scale
+    return out
+
+
+def isin_sorted(a: torch.Tensor, b_sorted: torch.Tensor) -> torch.Tensor:
+    """Boolean mask: which elements of a are present in ascending-sorted b.
+
+    Reference: np.intersect1d for residual credit, VGG/allreducer.py:844,1051
+    (BERT uses the same searchsorted trick, BERT/bert/allreducer.py:21-24)."""
+    if b_sorted.numel() == 0:
+        return torch.zeros(a.numel(), dtype=torch.bool, device=a.device)
+    a64 = a.long()
+    b64 = b_sorted.long()
+    pos = torch.searchsorted(b64, a64).clamp_(max=b64.numel() - 1)
+    return b64[pos] == a64
+
+
+def ef_restore_snapshot_(t: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
+    """t += residual; residual = t  (fused in HIP).
+
+    Reference: ratio2threshold / add2residual preamble,
+    VGG/compression.py:375-379,389-391."""
+    t.add_(residual)
+    residual.copy_(t)
+    return t
+
+
+def fused_sgd_(
+    param: torch.Tensor,
+    grad: torch.Tensor,
+    momentum_buf: torch.Tensor,
+    lr: float,
+    momentum: float,
+    weight_decay: float,
+    nesterov: bool,
+) -> None:
+    """SGD with momentum/nesterov/weight-decay.
+
+    Reference: _DistributedOptimizer._step, VGG/distributed_optimizer.py:107-145."""
+    d_p = grad
+    if weight_decay != 0:
+        d_p = d_p.add(param, alpha=weight_decay)
+    if momentum != 0:
+        momentum_buf.mul_(momentum).add_(d_p)
+        d_p = d_p.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
+    param.add_(d_p, alpha=-lr)
+
+
+def fused_adam_(
+    param: torch.Tensor,
+    grad: torch.Tensor,
+    exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor,
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+) -> None:
+    """BertAdam-style update: no bias correction, decoupled weight decay.
+
+    Reference: BertAdam.step, BERT/bert/transformers/optimization.py:183-224."""
+    exp_avg.mul_(beta1).add_(grad, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+    update = exp_avg / (exp_avg_sq.sqrt() + eps)
+    if weight_decay != 0:
+        update = update + weight_decay * param
+    param.add_(update, alpha=-lr)
+
+
+def l2norm(t: torch.Tensor) -> float:
+    return float(t.reshape(-1).norm(p=2).item())

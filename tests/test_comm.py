@@ -1,0 +1,76 @@
+"""Comm-layer tests over gloo world 2 (the CPU plumbing config of
+BASELINE.json #1). Mirrors the reference's 2-process comm micro-tests
+(/root/reference/BERT/tests/communication/)."""
+import torch
+
+from conftest import run_dist
+
+
+def _allgatherv(rank):
+    from oktopk_amd.comm import Comm
+    import torch.distributed as dist
+
+    c = Comm(dist.group.WORLD)
+    t = torch.arange(3 + rank * 2, dtype=torch.float32) + rank * 100
+    buf, sizes = c.allgatherv(t)
+    assert sizes == [3, 5]
+    assert torch.equal(buf[:3], torch.tensor([0.0, 1.0, 2.0]))
+    assert torch.equal(buf[3:], torch.tensor([100.0, 101.0, 102.0, 103.0, 104.0]))
+
+
+def test_allgatherv():
+    run_dist(_allgatherv, 2)
+
+
+def _alltoallv(rank):
+    from oktopk_amd.comm import Comm
+    import torch.distributed as dist
+
+    c = Comm(dist.group.WORLD)
+    # rank r sends [r*10+d] of size (d+1) to dest d
+    if rank == 0:
+        send = torch.tensor([0, 1, 1], dtype=torch.int32)
+        send_splits = [1, 2]
+    else:
+        send = torch.tensor([10, 11, 11], dtype=torch.int32)
+        send_splits = [1, 2]
+    recv_splits = c.alltoall_sizes(send_splits, c.device)
+    out = c.alltoallv(send, send_splits, recv_splits)
+    if rank == 0:
+        assert recv_splits == [1, 1]
+        assert out.tolist() == [0, 10]
+    else:
+        assert recv_splits == [2, 2]
+        assert out.tolist() == [1, 1, 11, 11]
+
+
+def test_alltoallv():
+    run_dist(_alltoallv, 2)
+
+
+def _sizes(rank):
+    from oktopk_amd.comm import Comm
+    import torch.distributed as dist
+
+    c = Comm(dist.group.WORLD)
+    sizes = c.allgather_sizes(5 + rank, c.device)
+    assert sizes.tolist() == [5, 6]
+
+
+def test_allgather_sizes():
+    run_dist(_sizes, 2)
+
+
+def test_world1_noop_comm():
+    from oktopk_amd.comm import Comm
+
+    c = Comm(None)
+    assert c.size == 1 and c.rank == 0
+    t = torch.randn(4)
+    ref = t.clone()
+    c.allreduce_(t)
+    assert torch.equal(t, ref)
+    buf, sizes = c.allgatherv(t)
+    assert torch.equal(buf, ref) and sizes == [4]
+    out = c.alltoallv(t, [4], [4])
+    assert torch.equal(out, ref)
