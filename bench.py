@@ -1,0 +1,154 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: BERT-base seq=128, bs=8/GPU, Ok-Topk density=0.1%,
+bf16 autocast, synthetic data — the BASELINE.json headline config — plus an
+in-run dense-allreduce baseline for the speedup-vs-dense figure.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`, launched
+under torch.distributed.run for N>1 (one rank per GPU over RCCL).  Rank 0
+prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", type=str, default="bert_base")
+    p.add_argument("--batch-size", type=int, default=8)
+    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--density", type=float, default=0.001)
+    p.add_argument("--compressor", type=str, default="oktopk")
+    p.add_argument("--dense-baseline-steps", type=int, default=-1,
+                   help="steps for the in-run dense baseline (-1: min(steps,10); 0: skip)")
+    return p.parse_args()
+
+
+def timed_steps(trainer, comm, steps):
+    """Barrier + device sync bracketed timing of exactly `steps` steps;
+    returns max-over-ranks elapsed seconds."""
+    dev_sync = torch.cuda.synchronize if torch.cuda.is_available() else (lambda: None)
+    comm.barrier()
+    dev_sync()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        trainer.step()
+    dev_sync()
+    comm.barrier()
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([elapsed], device=comm.device)
+    comm.allreduce_(t, op="max")
+    return float(t.cpu().item())
+
+
+def build_trainer(args, comm, compressor):
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.trainer import Trainer
+
+    preset = "bert" if args.model.startswith("bert") else (
+        "lstm" if args.model == "lstman4" else "vgg"
+    )
+    cfg = EngineConfig.preset(preset, compressor=compressor, density=args.density,
+                              dense_warmup_iters=0)
+    dtype = "bf16" if torch.cuda.is_available() else "fp32"
+    return Trainer(
+        model_name=args.model,
+        batch_size=args.batch_size,
+        seq_len=args.seq_len,
+        comm=comm,
+        cfg=cfg,
+        dtype=dtype,
+    )
+
+
+def reducer_phase_ms(trainer, steps):
+    red = getattr(trainer.opt, "reducer", None)
+    if red is None:
+        return {}
+    totals = {}
+    for name, phases in red.timers.items():
+        for ph, secs in phases.items():
+            totals[ph] = totals.get(ph, 0.0) + secs
+    return {ph: 1000.0 * s / max(steps, 1) for ph, s in totals.items()}
+
+
+def main():
+    args = parse_args()
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+    from oktopk_amd.comm import init_from_env
+
+    comm = init_from_env()
+    n_gpus = comm.size if comm.size > 1 else args.gpus
+    rank = comm.rank
+
+    trainer = build_trainer(args, comm, args.compressor)
+    for _ in range(args.warmup):
+        trainer.step()
+    # reset phase timers after warmup so per-step phase ms is steady-state
+    if getattr(trainer.opt, "reducer", None) is not None:
+        trainer.opt.reducer.timers = {}
+    elapsed = timed_steps(trainer, comm, args.steps)
+    ms_per_step = 1000.0 * elapsed / args.steps
+    phases = reducer_phase_ms(trainer, args.steps)
+    comm_ms = sum(phases.get(p, 0.0) for p in ("alltoall", "allgather", "allreduce"))
+    allreduce_ms = sum(phases.values())  # whole sparse-allreduce pipeline
+
+    dense_steps = args.dense_baseline_steps
+    if dense_steps < 0:
+        dense_steps = min(args.steps, 10)
+    dense_ms = None
+    speedup = None
+    if dense_steps > 0 and args.compressor not in ("dense", "none"):
+        dense_tr = build_trainer(args, comm, "dense")
+        for _ in range(min(args.warmup, 3)):
+            dense_tr.step()
+        dense_elapsed = timed_steps(dense_tr, comm, dense_steps)
+        dense_ms = 1000.0 * dense_elapsed / dense_steps
+        speedup = dense_ms / ms_per_step
+
+    tokens_per_step = n_gpus * args.batch_size * args.seq_len
+    value = tokens_per_step / (ms_per_step / 1000.0)
+
+    if rank == 0:
+        out = {
+            "metric": "tokens/s (BERT-base seq128 bs8/GPU, Ok-Topk density=0.1%, step time & allreduce ms; speedup vs dense)",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if torch.cuda.is_available() else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch_size * n_gpus,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{n_gpus}",
+                "compressor": args.compressor,
+                "density": args.density,
+                "allreduce_ms_per_step": round(allreduce_ms, 3),
+                "comm_ms_per_step": round(comm_ms, 3),
+                "dense_ms_per_step": round(dense_ms, 3) if dense_ms else None,
+                "speedup_vs_dense": round(speedup, 3) if speedup else None,
+                "phase_ms": {k: round(v, 3) for k, v in phases.items()},
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

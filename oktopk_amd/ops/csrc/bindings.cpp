@@ -1,0 +1,231 @@
+// PyTorch-ROCm bindings for the CDNA4 kernel library (kernels.hip).
+// API mirrors oktopk_amd/ops/reference.py one-to-one so the dispatch layer
+// can swap backends and the numerics tests can A/B them.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+
+#include <cstdint>
+#include <vector>
+
+// launchers from kernels.hip
+extern "C" {
+void launch_count_gt(const float*, int64_t, float, unsigned long long*, hipStream_t);
+void launch_compact_count(const float*, int64_t, float, int64_t, int, int*, hipStream_t);
+void launch_compact_write(const float*, int64_t, float, int64_t, int, const int*,
+                          int32_t*, float*, hipStream_t);
+void launch_hist(const float*, int64_t, uint32_t, uint32_t, int, int, unsigned int*,
+                 hipStream_t);
+void launch_scatter_add(float*, const int32_t*, const float*, int64_t, hipStream_t);
+void launch_scatter_set_scaled(float*, const int32_t*, const float*, float, int64_t,
+                               hipStream_t);
+void launch_zero_at(float*, const int32_t*, int64_t, hipStream_t);
+void launch_isin_sorted(const int32_t*, int64_t, const int32_t*, int64_t, bool*,
+                        hipStream_t);
+void launch_ef_restore(float*, float*, int64_t, hipStream_t);
+void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
+                hipStream_t);
+void launch_adam(float*, const float*, float*, float*, int64_t, float, float, float,
+                 float, float, hipStream_t);
+void launch_sumsq(const float*, int64_t, double*, hipStream_t);
+}
+
+namespace {
+
+inline hipStream_t cur_stream() {
+    return at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_f32_1d(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+    TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void check_i32_1d(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+    TORCH_CHECK(t.scalar_type() == torch::kInt32, name, " must be int32");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+}  // namespace
+
+static int64_t count_gt(torch::Tensor t, double tau) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    auto out = torch::zeros({1}, t.options().dtype(torch::kInt64));
+    launch_count_gt(t.data_ptr<float>(), t.numel(), (float)tau,
+                    reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()),
+                    cur_stream());
+    return out.cpu().item<int64_t>();
+}
+
+static std::vector<torch::Tensor> compact_gt(torch::Tensor t, double tau) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    int64_t n = t.numel();
+    // chunk per block: multiple of BLOCK*COMPACT_VEC (1024) covering n
+    const int64_t unit = 1024;
+    int64_t nchunks = (n + unit - 1) / unit;
+    if (nchunks < 1) nchunks = 1;
+    if (nchunks > 2048) nchunks = 2048;
+    int64_t chunk = ((n + nchunks - 1) / nchunks + unit - 1) / unit * unit;
+    int nblocks = (int)((n + chunk - 1) / chunk);
+    if (nblocks < 1) nblocks = 1;
+
+    auto counts = torch::empty({nblocks}, t.options().dtype(torch::kInt32));
+    launch_compact_count(t.data_ptr<float>(), n, (float)tau, chunk, nblocks,
+                         counts.data_ptr<int>(), cur_stream());
+    auto csum = counts.cumsum(0, torch::kInt32);
+    auto offsets = csum - counts;  // exclusive scan
+    int64_t total = csum[nblocks - 1].cpu().item<int>();
+    auto idx = torch::empty({total}, t.options().dtype(torch::kInt32));
+    auto val = torch::empty({total}, t.options());
+    if (total > 0) {
+        launch_compact_write(t.data_ptr<float>(), n, (float)tau, chunk, nblocks,
+                             offsets.contiguous().data_ptr<int>(),
+                             idx.data_ptr<int32_t>(), val.data_ptr<float>(),
+                             cur_stream());
+    }
+    return {idx, val};
+}
+
+static double kth_abs_value(torch::Tensor t, int64_t k) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    int64_t n = t.numel();
+    TORCH_CHECK(n > 0, "kth_abs_value on empty tensor");
+    if (k < 1) k = 1;
+    if (k > n) k = n;
+    // 3 histogram levels over |x| bits: 11 + 11 + 10
+    const int shifts[3] = {21, 10, 0};
+    const int bits[3] = {11, 11, 10};
+    uint32_t prefix_mask = 0, prefix_val = 0;
+    int64_t remaining = k;
+    auto hist = torch::empty({2048}, t.options().dtype(torch::kInt32));
+    for (int lvl = 0; lvl < 3; ++lvl) {
+        int nbins = 1 << bits[lvl];
+        hist.zero_();
+        launch_hist(t.data_ptr<float>(), n, prefix_mask, prefix_val, shifts[lvl], nbins,
+                    reinterpret_cast<unsigned int*>(hist.data_ptr<int>()), cur_stream());
+        auto h = hist.narrow(0, 0, nbins).cpu();
+        const int* hp = h.data_ptr<int>();
+        int64_t b = nbins - 1;
+        for (; b >= 0; --b) {
+            if (remaining <= hp[b]) break;
+            remaining -= hp[b];
+        }
+        if (b < 0) b = 0;  // fewer matching elements than k (ties/degenerate)
+        prefix_val |= ((uint32_t)b) << shifts[lvl];
+        prefix_mask |= ((uint32_t)(nbins - 1)) << shifts[lvl];
+    }
+    union { uint32_t u; float f; } c;
+    c.u = prefix_val;
+    return (double)c.f;
+}
+
+static torch::Tensor scatter_add_(torch::Tensor dest, torch::Tensor idx,
+                                  torch::Tensor val) {
+    check_f32_1d(dest, "dest");
+    check_f32_1d(val, "val");
+    check_i32_1d(idx, "idx");
+    const at::cuda::CUDAGuard guard(dest.device());
+    TORCH_CHECK(idx.numel() == val.numel(), "idx/val size mismatch");
+    if (idx.numel())
+        launch_scatter_add(dest.data_ptr<float>(), idx.data_ptr<int32_t>(),
+                           val.data_ptr<float>(), idx.numel(), cur_stream());
+    return dest;
+}
+
+static torch::Tensor zero_at_(torch::Tensor t, torch::Tensor idx) {
+    check_f32_1d(t, "t");
+    check_i32_1d(idx, "idx");
+    const at::cuda::CUDAGuard guard(t.device());
+    if (idx.numel())
+        launch_zero_at(t.data_ptr<float>(), idx.data_ptr<int32_t>(), idx.numel(),
+                       cur_stream());
+    return t;
+}
+
+static torch::Tensor fill_sparse_scaled_(torch::Tensor out, torch::Tensor idx,
+                                         torch::Tensor val, double scale) {
+    check_f32_1d(out, "out");
+    check_i32_1d(idx, "idx");
+    check_f32_1d(val, "val");
+    const at::cuda::CUDAGuard guard(out.device());
+    out.zero_();
+    if (idx.numel())
+        launch_scatter_set_scaled(out.data_ptr<float>(), idx.data_ptr<int32_t>(),
+                                  val.data_ptr<float>(), (float)scale, idx.numel(),
+                                  cur_stream());
+    return out;
+}
+
+static torch::Tensor isin_sorted(torch::Tensor a, torch::Tensor b_sorted) {
+    check_i32_1d(a, "a");
+    check_i32_1d(b_sorted, "b_sorted");
+    const at::cuda::CUDAGuard guard(a.device());
+    auto out = torch::zeros({a.numel()}, a.options().dtype(torch::kBool));
+    if (a.numel() && b_sorted.numel())
+        launch_isin_sorted(a.data_ptr<int32_t>(), a.numel(),
+                           b_sorted.data_ptr<int32_t>(), b_sorted.numel(),
+                           out.data_ptr<bool>(), cur_stream());
+    return out;
+}
+
+static torch::Tensor ef_restore_snapshot_(torch::Tensor t, torch::Tensor r) {
+    check_f32_1d(t, "t");
+    check_f32_1d(r, "r");
+    const at::cuda::CUDAGuard guard(t.device());
+    TORCH_CHECK(t.numel() == r.numel(), "t/r size mismatch");
+    launch_ef_restore(t.data_ptr<float>(), r.data_ptr<float>(), t.numel(), cur_stream());
+    return t;
+}
+
+static void fused_sgd_(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
+                       double momentum, double wd, bool nesterov) {
+    check_f32_1d(p, "p");
+    check_f32_1d(g, "g");
+    check_f32_1d(buf, "buf");
+    const at::cuda::CUDAGuard guard(p.device());
+    launch_sgd(p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
+               p.numel(), (float)lr, (float)momentum, (float)wd, nesterov ? 1 : 0,
+               cur_stream());
+}
+
+static void fused_adam_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                        torch::Tensor v, double lr, double b1, double b2, double eps,
+                        double wd) {
+    check_f32_1d(p, "p");
+    check_f32_1d(g, "g");
+    check_f32_1d(m, "m");
+    check_f32_1d(v, "v");
+    const at::cuda::CUDAGuard guard(p.device());
+    launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                v.data_ptr<float>(), p.numel(), (float)lr, (float)b1, (float)b2,
+                (float)eps, (float)wd, cur_stream());
+}
+
+static double l2norm(torch::Tensor t) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    auto out = torch::zeros({1}, t.options().dtype(torch::kFloat64));
+    launch_sumsq(t.data_ptr<float>(), t.numel(), out.data_ptr<double>(), cur_stream());
+    return std::sqrt(out.cpu().item<double>());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "oktopk_amd CDNA4 HIP kernels (gfx950)";
+    m.def("count_gt", &count_gt, "count |t| > tau");
+    m.def("compact_gt", &compact_gt, "ascending (idx,val) where |t| > tau");
+    m.def("kth_abs_value", &kth_abs_value, "exact k-th largest |t| via radix select");
+    m.def("scatter_add_", &scatter_add_, "dest[idx] += val");
+    m.def("zero_at_", &zero_at_, "t[idx] = 0");
+    m.def("fill_sparse_scaled_", &fill_sparse_scaled_, "out=0; out[idx]=val*scale");
+    m.def("isin_sorted", &isin_sorted, "membership of a in sorted b");
+    m.def("ef_restore_snapshot_", &ef_restore_snapshot_, "t+=r; r=t (fused)");
+    m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
+    m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
+    m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
+}

@@ -1,0 +1,382 @@
+// CDNA4 (gfx950) kernels for the Ok-Topk sparse-allreduce engine.
+//
+// Replaces the torch/CUDA op call-sites of the reference implementation
+// (inventory: SURVEY.md §2.4a):
+//   torch.topk threshold      -> 3-level histogram radix-select (kth_abs_bits)
+//   abs>tau -> nonzero/gather -> deterministic 2-pass block-scan compaction
+//   result[idx] += vals       -> scatter_add (device-scope atomics)
+//   residual EF update        -> fused ef_restore_snapshot (t+=r; r=t)
+//   SGD / Adam steps          -> fused elementwise kernels
+//   clip_grad_norm            -> l2norm partial-sum reduction
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//   * wave64: ballots are 64-bit, popcount via __popcll
+//   * memory-bound streaming kernels: fp32 scalar lane loads are 256 B/wave
+//     (full line); elementwise kernels use float4 when 16B-aligned
+//   * grid-stride with <=2048 blocks (G11); block size 256
+//   * LDS histograms (8 KiB) + one global atomicAdd per block per bin (G12)
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define BLOCK 256
+#define WAVES_PER_BLOCK (BLOCK / 64)
+#define MAX_BLOCKS 2048
+
+static inline uint32_t f32_abs_bits_host(float x) {
+    union { float f; uint32_t u; } c;
+    c.f = x;
+    return c.u & 0x7fffffffu;
+}
+
+static inline int n_blocks(int64_t work, int per_thread = 1) {
+    int64_t b = (work + (int64_t)BLOCK * per_thread - 1) / ((int64_t)BLOCK * per_thread);
+    if (b < 1) b = 1;
+    if (b > MAX_BLOCKS) b = MAX_BLOCKS;
+    return (int)b;
+}
+
+__device__ __forceinline__ uint32_t abs_bits(float x) {
+    return __float_as_uint(x) & 0x7fffffffu;
+}
+
+// ---------------------------------------------------------------------------
+// count_gt: count elements with |t| > tau
+// ---------------------------------------------------------------------------
+__global__ void count_gt_kernel(const float* __restrict__ t, int64_t n,
+                                uint32_t tau_bits, unsigned long long* __restrict__ out) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    unsigned long long cnt = 0;
+    for (; i < n; i += stride) cnt += (abs_bits(t[i]) > tau_bits);
+    // wave reduce
+    for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
+    __shared__ unsigned long long ws[WAVES_PER_BLOCK];
+    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    if (lane == 0) ws[wave] = cnt;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long s = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
+        atomicAdd(out, s);
+    }
+}
+
+extern "C" void launch_count_gt(const float* t, int64_t n, float tau,
+                                unsigned long long* out, hipStream_t stream) {
+    uint32_t tb = f32_abs_bits_host(tau);
+    hipLaunchKernelGGL(count_gt_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0, stream,
+                       t, n, tb, out);
+}
+
+// ---------------------------------------------------------------------------
+// compact_gt: deterministic (index-ordered) compaction of |t| > tau.
+// Pass A: per-block counts over contiguous chunks.
+// Pass B: stable intra-block scan (wave ballots + LDS) writing idx+val at
+//         exclusive-scanned block offsets -> output ascending by index.
+// ---------------------------------------------------------------------------
+#define COMPACT_VEC 4  // elements per thread per iteration
+
+__global__ void compact_count_kernel(const float* __restrict__ t, int64_t n,
+                                     uint32_t tau_bits, int64_t chunk,
+                                     int* __restrict__ block_counts) {
+    int64_t start = (int64_t)blockIdx.x * chunk;
+    int64_t end = (start + chunk < n) ? start + chunk : n;
+    int cnt = 0;
+    for (int64_t i = start + threadIdx.x; i < end; i += BLOCK)
+        cnt += (abs_bits(t[i]) > tau_bits);
+    for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
+    __shared__ int ws[WAVES_PER_BLOCK];
+    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    if (lane == 0) ws[wave] = cnt;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int s = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
+        block_counts[blockIdx.x] = s;
+    }
+}
+
+__global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
+                                     uint32_t tau_bits, int64_t chunk,
+                                     const int* __restrict__ block_offsets,
+                                     int32_t* __restrict__ out_idx,
+                                     float* __restrict__ out_val) {
+    int64_t start = (int64_t)blockIdx.x * chunk;
+    int64_t end = (start + chunk < n) ? start + chunk : n;
+    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    uint64_t lt_mask = ((uint64_t)1 << lane) - 1;
+    __shared__ int wsum[WAVES_PER_BLOCK];
+    __shared__ int running;
+    if (threadIdx.x == 0) running = block_offsets[blockIdx.x];
+    __syncthreads();
+
+    // iteration covers BLOCK*COMPACT_VEC consecutive elements; thread tid owns
+    // elements base+tid*4 .. +3, so (wave, lane, j) order == index order.
+    for (int64_t base = start; base < end; base += (int64_t)BLOCK * COMPACT_VEC) {
+        int64_t my = base + (int64_t)threadIdx.x * COMPACT_VEC;
+        bool p[COMPACT_VEC];
+        float v[COMPACT_VEC];
+        int own = 0;
+        #pragma unroll
+        for (int j = 0; j < COMPACT_VEC; ++j) {
+            int64_t i = my + j;
+            bool ok = i < end;
+            float x = ok ? t[i] : 0.f;
+            p[j] = ok && (abs_bits(x) > tau_bits);
+            v[j] = x;
+            own += p[j];
+        }
+        int lane_prefix = 0, wave_total = 0;
+        #pragma unroll
+        for (int j = 0; j < COMPACT_VEC; ++j) {
+            uint64_t b = __ballot(p[j]);
+            lane_prefix += __popcll(b & lt_mask);
+            wave_total += __popcll(b);
+        }
+        if (lane == 0) wsum[wave] = wave_total;
+        __syncthreads();
+        int wave_prefix = 0, iter_total = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; ++w) {
+            if (w < wave) wave_prefix += wsum[w];
+            iter_total += wsum[w];
+        }
+        int pos = running + wave_prefix + lane_prefix;
+        #pragma unroll
+        for (int j = 0; j < COMPACT_VEC; ++j) {
+            if (p[j]) {
+                out_idx[pos] = (int32_t)(my + j);
+                out_val[pos] = v[j];
+                ++pos;
+            }
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) running += iter_total;
+        __syncthreads();
+    }
+}
+
+extern "C" void launch_compact_count(const float* t, int64_t n, float tau,
+                                     int64_t chunk, int nblocks,
+                                     int* block_counts, hipStream_t stream) {
+    uint32_t tb = f32_abs_bits_host(tau);
+    hipLaunchKernelGGL(compact_count_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
+                       t, n, tb, chunk, block_counts);
+}
+
+extern "C" void launch_compact_write(const float* t, int64_t n, float tau,
+                                     int64_t chunk, int nblocks,
+                                     const int* block_offsets, int32_t* out_idx,
+                                     float* out_val, hipStream_t stream) {
+    uint32_t tb = f32_abs_bits_host(tau);
+    hipLaunchKernelGGL(compact_write_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
+                       t, n, tb, chunk, block_offsets, out_idx, out_val);
+}
+
+// ---------------------------------------------------------------------------
+// kth_abs_bits: histogram radix-select over |x| bit patterns (nonnegative
+// float bits order == uint order).  Host drives 3 levels (11/11/10 bits).
+// ---------------------------------------------------------------------------
+__global__ void hist_kernel(const float* __restrict__ t, int64_t n,
+                            uint32_t prefix_mask, uint32_t prefix_val,
+                            int shift, int nbins,
+                            unsigned int* __restrict__ ghist) {
+    extern __shared__ unsigned int lhist[];
+    for (int b = threadIdx.x; b < nbins; b += BLOCK) lhist[b] = 0;
+    __syncthreads();
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < n; i += stride) {
+        uint32_t u = abs_bits(t[i]);
+        if ((u & prefix_mask) == prefix_val)
+            atomicAdd(&lhist[(u >> shift) & (nbins - 1)], 1u);
+    }
+    __syncthreads();
+    for (int b = threadIdx.x; b < nbins; b += BLOCK)
+        if (lhist[b]) atomicAdd(&ghist[b], lhist[b]);
+}
+
+extern "C" void launch_hist(const float* t, int64_t n, uint32_t prefix_mask,
+                            uint32_t prefix_val, int shift, int nbins,
+                            unsigned int* ghist, hipStream_t stream) {
+    hipLaunchKernelGGL(hist_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK),
+                       nbins * sizeof(unsigned int), stream,
+                       t, n, prefix_mask, prefix_val, shift, nbins, ghist);
+}
+
+// ---------------------------------------------------------------------------
+// scatter ops
+// ---------------------------------------------------------------------------
+__global__ void scatter_add_kernel(float* __restrict__ dest,
+                                   const int32_t* __restrict__ idx,
+                                   const float* __restrict__ val, int64_t m) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < m; i += stride) atomicAdd(&dest[idx[i]], val[i]);
+}
+
+extern "C" void launch_scatter_add(float* dest, const int32_t* idx, const float* val,
+                                   int64_t m, hipStream_t stream) {
+    hipLaunchKernelGGL(scatter_add_kernel, dim3(n_blocks(m, 4)), dim3(BLOCK), 0, stream,
+                       dest, idx, val, m);
+}
+
+__global__ void scatter_set_scaled_kernel(float* __restrict__ dest,
+                                          const int32_t* __restrict__ idx,
+                                          const float* __restrict__ val,
+                                          float scale, int64_t m) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < m; i += stride) dest[idx[i]] = val[i] * scale;
+}
+
+extern "C" void launch_scatter_set_scaled(float* dest, const int32_t* idx,
+                                          const float* val, float scale, int64_t m,
+                                          hipStream_t stream) {
+    hipLaunchKernelGGL(scatter_set_scaled_kernel, dim3(n_blocks(m, 4)), dim3(BLOCK), 0,
+                       stream, dest, idx, val, scale, m);
+}
+
+__global__ void zero_at_kernel(float* __restrict__ dest,
+                               const int32_t* __restrict__ idx, int64_t m) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < m; i += stride) dest[idx[i]] = 0.f;
+}
+
+extern "C" void launch_zero_at(float* dest, const int32_t* idx, int64_t m,
+                               hipStream_t stream) {
+    hipLaunchKernelGGL(zero_at_kernel, dim3(n_blocks(m, 4)), dim3(BLOCK), 0, stream,
+                       dest, idx, m);
+}
+
+// ---------------------------------------------------------------------------
+// isin_sorted: binary search of each a[i] in ascending b
+// ---------------------------------------------------------------------------
+__global__ void isin_sorted_kernel(const int32_t* __restrict__ a, int64_t m,
+                                   const int32_t* __restrict__ b, int64_t nb,
+                                   bool* __restrict__ out) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < m; i += stride) {
+        int32_t x = a[i];
+        int64_t lo = 0, hi = nb;
+        while (lo < hi) {
+            int64_t mid = (lo + hi) >> 1;
+            if (b[mid] < x) lo = mid + 1; else hi = mid;
+        }
+        out[i] = (lo < nb) && (b[lo] == x);
+    }
+}
+
+extern "C" void launch_isin_sorted(const int32_t* a, int64_t m, const int32_t* b,
+                                   int64_t nb, bool* out, hipStream_t stream) {
+    hipLaunchKernelGGL(isin_sorted_kernel, dim3(n_blocks(m, 4)), dim3(BLOCK), 0, stream,
+                       a, m, b, nb, out);
+}
+
+// ---------------------------------------------------------------------------
+// fused elementwise streaming kernels (float4 when 16B-aligned)
+// ---------------------------------------------------------------------------
+__global__ void ef_restore_vec_kernel(float4* __restrict__ t, float4* __restrict__ r,
+                                      int64_t n4) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < n4; i += stride) {
+        float4 a = t[i], b = r[i];
+        a.x += b.x; a.y += b.y; a.z += b.z; a.w += b.w;
+        t[i] = a; r[i] = a;
+    }
+}
+
+__global__ void ef_restore_scalar_kernel(float* __restrict__ t, float* __restrict__ r,
+                                         int64_t lo, int64_t n) {
+    int64_t i = lo + (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < n; i += stride) {
+        float a = t[i] + r[i];
+        t[i] = a; r[i] = a;
+    }
+}
+
+extern "C" void launch_ef_restore(float* t, float* r, int64_t n, hipStream_t stream) {
+    bool aligned = (((uintptr_t)t | (uintptr_t)r) & 15) == 0;
+    int64_t n4 = aligned ? n / 4 : 0;
+    if (n4 > 0)
+        hipLaunchKernelGGL(ef_restore_vec_kernel, dim3(n_blocks(n4, 4)), dim3(BLOCK), 0,
+                           stream, (float4*)t, (float4*)r, n4);
+    if (n4 * 4 < n)
+        hipLaunchKernelGGL(ef_restore_scalar_kernel, dim3(n_blocks(n - n4 * 4, 1)),
+                           dim3(BLOCK), 0, stream, t, r, n4 * 4, n);
+}
+
+__global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
+                           float* __restrict__ buf, int64_t n, float lr, float mom,
+                           float wd, int nesterov, int use_mom) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < n; i += stride) {
+        float d = g[i] + wd * p[i];
+        if (use_mom) {
+            float b = buf[i] * mom + d;
+            buf[i] = b;
+            d = nesterov ? d + mom * b : b;
+        }
+        p[i] -= lr * d;
+    }
+}
+
+extern "C" void launch_sgd(float* p, const float* g, float* buf, int64_t n, float lr,
+                           float mom, float wd, int nesterov, hipStream_t stream) {
+    hipLaunchKernelGGL(sgd_kernel, dim3(n_blocks(n, 4)), dim3(BLOCK), 0, stream,
+                       p, g, buf, n, lr, mom, wd, nesterov, mom != 0.f);
+}
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v, int64_t n,
+                            float lr, float b1, float b2, float eps, float wd) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < n; i += stride) {
+        float gi = g[i];
+        float mi = m[i] * b1 + (1.f - b1) * gi;
+        float vi = v[i] * b2 + (1.f - b2) * gi * gi;
+        m[i] = mi; v[i] = vi;
+        float up = mi / (sqrtf(vi) + eps) + wd * p[i];
+        p[i] -= lr * up;
+    }
+}
+
+extern "C" void launch_adam(float* p, const float* g, float* m, float* v, int64_t n,
+                            float lr, float b1, float b2, float eps, float wd,
+                            hipStream_t stream) {
+    hipLaunchKernelGGL(adam_kernel, dim3(n_blocks(n, 4)), dim3(BLOCK), 0, stream,
+                       p, g, m, v, n, lr, b1, b2, eps, wd);
+}
+
+// ---------------------------------------------------------------------------
+// l2 norm (sum of squares in double, sqrt on host)
+// ---------------------------------------------------------------------------
+__global__ void sumsq_kernel(const float* __restrict__ t, int64_t n,
+                             double* __restrict__ out) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    double acc = 0.0;
+    for (; i < n; i += stride) { double x = t[i]; acc += x * x; }
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, 64);
+    __shared__ double ws[WAVES_PER_BLOCK];
+    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    if (lane == 0) ws[wave] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        double s = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
+        atomicAdd(out, s);
+    }
+}
+
+extern "C" void launch_sumsq(const float* t, int64_t n, double* out, hipStream_t stream) {
+    hipLaunchKernelGGL(sumsq_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0, stream,
+                       t, n, out);
+}

@@ -1,0 +1,158 @@
+"""Training driver with synthetic data (bench + smoke path).
+
+Capability parity with the reference trainers (VGG/dl_trainer.py DLTrainer,
+BERT/bert/main_bert.py): model build, fwd/bwd, distributed optimizer, LR
+schedule hooks, per-phase timing.  Real-dataset loaders are out of scope for
+the benchmark contract (no network): batches are synthetic with the exact
+shapes of the reference configs.
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+
+from . import models
+from .comm import Comm
+from .config import EngineConfig
+from .optimizer import DistributedOptimizer, FlatBertAdam
+
+
+class SyntheticBatches:
+    """Deterministic per-rank synthetic batches with reference shapes."""
+
+    def __init__(self, model_name: str, batch_size: int, device, seq_len: int = 128,
+                 rank: int = 0, vocab_size: int = 30522):
+        self.model_name = model_name
+        self.bs = batch_size
+        self.device = device
+        self.seq_len = seq_len
+        self.vocab = vocab_size
+        g = torch.Generator().manual_seed(1234 + rank)
+        if model_name == "vgg16":
+            # CIFAR-10 shape, bs 16/rank in the reference (vgg16_oktopk.sh)
+            self.x = torch.randn(batch_size, 3, 32, 32, generator=g)
+            self.y = torch.randint(0, 10, (batch_size,), generator=g)
+        elif model_name == "lstman4":
+            # AN4 spectrograms: (N, 1, freq=161, time~200), bs 2/rank
+            self.x = torch.randn(batch_size, 1, 161, 201, generator=g)
+            t_out = 101  # conv time stride 2 once
+            self.target = torch.randint(1, 29, (batch_size, 20), generator=g)
+            self.in_len = torch.full((batch_size,), t_out, dtype=torch.int32)
+            self.tgt_len = torch.full((batch_size,), 20, dtype=torch.int32)
+        elif model_name.startswith("bert"):
+            # seq 128, bs 8/rank (bert_oktopk.sh)
+            self.input_ids = torch.randint(0, vocab_size, (batch_size, seq_len), generator=g)
+            self.token_type = torch.randint(0, 2, (batch_size, seq_len), generator=g)
+            self.attn = torch.ones(batch_size, seq_len, dtype=torch.long)
+            labels = torch.full((batch_size, seq_len), -1, dtype=torch.long)
+            mask_pos = torch.rand(batch_size, seq_len, generator=g) < 0.15
+            labels[mask_pos] = torch.randint(0, vocab_size, (int(mask_pos.sum()),), generator=g)
+            self.mlm_labels = labels
+            self.nsp = torch.randint(0, 2, (batch_size,), generator=g)
+        else:
+            raise ValueError(model_name)
+        for k, v in list(self.__dict__.items()):
+            if isinstance(v, torch.Tensor):
+                setattr(self, k, v.to(device))
+
+
+class Trainer:
+    def __init__(
+        self,
+        model_name: str = "bert_base",
+        batch_size: int = 8,
+        seq_len: int = 128,
+        device: Optional[torch.device] = None,
+        comm: Optional[Comm] = None,
+        cfg: Optional[EngineConfig] = None,
+        optimizer: str = "auto",  # auto | sgd | adam
+        lr: float = None,
+        dtype: str = "bf16",
+        nsteps_update: int = 1,
+        model_kwargs: Optional[dict] = None,
+    ):
+        self.comm = comm or Comm(None)
+        self.device = device or (
+            torch.device("cuda", torch.cuda.current_device())
+            if torch.cuda.is_available()
+            else torch.device("cpu")
+        )
+        self.model_name = model_name
+        self.cfg = cfg or EngineConfig.preset(
+            "bert" if model_name.startswith("bert") else
+            ("lstm" if model_name == "lstman4" else "vgg")
+        )
+        self.autocast = dtype == "bf16" and self.device.type == "cuda"
+        self.nsteps_update = max(1, nsteps_update)
+
+        self.model = models.create_net(model_name, **(model_kwargs or {})).to(self.device)
+        # broadcast initial weights (reference comm.bcast(state_dict),
+        # VGG/main_trainer.py:52)
+        if self.comm.size > 1:
+            for p in self.model.parameters():
+                self.comm.broadcast_(p.data, src=0)
+
+        if optimizer == "auto":
+            optimizer = "adam" if model_name.startswith("bert") else "sgd"
+        self.opt_kind = optimizer
+        if optimizer == "adam":
+            self.opt = FlatBertAdam(
+                self.model.named_parameters(),
+                comm=self.comm,
+                cfg=self.cfg,
+                lr=lr or 2e-4,
+            )
+        else:
+            inner = torch.optim.SGD(
+                self.model.parameters(), lr=lr or 0.1, momentum=0.9, weight_decay=5e-4
+            )
+            self.opt = DistributedOptimizer(
+                inner, self.model.named_parameters(), comm=self.comm, cfg=self.cfg
+            )
+
+        self.batches = SyntheticBatches(
+            model_name, batch_size, self.device, seq_len=seq_len, rank=self.comm.rank
+        )
+        self.iteration = 0
+        self.last_loss = 0.0
+
+    # ------------------------------------------------------------------
+    def _forward_loss(self) -> torch.Tensor:
+        b = self.batches
+        if self.model_name == "vgg16":
+            out = self.model(b.x)
+            return torch.nn.functional.cross_entropy(out, b.y)
+        if self.model_name == "lstman4":
+            logits = self.model(b.x)  # (T, N, C)
+            logp = torch.nn.functional.log_softmax(logits, dim=-1)
+            t = logp.size(0)
+            in_len = torch.clamp(b.in_len, max=t)
+            return torch.nn.functional.ctc_loss(
+                logp.float(), b.target, in_len, b.tgt_len, blank=0, zero_infinity=True
+            )
+        return self.model(
+            b.input_ids,
+            token_type_ids=b.token_type,
+            attention_mask=b.attn,
+            masked_lm_labels=b.mlm_labels,
+            next_sentence_label=b.nsp,
+        )
+
+    def step(self) -> float:
+        """One optimizer step (with nsteps_update grad-accumulation substeps)."""
+        self.opt.zero_grad()
+        for sub in range(self.nsteps_update):
+            if hasattr(self.opt, "local"):
+                self.opt.local = sub < self.nsteps_update - 1
+            if self.autocast:
+                with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+                    loss = self._forward_loss()
+            else:
+                loss = self._forward_loss()
+            loss.backward()
+        self.opt.step()
+        self.iteration += 1
+        self.last_loss = float(loss.detach().float().item())
+        return self.last_loss
