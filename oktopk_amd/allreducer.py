@@ -290,7 +290,8 @@ class AllReducer:
         s2 = time.perf_counter()
         reduced = torch.zeros(hi - lo, dtype=t.dtype, device=t.device)
         if r_idx.numel():
-            ops.scatter_add_(reduced, r_idx.to(t.device).long() - lo, r_val.to(t.device))
+            # int32 arithmetic keeps the wire dtype end-to-end (n < 2^31)
+            ops.scatter_add_(reduced, r_idx.to(t.device) - lo, r_val.to(t.device))
         self._time(name, "reduce", time.perf_counter() - s2)
 
         # --- 4. round 2: global top-k + sparse allgather -----------------
@@ -544,7 +545,7 @@ class AllReducer:
         s2 = time.perf_counter()
         reduced = torch.zeros(hi - lo, dtype=t.dtype, device=t.device)
         if r_idx.numel():
-            ops.scatter_add_(reduced, r_idx.to(t.device).long() - lo, r_val.to(t.device))
+            ops.scatter_add_(reduced, r_idx.to(t.device) - lo, r_val.to(t.device))
         gidx, gval = ops.compact_gt(reduced, 0.0)
         self._time(name, "reduce", time.perf_counter() - s2)
 

@@ -23,10 +23,17 @@
 #define WAVES_PER_BLOCK (BLOCK / 64)
 #define MAX_BLOCKS 2048
 
-static inline uint32_t f32_abs_bits_host(float x) {
+static inline uint32_t tau_to_bits(float tau) {
+    // |x| > tau as a uint compare on abs bit patterns; tau < 0 selects ALL
+    // elements (including zeros), flagged by the sentinel 0xFFFFFFFF.
+    if (tau < 0.f) return 0xFFFFFFFFu;
     union { float f; uint32_t u; } c;
-    c.f = x;
+    c.f = tau;
     return c.u & 0x7fffffffu;
+}
+
+__device__ __forceinline__ bool sel_gt(uint32_t abits, uint32_t tau_bits) {
+    return (tau_bits == 0xFFFFFFFFu) | (abits > tau_bits);
 }
 
 static inline int n_blocks(int64_t work, int per_thread = 1) {
@@ -48,7 +55,7 @@ __global__ void count_gt_kernel(const float* __restrict__ t, int64_t n,
     int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * BLOCK;
     unsigned long long cnt = 0;
-    for (; i < n; i += stride) cnt += (abs_bits(t[i]) > tau_bits);
+    for (; i < n; i += stride) cnt += sel_gt(abs_bits(t[i]), tau_bits);
     // wave reduce
     for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
     __shared__ unsigned long long ws[WAVES_PER_BLOCK];
@@ -64,7 +71,7 @@ __global__ void count_gt_kernel(const float* __restrict__ t, int64_t n,
 
 extern "C" void launch_count_gt(const float* t, int64_t n, float tau,
                                 unsigned long long* out, hipStream_t stream) {
-    uint32_t tb = f32_abs_bits_host(tau);
+    uint32_t tb = tau_to_bits(tau);
     hipLaunchKernelGGL(count_gt_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0, stream,
                        t, n, tb, out);
 }
@@ -84,7 +91,7 @@ __global__ void compact_count_kernel(const float* __restrict__ t, int64_t n,
     int64_t end = (start + chunk < n) ? start + chunk : n;
     int cnt = 0;
     for (int64_t i = start + threadIdx.x; i < end; i += BLOCK)
-        cnt += (abs_bits(t[i]) > tau_bits);
+        cnt += sel_gt(abs_bits(t[i]), tau_bits);
     for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
     __shared__ int ws[WAVES_PER_BLOCK];
     int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
@@ -123,7 +130,7 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
             int64_t i = my + j;
             bool ok = i < end;
             float x = ok ? t[i] : 0.f;
-            p[j] = ok && (abs_bits(x) > tau_bits);
+            p[j] = ok && sel_gt(abs_bits(x), tau_bits);
             v[j] = x;
             own += p[j];
         }
@@ -159,7 +166,7 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
 extern "C" void launch_compact_count(const float* t, int64_t n, float tau,
                                      int64_t chunk, int nblocks,
                                      int* block_counts, hipStream_t stream) {
-    uint32_t tb = f32_abs_bits_host(tau);
+    uint32_t tb = tau_to_bits(tau);
     hipLaunchKernelGGL(compact_count_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
                        t, n, tb, chunk, block_counts);
 }
@@ -168,7 +175,7 @@ extern "C" void launch_compact_write(const float* t, int64_t n, float tau,
                                      int64_t chunk, int nblocks,
                                      const int* block_offsets, int32_t* out_idx,
                                      float* out_val, hipStream_t stream) {
-    uint32_t tb = f32_abs_bits_host(tau);
+    uint32_t tb = tau_to_bits(tau);
     hipLaunchKernelGGL(compact_write_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
                        t, n, tb, chunk, block_offsets, out_idx, out_val);
 }
