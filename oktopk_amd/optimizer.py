@@ -262,20 +262,37 @@ class FlatBertAdam:
         self.step_count = 0
         device = self.params[0].device
         self.numel = sum(p.numel() for p in self.params)
+
+        # Fully-flat layout: params are reordered [decay group | no-decay
+        # group] so the Adam step is TWO fused kernel launches over
+        # contiguous slices instead of one launch per parameter (the
+        # reference pays a python loop over ~200 params,
+        # optimization.py:170-224, plus a torch.cat of 109.5M floats per
+        # step at :160-163 — here both gather and update are zero-copy).
+        def _decays(n: str) -> bool:
+            low = n.lower()
+            return not any(k in low for k in no_decay_keys)
+
+        order = sorted(range(len(named)), key=lambda i: 0 if _decays(named[i][0]) else 1)
+        self.names = [named[i][0] for i in order]
+        self.params = [named[i][1] for i in order]
+        self.decay_numel = sum(
+            named[i][1].numel() for i in order if _decays(named[i][0])
+        )
         self.flat_grad = torch.zeros(self.numel, dtype=torch.float32, device=device)
+        self.flat_param = torch.zeros_like(self.flat_grad)
         self.exp_avg = torch.zeros_like(self.flat_grad)
         self.exp_avg_sq = torch.zeros_like(self.flat_grad)
         self.slots = []
-        self.decay_mask = []
         off = 0
-        for n, p in named:
-            self.slots.append((off, p.numel()))
-            low = n.lower()
-            self.decay_mask.append(not any(k in low for k in no_decay_keys))
-            off += p.numel()
-        # point grads at the flat buffer: backward accumulates in place and
-        # the whole-model reduce needs no gather (reference pays a
-        # torch.cat of 109.5M floats every step, optimization.py:160-163)
+        for p in self.params:
+            n_ = p.numel()
+            self.slots.append((off, n_))
+            self.flat_param[off : off + n_].copy_(p.data.view(-1))
+            # alias param storage into the flat buffer (module keeps the
+            # same Parameter objects; their .data becomes a view)
+            p.data = self.flat_param[off : off + n_].view_as(p)
+            off += n_
         for p, (o, n_) in zip(self.params, self.slots):
             p.grad = self.flat_grad[o : o + n_].view_as(p)
 
@@ -297,21 +314,19 @@ class FlatBertAdam:
             gn = ops.l2norm(self.flat_grad)
             if gn > self.max_grad_norm:
                 self.flat_grad.mul_(self.max_grad_norm / (gn + 1e-6))
-        # 3. fused Adam per param slice
+        # 3. fused Adam: one launch per weight-decay group (2 total)
         lr = self.current_lr()
         b1, b2 = self.betas
-        for p, (o, n_), decay in zip(self.params, self.slots, self.decay_mask):
-            wd = self.weight_decay if decay else 0.0
+        d = self.decay_numel
+        if d > 0:
             ops.fused_adam_(
-                p.data.view(-1),
-                self.flat_grad[o : o + n_],
-                self.exp_avg[o : o + n_],
-                self.exp_avg_sq[o : o + n_],
-                lr,
-                b1,
-                b2,
-                self.eps,
-                wd,
+                self.flat_param[:d], self.flat_grad[:d], self.exp_avg[:d],
+                self.exp_avg_sq[:d], lr, b1, b2, self.eps, self.weight_decay,
+            )
+        if d < self.numel:
+            ops.fused_adam_(
+                self.flat_param[d:], self.flat_grad[d:], self.exp_avg[d:],
+                self.exp_avg_sq[d:], lr, b1, b2, self.eps, 0.0,
             )
         self.step_count += 1
 
