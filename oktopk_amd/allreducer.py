@@ -205,15 +205,20 @@ class AllReducer:
         )
 
     def _adaptive_bump(self, t: torch.Tensor, tau: float, k: int) -> float:
-        """Raise tau while too many elements are selected
-        (reference add2residual, VGG/compression.py:384-404)."""
+        """Raise tau while too many elements are selected (reference
+        add2residual, VGG/compression.py:384-404).  All candidate thresholds
+        tau * scale^i are counted in ONE kernel pass (count_multi_gt) — the
+        sequential 5-iteration loop of the reference costs one full tensor
+        read plus a device sync per iteration."""
         ok = self.cfg.oktopk
-        for _ in range(ok.bump_max_loops):
-            if ops.count_gt(t, tau) > 4 * k // 3:
-                tau *= ok.bump_scale
-            else:
-                break
-        return tau
+        m = ok.bump_max_loops
+        taus = [tau * ok.bump_scale ** i for i in range(m + 1)]
+        counts = ops.count_multi_gt(t, taus)
+        # loop semantics: bump while count(current) > 4k/3, at most m times
+        i = 0
+        while i < m and counts[i] > 4 * k // 3:
+            i += 1
+        return taus[i]
 
     # -- Ok-Topk (SURVEY.md section 2.5) --------------------------------
     def _oktopk(self, name: str, t: torch.Tensor, st: TensorState) -> torch.Tensor:

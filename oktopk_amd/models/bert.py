@@ -117,8 +117,9 @@ class BertModel(nn.Module):
     def forward(self, input_ids, token_type_ids=None, attention_mask=None):
         mask = None
         if attention_mask is not None:
-            # (b, s) {0,1} -> additive (b, 1, 1, s)
-            mask = (1.0 - attention_mask[:, None, None, :].float()) * -10000.0
+            # (b, s) {0,1} -> additive (b, 1, 1, s), in the compute dtype
+            wdt = self.embeddings.word_embeddings.weight.dtype
+            mask = (1.0 - attention_mask[:, None, None, :].to(wdt)) * -10000.0
         x = self.embeddings(input_ids, token_type_ids)
         for lyr in self.layer:
             x = lyr(x, mask)

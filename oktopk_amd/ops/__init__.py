@@ -65,6 +65,10 @@ def count_gt(t: torch.Tensor, tau: float) -> int:
     return int(_backend(t).count_gt(t, float(tau)))
 
 
+def count_multi_gt(t: torch.Tensor, taus) -> list:
+    return [int(x) for x in _backend(t).count_multi_gt(t, [float(x) for x in taus])]
+
+
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
     return _backend(dest).scatter_add_(dest, idx, val)
 

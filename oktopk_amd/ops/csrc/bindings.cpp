@@ -12,6 +12,8 @@
 // launchers from kernels.hip
 extern "C" {
 void launch_count_gt(const float*, int64_t, float, unsigned long long*, hipStream_t);
+void launch_count_multi_gt(const float*, int64_t, const float*, int, unsigned long long*,
+                           hipStream_t);
 void launch_compact_count(const float*, int64_t, float, int64_t, int, int*, hipStream_t);
 void launch_compact_write(const float*, int64_t, float, int64_t, int, const int*,
                           int32_t*, float*, hipStream_t);
@@ -59,6 +61,22 @@ static int64_t count_gt(torch::Tensor t, double tau) {
                     reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()),
                     cur_stream());
     return out.cpu().item<int64_t>();
+}
+
+static std::vector<int64_t> count_multi_gt(torch::Tensor t, std::vector<double> taus) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    TORCH_CHECK(taus.size() >= 1 && taus.size() <= 8, "1..8 thresholds");
+    float tf[8];
+    for (size_t j = 0; j < taus.size(); ++j) tf[j] = (float)taus[j];
+    auto out = torch::zeros({(int64_t)taus.size()}, t.options().dtype(torch::kInt64));
+    launch_count_multi_gt(t.data_ptr<float>(), t.numel(), tf, (int)taus.size(),
+                          reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()),
+                          cur_stream());
+    auto cpu = out.cpu();
+    std::vector<int64_t> res(taus.size());
+    for (size_t j = 0; j < taus.size(); ++j) res[j] = cpu[j].item<int64_t>();
+    return res;
 }
 
 static std::vector<torch::Tensor> compact_gt(torch::Tensor t, double tau) {
@@ -218,6 +236,7 @@ static double l2norm(torch::Tensor t) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "oktopk_amd CDNA4 HIP kernels (gfx950)";
     m.def("count_gt", &count_gt, "count |t| > tau");
+    m.def("count_multi_gt", &count_multi_gt, "counts for up to 8 thresholds, one pass");
     m.def("compact_gt", &compact_gt, "ascending (idx,val) where |t| > tau");
     m.def("kth_abs_value", &kth_abs_value, "exact k-th largest |t| via radix select");
     m.def("scatter_add_", &scatter_add_, "dest[idx] += val");

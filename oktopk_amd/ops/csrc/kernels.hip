@@ -50,12 +50,19 @@ __device__ __forceinline__ uint32_t abs_bits(float x) {
 // ---------------------------------------------------------------------------
 // count_gt: count elements with |t| > tau
 // ---------------------------------------------------------------------------
-__global__ void count_gt_kernel(const float* __restrict__ t, int64_t n,
+__global__ void count_gt_kernel(const float* __restrict__ t, int64_t n, int64_t n4,
                                 uint32_t tau_bits, unsigned long long* __restrict__ out) {
+    const float4* t4 = reinterpret_cast<const float4*>(t);
     int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * BLOCK;
     unsigned long long cnt = 0;
-    for (; i < n; i += stride) cnt += sel_gt(abs_bits(t[i]), tau_bits);
+    for (; i < n4; i += stride) {
+        float4 v = t4[i];
+        cnt += sel_gt(abs_bits(v.x), tau_bits) + sel_gt(abs_bits(v.y), tau_bits)
+             + sel_gt(abs_bits(v.z), tau_bits) + sel_gt(abs_bits(v.w), tau_bits);
+    }
+    for (i = n4 * 4 + (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride)
+        cnt += sel_gt(abs_bits(t[i]), tau_bits);
     // wave reduce
     for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
     __shared__ unsigned long long ws[WAVES_PER_BLOCK];
@@ -72,8 +79,54 @@ __global__ void count_gt_kernel(const float* __restrict__ t, int64_t n,
 extern "C" void launch_count_gt(const float* t, int64_t n, float tau,
                                 unsigned long long* out, hipStream_t stream) {
     uint32_t tb = tau_to_bits(tau);
+    int64_t n4 = ((uintptr_t)t & 15) == 0 ? n >> 2 : 0;
     hipLaunchKernelGGL(count_gt_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0, stream,
-                       t, n, tb, out);
+                       t, n, n4, tb, out);
+}
+
+// count_multi_gt: counts for up to 8 candidate thresholds in ONE pass
+// (feeds the adaptive-bump loop of add2residual, VGG/compression.py:384-404,
+// with a single kernel + single D2H instead of one pass per candidate).
+struct TauSet { uint32_t tb[8]; int n; };
+
+__global__ void count_multi_gt_kernel(const float* __restrict__ t, int64_t n,
+                                      TauSet taus,
+                                      unsigned long long* __restrict__ out) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    unsigned long long cnt[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) cnt[j] = 0;
+    for (; i < n; i += stride) {
+        uint32_t u = abs_bits(t[i]);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            if (j < taus.n) cnt[j] += sel_gt(u, taus.tb[j]);
+    }
+    __shared__ unsigned long long ws[WAVES_PER_BLOCK];
+    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    for (int j = 0; j < taus.n; ++j) {
+        unsigned long long c = cnt[j];
+        for (int off = 32; off > 0; off >>= 1) c += __shfl_down(c, off, 64);
+        if (lane == 0) ws[wave] = c;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            unsigned long long s = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
+            atomicAdd(&out[j], s);
+        }
+        __syncthreads();
+    }
+}
+
+extern "C" void launch_count_multi_gt(const float* t, int64_t n, const float* taus,
+                                      int ntau, unsigned long long* out,
+                                      hipStream_t stream) {
+    TauSet ts;
+    ts.n = ntau;
+    for (int j = 0; j < 8; ++j) ts.tb[j] = j < ntau ? tau_to_bits(taus[j]) : 0;
+    hipLaunchKernelGGL(count_multi_gt_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0,
+                       stream, t, n, ts, out);
 }
 
 // ---------------------------------------------------------------------------
@@ -364,12 +417,21 @@ extern "C" void launch_adam(float* p, const float* g, float* m, float* v, int64_
 // ---------------------------------------------------------------------------
 // l2 norm (sum of squares in double, sqrt on host)
 // ---------------------------------------------------------------------------
-__global__ void sumsq_kernel(const float* __restrict__ t, int64_t n,
+__global__ void sumsq_kernel(const float* __restrict__ t, int64_t n, int64_t n4,
                              double* __restrict__ out) {
+    const float4* t4 = reinterpret_cast<const float4*>(t);
     int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * BLOCK;
     double acc = 0.0;
-    for (; i < n; i += stride) { double x = t[i]; acc += x * x; }
+    for (; i < n4; i += stride) {
+        float4 v = t4[i];
+        acc += (double)v.x * v.x + (double)v.y * v.y
+             + (double)v.z * v.z + (double)v.w * v.w;
+    }
+    for (i = n4 * 4 + (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
+        double x = t[i];
+        acc += x * x;
+    }
     for (int off = 32; off > 0; off >>= 1)
         acc += __shfl_down(acc, off, 64);
     __shared__ double ws[WAVES_PER_BLOCK];
@@ -384,6 +446,7 @@ __global__ void sumsq_kernel(const float* __restrict__ t, int64_t n,
 }
 
 extern "C" void launch_sumsq(const float* t, int64_t n, double* out, hipStream_t stream) {
+    int64_t n4 = ((uintptr_t)t & 15) == 0 ? n >> 2 : 0;
     hipLaunchKernelGGL(sumsq_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0, stream,
-                       t, n, out);
+                       t, n, n4, out);
 }

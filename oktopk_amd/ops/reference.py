@@ -19,6 +19,7 @@ __all__ = [
     "kth_abs_value",
     "compact_gt",
     "count_gt",
+    "count_multi_gt",
     "scatter_add_",
     "zero_at_",
     "fill_sparse_scaled_",
@@ -50,6 +51,11 @@ def compact_gt(t: torch.Tensor, tau: float) -> Tuple[torch.Tensor, torch.Tensor]
 
 def count_gt(t: torch.Tensor, tau: float) -> int:
     return int((t.reshape(-1).abs() > tau).sum().item())
+
+
+def count_multi_gt(t: torch.Tensor, taus) -> list:
+    a = t.reshape(-1).abs()
+    return [int((a > float(x)).sum().item()) for x in taus]
 
 
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:

@@ -279,27 +279,41 @@ class FlatBertAdam:
         self.decay_numel = sum(
             named[i][1].numel() for i in order if _decays(named[i][0])
         )
+        # Mixed precision: when the model runs in pure bf16 (no autocast —
+        # the per-layer weight-cast kernels of autocast were ~1000 extra
+        # launches/step on BERT-base, measured in profiles/), keep fp32
+        # MASTER weights + Adam state here and mirror to the bf16 model
+        # copy with one bulk cast per step.
+        self.model_dtype = self.params[0].dtype
         self.flat_grad = torch.zeros(self.numel, dtype=torch.float32, device=device)
         self.flat_param = torch.zeros_like(self.flat_grad)
         self.exp_avg = torch.zeros_like(self.flat_grad)
         self.exp_avg_sq = torch.zeros_like(self.flat_grad)
+        if self.model_dtype != torch.float32:
+            self.flat_param_model = torch.zeros(self.numel, dtype=self.model_dtype, device=device)
+            self.flat_grad_model = torch.zeros(self.numel, dtype=self.model_dtype, device=device)
+        else:
+            self.flat_param_model = self.flat_param
+            self.flat_grad_model = self.flat_grad
         self.slots = []
         off = 0
         for p in self.params:
             n_ = p.numel()
             self.slots.append((off, n_))
-            self.flat_param[off : off + n_].copy_(p.data.view(-1))
-            # alias param storage into the flat buffer (module keeps the
-            # same Parameter objects; their .data becomes a view)
-            p.data = self.flat_param[off : off + n_].view_as(p)
+            self.flat_param[off : off + n_].copy_(p.data.view(-1).float())
+            # alias param storage into the flat model-dtype buffer (module
+            # keeps the same Parameter objects; their .data becomes a view)
+            p.data = self.flat_param_model[off : off + n_].view_as(p)
             off += n_
+        if self.model_dtype != torch.float32:
+            self.flat_param_model.copy_(self.flat_param)
         for p, (o, n_) in zip(self.params, self.slots):
-            p.grad = self.flat_grad[o : o + n_].view_as(p)
+            p.grad = self.flat_grad_model[o : o + n_].view_as(p)
 
     def zero_grad(self, set_to_none: bool = False):
-        self.flat_grad.zero_()
+        self.flat_grad_model.zero_()
         for p, (o, n_) in zip(self.params, self.slots):
-            p.grad = self.flat_grad[o : o + n_].view_as(p)
+            p.grad = self.flat_grad_model[o : o + n_].view_as(p)
 
     def current_lr(self) -> float:
         if self.t_total > 0:
@@ -308,7 +322,15 @@ class FlatBertAdam:
 
     def step(self):
         # 1. sparse allreduce of the whole flat gradient
-        self.reducer.run("flat", self.flat_grad)
+        if self.cfg.compressor in ("dense", "none") and self.model_dtype != torch.float32:
+            # dense baseline: allreduce the bf16 grads directly (half wire
+            # volume — the right dense baseline on MI355X), then upcast
+            self.reducer.run("flat", self.flat_grad_model)
+            self.flat_grad.copy_(self.flat_grad_model)
+        else:
+            if self.flat_grad_model is not self.flat_grad:
+                self.flat_grad.copy_(self.flat_grad_model)  # one bulk upcast
+            self.reducer.run("flat", self.flat_grad)
         # 2. grad clip on the reduced gradient (reference optimization.py:197)
         if self.max_grad_norm and self.max_grad_norm > 0:
             gn = ops.l2norm(self.flat_grad)
@@ -328,6 +350,9 @@ class FlatBertAdam:
                 self.flat_param[d:], self.flat_grad[d:], self.exp_avg[d:],
                 self.exp_avg_sq[d:], lr, b1, b2, self.eps, 0.0,
             )
+        # 4. mirror fp32 master -> model-dtype weights (one bulk cast)
+        if self.flat_param_model is not self.flat_param:
+            self.flat_param_model.copy_(self.flat_param)
         self.step_count += 1
 
     def state_dict(self):

@@ -84,7 +84,13 @@ class Trainer:
             "bert" if model_name.startswith("bert") else
             ("lstm" if model_name == "lstman4" else "vgg")
         )
-        self.autocast = dtype == "bf16" and self.device.type == "cuda"
+        # BERT runs in PURE bf16 with fp32 master weights in FlatBertAdam
+        # (autocast's per-layer weight casts are ~1000 extra kernel launches
+        # per step — measured in profiles/); conv/LSTM recipes use autocast.
+        self.pure_bf16 = (
+            dtype == "bf16" and self.device.type == "cuda" and model_name.startswith("bert")
+        )
+        self.autocast = dtype == "bf16" and self.device.type == "cuda" and not self.pure_bf16
         self.nsteps_update = max(1, nsteps_update)
 
         self.model = models.create_net(model_name, **(model_kwargs or {})).to(self.device)
@@ -93,6 +99,8 @@ class Trainer:
         if self.comm.size > 1:
             for p in self.model.parameters():
                 self.comm.broadcast_(p.data, src=0)
+        if self.pure_bf16:
+            self.model = self.model.to(torch.bfloat16)
 
         if optimizer == "auto":
             optimizer = "adam" if model_name.startswith("bert") else "sgd"
