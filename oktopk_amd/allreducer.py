@@ -248,7 +248,7 @@ class AllReducer:
             if m >= P:
                 step = m // P
                 pos = torch.arange(1, P, dtype=torch.int64, device=idx.device) * step
-                q = idx.long()[pos].to(comm.device)
+                q = comm.to_comm(idx.long()[pos])
                 comm.allreduce_(q)
                 q = (q // P).cpu()
                 b = torch.empty(P + 1, dtype=torch.int64)
@@ -286,7 +286,7 @@ class AllReducer:
                 segs.append(val[a:b_].view(torch.int32))
             send = torch.cat(segs) if segs else idx.view(torch.int32)[:0]
             recv_counts = comm.alltoall_sizes(send_counts, comm.device)
-            recv = comm.alltoallv(send.to(comm.device), send_counts, recv_counts)
+            recv = comm.alltoallv(comm.to_comm(send), send_counts, recv_counts)
             r_idx, r_val = self._unpack(recv, recv_counts)
         else:
             r_idx, r_val = idx, val
@@ -308,7 +308,7 @@ class AllReducer:
             gidx, gval = ops.compact_gt(reduced, st.tau_global)
         gidx = gidx + lo  # absolute indices (int32 + int offset)
 
-        pack = self._pack(gidx, gval).to(comm.device)
+        pack = comm.to_comm(self._pack(gidx, gval))
         buf, sizes = comm.allgatherv(pack)
         all_idx, all_val = self._unpack(buf, sizes)
         all_idx = all_idx.to(t.device)
@@ -377,7 +377,7 @@ class AllReducer:
         self._time(name, "compress", time.perf_counter() - s0)
 
         s1 = time.perf_counter()
-        pack = self._pack(idx, val).to(comm.device)
+        pack = comm.to_comm(self._pack(idx, val))
         buf = comm.allgather_eq(pack)
         self._time(name, "allgather", time.perf_counter() - s1)
 
@@ -421,7 +421,7 @@ class AllReducer:
         self._time(name, "compress", time.perf_counter() - s0)
 
         s1 = time.perf_counter()
-        pack = self._pack(idx, val).to(comm.device)
+        pack = comm.to_comm(self._pack(idx, val))
         buf, sizes = comm.allgatherv(pack)
         all_idx, all_val = self._unpack(buf, sizes)
         self._time(name, "allgather", time.perf_counter() - s1)
@@ -461,7 +461,7 @@ class AllReducer:
         self._time(name, "compress", time.perf_counter() - s0)
 
         s1 = time.perf_counter()
-        pack = self._pack(idx, val).to(comm.device)
+        pack = comm.to_comm(self._pack(idx, val))
         buf, sizes = comm.allgatherv(pack)
         all_idx, all_val = self._unpack(buf, sizes)
         self._time(name, "allgather", time.perf_counter() - s1)
@@ -541,7 +541,7 @@ class AllReducer:
                 segs.append(val[a:b_].view(torch.int32))
             send = torch.cat(segs)
             recv_counts = comm.alltoall_sizes(send_counts, comm.device)
-            recv = comm.alltoallv(send.to(comm.device), send_counts, recv_counts)
+            recv = comm.alltoallv(comm.to_comm(send), send_counts, recv_counts)
             r_idx, r_val = self._unpack(recv, recv_counts)
         else:
             r_idx, r_val = idx, val
@@ -572,7 +572,7 @@ class AllReducer:
             result.div_(P)
         else:
             gidx = gidx + lo
-            pack = self._pack(gidx, gval).to(comm.device)
+            pack = comm.to_comm(self._pack(gidx, gval))
             buf, sizes = comm.allgatherv(pack)
             all_idx, all_val = self._unpack(buf, sizes)
             ops.fill_sparse_scaled_(
@@ -605,7 +605,7 @@ class AllReducer:
         if P > 1:
             if P & (P - 1):
                 raise ValueError("gtopk requires a power-of-two world size")
-            cur_idx, cur_val = idx.to(comm.device), val.to(comm.device)
+            cur_idx, cur_val = comm.to_comm(idx), comm.to_comm(val)
             nrounds = int(math.log2(P))
             peer_dist = 1
             alive = True
@@ -623,8 +623,8 @@ class AllReducer:
                         ops.scatter_add_(merged, cur_idx.to(t.device), cur_val.to(t.device))
                         ops.scatter_add_(merged, o_idx.to(t.device), o_val.to(t.device))
                         topm = torch.topk(merged.abs(), k, sorted=False)
-                        cur_idx = topm.indices.to(torch.int32).to(comm.device)
-                        cur_val = merged[topm.indices].to(comm.device)
+                        cur_idx = comm.to_comm(topm.indices.to(torch.int32))
+                        cur_val = comm.to_comm(merged[topm.indices])
                     else:
                         sr = comm.isend(pack, dst=peer, tag=7)
                         sr.wait()

@@ -92,6 +92,16 @@ class Comm:
             return torch.device("cuda", torch.cuda.current_device())
         return torch.device("cpu")
 
+    def to_comm(self, t: torch.Tensor) -> torch.Tensor:
+        """Move a payload to the comm device — identity for world-of-1.
+
+        (A naive `.to(cpu)` round-trip at world-1 costs intermittent ~85 ms
+        pageable-copy stalls on ROCm — measured; never move when no
+        communication will happen.)"""
+        if self._size == 1:
+            return t
+        return t.to(self.device)
+
     # -- collectives ------------------------------------------------------
     def allreduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         if self._size == 1:
