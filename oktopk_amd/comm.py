@@ -42,6 +42,8 @@ def init_from_env(backend: Optional[str] = None, timeout_s: float = 300.0) -> "C
         return Comm(group=None)
     if not dist.is_initialized():
         if backend is None:
+            backend = os.environ.get("OKTOPK_BACKEND")
+        if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         rank = int(os.environ.get("RANK", "0"))
         local_rank = int(os.environ.get("LOCAL_RANK", "0"))
