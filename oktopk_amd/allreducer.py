@@ -103,6 +103,13 @@ class AllReducer:
         self.timers: Dict[str, Dict[str, float]] = {}
         self.eps_log: List[Tuple[int, float]] = []
 
+    def set_comm(self, comm: Comm) -> None:
+        """Swap the communicator after an elastic shrink; per-world state
+        (region boundaries) resets and re-derives on the next repartition."""
+        self.comm = comm
+        for st in self.states.values():
+            st.boundaries = None
+
     # ------------------------------------------------------------------
     def state(self, name: str, t: torch.Tensor) -> TensorState:
         st = self.states.get(name)

@@ -148,6 +148,21 @@ class Trainer:
             next_sentence_label=b.nsp,
         )
 
+    def update_nworker(self, new_comm) -> None:
+        """Continue training in a shrunken world (reference
+        DLTrainer.update_nworker, VGG/dl_trainer.py:472-493: rebuilds the
+        DistributedSampler; here the per-rank synthetic shard is re-seeded
+        and the engine communicator swapped)."""
+        self.comm = new_comm
+        if hasattr(self.opt, "reducer"):
+            self.opt.reducer.set_comm(new_comm)
+        if hasattr(self.opt, "comm"):
+            self.opt.comm = new_comm
+        self.batches = SyntheticBatches(
+            self.model_name, self.batches.bs, self.device,
+            seq_len=self.batches.seq_len, rank=new_comm.rank,
+        )
+
     def step(self) -> float:
         """One optimizer step (with nsteps_update grad-accumulation substeps)."""
         self.opt.zero_grad()
