@@ -7,10 +7,39 @@ These are fresh implementations sized to the reference configs.
 from .vgg import VGG, vgg16
 from .lstm import DeepSpeech, deepspeech_an4
 from .bert import BertConfig, BertForPreTraining, bert_base, bert_large
+from .resnet import (
+    resnet20, resnet32, resnet44, resnet56, resnet110,
+    resnet18, resnet34, resnet50, resnet101, resnet152,
+)
+from .small_cnns import (
+    AlexNet, CaffeCifar, DenseNetCifar, ResNeXtCifar, PTBLSTM,
+)
 
 _REGISTRY = {
+    # CIFAR CNNs (reference VGG/models registry)
+    "vgg11": lambda **kw: VGG("vgg11", **kw),
+    "vgg13": lambda **kw: VGG("vgg13", **kw),
     "vgg16": vgg16,
-    "lstman4": deepspeech_an4,
+    "vgg19": lambda **kw: VGG("vgg19", **kw),
+    "resnet20": resnet20,
+    "resnet32": resnet32,
+    "resnet44": resnet44,
+    "resnet56": resnet56,
+    "resnet110": resnet110,
+    "alexnet": AlexNet,
+    "caffe_cifar": CaffeCifar,
+    "densenet": DenseNetCifar,
+    "resnext": ResNeXtCifar,
+    # ImageNet CNNs
+    "resnet18": resnet18,
+    "resnet34": resnet34,
+    "resnet50": resnet50,
+    "resnet101": resnet101,
+    "resnet152": resnet152,
+    # RNN workloads
+    "lstm": PTBLSTM,        # PTB word LM (reference models/lstm.py)
+    "lstman4": deepspeech_an4,  # DeepSpeech-style AN4 (reference models/lstman4.py)
+    # BERT
     "bert_base": bert_base,
     "bert_large": bert_large,
 }

@@ -74,4 +74,6 @@ class DeepSpeech(nn.Module):
 
 
 def deepspeech_an4(**kw) -> DeepSpeech:
-    return DeepSpeech(rnn_hidden_size=800, nb_layers=5, **kw)
+    kw.setdefault("rnn_hidden_size", 800)
+    kw.setdefault("nb_layers", 5)
+    return DeepSpeech(**kw)

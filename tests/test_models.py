@@ -1,0 +1,76 @@
+"""Model zoo: every registry entry builds and runs forward/backward on CPU."""
+import pytest
+import torch
+
+from oktopk_amd import models
+
+CIFAR_MODELS = ["vgg11", "vgg16", "resnet20", "resnet56", "alexnet",
+                "caffe_cifar", "densenet", "resnext"]
+
+
+@pytest.mark.parametrize("name", CIFAR_MODELS)
+def test_cifar_models_forward_backward(name):
+    torch.manual_seed(0)
+    m = models.create_net(name)
+    x = torch.randn(2, 3, 32, 32)
+    out = m(x)
+    assert out.shape == (2, 10)
+    loss = torch.nn.functional.cross_entropy(out, torch.tensor([1, 3]))
+    loss.backward()
+    assert all(p.grad is not None for p in m.parameters() if p.requires_grad)
+
+
+@pytest.mark.parametrize("name", ["resnet18", "resnet50"])
+def test_imagenet_models_forward(name):
+    m = models.create_net(name)
+    out = m(torch.randn(1, 3, 64, 64))
+    assert out.shape == (1, 1000)
+
+
+def test_ptb_lstm():
+    m = models.create_net("lstm", vocab_size=200, emb=32, hidden=32)
+    x = torch.randint(0, 200, (5, 3))
+    logits, hidden = m(x)
+    assert logits.shape == (5, 3, 200)
+    loss = torch.nn.functional.cross_entropy(
+        logits.view(-1, 200), torch.randint(0, 200, (15,))
+    )
+    loss.backward()
+
+
+def test_deepspeech_shapes():
+    m = models.create_net("lstman4", rnn_hidden_size=64)
+    x = torch.randn(2, 1, 161, 101)
+    out = m(x)
+    assert out.dim() == 3 and out.size(1) == 2 and out.size(2) == 29
+
+
+def test_deepspeech_param_count_reference_scale():
+    """Reference DeepSpeech an4: 5x800 bi-LSTM ~= 27M params
+    (LSTM/models/lstman4.py:7)."""
+    m = models.create_net("lstman4")
+    n = sum(p.numel() for p in m.parameters())
+    assert 20e6 < n < 50e6, n
+
+
+def test_bert_base_param_count():
+    """Reference BERT-base: 109.5M params."""
+    m = models.create_net("bert_base")
+    n = sum(p.numel() for p in m.parameters())
+    assert 105e6 < n < 115e6, n
+
+
+def test_bert_large_param_count():
+    m = models.create_net("bert_large")
+    n = sum(p.numel() for p in m.parameters())
+    assert 300e6 < n < 360e6, n
+
+
+def test_flops_counter():
+    from oktopk_amd.utils import get_model_complexity_info
+
+    m = models.create_net("vgg16")
+    flops, params = get_model_complexity_info(m, (3, 32, 32))
+    # VGG-16 on 32x32 is ~0.31 GMac; params ~15M
+    assert 2e8 < flops < 5e8, flops
+    assert 14e6 < params < 16e6, params
