@@ -11,14 +11,19 @@ import torch.nn as nn
 
 
 class BatchRNN(nn.Module):
-    """One bidirectional LSTM layer with BN on input and summed directions."""
+    """One LSTM layer with BN on input; bidirectional variants sum the two
+    directions (reference LSTM/models/lstm_models.py:84-116 BatchRNN; the
+    an4 recipe builds it UNIdirectional, LSTM/models/lstman4.py:8)."""
 
-    def __init__(self, input_size: int, hidden_size: int, batch_norm: bool = True):
+    def __init__(self, input_size: int, hidden_size: int, batch_norm: bool = True,
+                 bidirectional: bool = False):
         super().__init__()
         self.batch_norm = (
             nn.BatchNorm1d(input_size) if batch_norm else None
         )
-        self.rnn = nn.LSTM(input_size, hidden_size, bidirectional=True, bias=True)
+        self.bidirectional = bidirectional
+        self.rnn = nn.LSTM(input_size, hidden_size, bidirectional=bidirectional,
+                           bias=True)
         self.hidden_size = hidden_size
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:  # x: (T, N, F)
@@ -26,8 +31,9 @@ class BatchRNN(nn.Module):
             t, n = x.size(0), x.size(1)
             x = self.batch_norm(x.view(t * n, -1)).view(t, n, -1)
         x, _ = self.rnn(x)
-        # sum forward/backward directions (reference lstm_models.py BatchRNN)
-        x = x.view(x.size(0), x.size(1), 2, -1).sum(2)
+        if self.bidirectional:
+            # sum forward/backward directions (reference BatchRNN)
+            x = x.view(x.size(0), x.size(1), 2, -1).sum(2)
         return x
 
 
@@ -39,6 +45,7 @@ class DeepSpeech(nn.Module):
         num_classes: int = 29,
         sample_rate: int = 16000,
         window_size: float = 0.02,
+        bidirectional: bool = False,
     ):
         super().__init__()
         self.conv = nn.Sequential(
@@ -53,9 +60,11 @@ class DeepSpeech(nn.Module):
         freq = (freq + 2 * 20 - 41) // 2 + 1
         freq = (freq + 2 * 10 - 21) // 2 + 1
         rnn_in = freq * 32
-        rnns = [BatchRNN(rnn_in, rnn_hidden_size, batch_norm=False)]
+        rnns = [BatchRNN(rnn_in, rnn_hidden_size, batch_norm=False,
+                         bidirectional=bidirectional)]
         for _ in range(nb_layers - 1):
-            rnns.append(BatchRNN(rnn_hidden_size, rnn_hidden_size))
+            rnns.append(BatchRNN(rnn_hidden_size, rnn_hidden_size,
+                                 bidirectional=bidirectional))
         self.rnns = nn.Sequential(*rnns)
         self.fc = nn.Sequential(
             nn.BatchNorm1d(rnn_hidden_size),
