@@ -19,6 +19,25 @@ from .config import EngineConfig
 from .optimizer import DistributedOptimizer, FlatBertAdam
 
 
+_CIFAR = {"vgg11", "vgg13", "vgg16", "vgg19", "resnet20", "resnet32", "resnet44",
+          "resnet56", "resnet110", "alexnet", "caffe_cifar", "densenet", "resnext"}
+_IMAGENET = {"resnet18", "resnet34", "resnet50", "resnet101", "resnet152"}
+
+
+def model_family(name: str) -> str:
+    if name in _CIFAR:
+        return "cifar"
+    if name in _IMAGENET:
+        return "imagenet"
+    if name == "lstm":
+        return "ptb"
+    if name == "lstman4":
+        return "an4"
+    if name.startswith("bert"):
+        return "bert"
+    raise ValueError(name)
+
+
 class SyntheticBatches:
     """Deterministic per-rank synthetic batches with reference shapes."""
 
@@ -30,10 +49,18 @@ class SyntheticBatches:
         self.seq_len = seq_len
         self.vocab = vocab_size
         g = torch.Generator().manual_seed(1234 + rank)
-        if model_name == "vgg16":
+        self.family = model_family(model_name)
+        if self.family == "cifar":
             # CIFAR-10 shape, bs 16/rank in the reference (vgg16_oktopk.sh)
             self.x = torch.randn(batch_size, 3, 32, 32, generator=g)
             self.y = torch.randint(0, 10, (batch_size,), generator=g)
+        elif self.family == "imagenet":
+            self.x = torch.randn(batch_size, 3, 224, 224, generator=g)
+            self.y = torch.randint(0, 1000, (batch_size,), generator=g)
+        elif self.family == "ptb":
+            # PTB word-LM: (seq, batch) int tokens + shifted targets
+            self.tokens = torch.randint(0, 10000, (35, batch_size), generator=g)
+            self.targets = torch.randint(0, 10000, (35, batch_size), generator=g)
         elif model_name == "lstman4":
             # AN4 spectrograms: (N, 1, freq=161, time~200), bs 2/rank
             self.x = torch.randn(batch_size, 1, 161, 201, generator=g)
@@ -82,7 +109,7 @@ class Trainer:
         self.model_name = model_name
         self.cfg = cfg or EngineConfig.preset(
             "bert" if model_name.startswith("bert") else
-            ("lstm" if model_name == "lstman4" else "vgg")
+            ("lstm" if model_name.startswith("lstm") else "vgg")
         )
         # BERT runs in PURE bf16 with fp32 master weights in FlatBertAdam
         # (autocast's per-layer weight casts are ~1000 extra kernel launches
@@ -129,9 +156,14 @@ class Trainer:
     # ------------------------------------------------------------------
     def _forward_loss(self) -> torch.Tensor:
         b = self.batches
-        if self.model_name == "vgg16":
+        if b.family in ("cifar", "imagenet"):
             out = self.model(b.x)
             return torch.nn.functional.cross_entropy(out, b.y)
+        if b.family == "ptb":
+            logits, _ = self.model(b.tokens)
+            return torch.nn.functional.cross_entropy(
+                logits.view(-1, logits.size(-1)).float(), b.targets.view(-1)
+            )
         if self.model_name == "lstman4":
             logits = self.model(b.x)  # (T, N, C)
             logp = torch.nn.functional.log_softmax(logits, dim=-1)
