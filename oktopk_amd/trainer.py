@@ -152,6 +152,13 @@ class Trainer:
         )
         self.iteration = 0
         self.last_loss = 0.0
+        # hipGraph capture of fwd+bwd (the step is launch-bound at small
+        # per-GPU batches — measured in profiles/): grads accumulate into
+        # the optimizer's flat views, so one graph replays the whole
+        # fwd+bwd; the sparse engine (data-dependent sizes + RCCL) stays
+        # outside the graph.
+        self._graph = None
+        self._graph_loss = None
 
     # ------------------------------------------------------------------
     def _forward_loss(self) -> torch.Tensor:
@@ -195,8 +202,53 @@ class Trainer:
             seq_len=self.batches.seq_len, rank=new_comm.rank,
         )
 
+    def capture_graph(self) -> bool:
+        """Capture fwd+bwd into a hipGraph (call after a few eager warmup
+        steps).  Returns False (and stays eager) when the model is not
+        capturable (e.g. CTC loss) or no GPU is present."""
+        if self.device.type != "cuda" or self.nsteps_update != 1:
+            return False
+        if getattr(self.opt, "local", False):
+            return False
+        # hook-driven optimizers run comm inside backward - not capturable
+        from .optimizer import _DistributedOptimizer
+
+        if isinstance(self.opt, _DistributedOptimizer):
+            return False
+        try:
+            torch.cuda.synchronize()
+            self.opt.zero_grad()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(2):  # warm up allocator on the side stream
+                    self.opt.zero_grad()
+                    loss = self._forward_loss()
+                    loss.backward()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            self.opt.zero_grad()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                loss = self._forward_loss()
+                loss.backward()
+            self._graph = g
+            self._graph_loss = loss
+            return True
+        except Exception:
+            self._graph = None
+            self._graph_loss = None
+            return False
+
     def step(self) -> float:
         """One optimizer step (with nsteps_update grad-accumulation substeps)."""
+        if self._graph is not None and self.nsteps_update == 1:
+            self.opt.zero_grad()
+            self._graph.replay()
+            self.opt.step()
+            self.iteration += 1
+            self.last_loss = float(self._graph_loss.detach().float().item())
+            return self.last_loss
         self.opt.zero_grad()
         for sub in range(self.nsteps_update):
             if hasattr(self.opt, "local"):
