@@ -237,20 +237,27 @@ class AllReducer:
         k = self._k(n)
         it = st.counter
 
-        # --- 1. error-feedback restore + local threshold maintenance ----
+        # --- 1. error-feedback restore + local selection -----------------
+        # threshold maintenance and compaction are FUSED: the count pass that
+        # chooses tau is the same pass that compaction's offsets need, so the
+        # reference's count-per-candidate loop + separate compact-count pass
+        # collapse into one tensor read (ops.compact_adaptive).
         s0 = time.perf_counter()
         ops.ef_restore_snapshot_(t, st.residual)
         if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
             st.tau_local = ops.kth_abs_value(t, k)
+            idx, val = ops.compact_gt(t, st.tau_local)
         else:
-            st.tau_local = self._adaptive_bump(t, st.tau_local, k)
+            taus = [st.tau_local * ok.bump_scale ** i
+                    for i in range(ok.bump_max_loops + 1)]
+            idx, val, chosen, _cnt = ops.compact_adaptive(t, taus, 4 * k // 3)
+            st.tau_local = taus[chosen]
         tau = st.tau_local
 
-        # --- 2. balanced region repartition ------------------------------
+        # --- 2. balanced region repartition (reuses the selection) -------
         if st.boundaries is None:
             st.boundaries = self._uniform_boundaries(n)
         if it % ok.region_repartition_interval == 0 and P > 1:
-            idx, _ = ops.compact_gt(t, tau)
             m = idx.numel()
             if m >= P:
                 step = m // P
@@ -268,8 +275,7 @@ class AllReducer:
         bounds = st.boundaries
         lo, hi = int(bounds[rank]), int(bounds[rank + 1])
 
-        # --- 3. round 1: select + sparse reduce-scatter to region owners -
-        idx, val = ops.compact_gt(t, tau)  # ascending idx, one pass
+        # --- 3. round 1: sparse reduce-scatter to region owners ----------
         sel = idx.numel()
         # local feedback controller (VGG/allreducer.py:696-699)
         if sel < ok.local_lo_num * k // ok.local_lo_den:
@@ -417,9 +423,12 @@ class AllReducer:
         ops.ef_restore_snapshot_(t, st.residual)
         if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
             st.tau_local = ops.kth_abs_value(t, k)
+            idx, val = ops.compact_gt(t, st.tau_local)
         else:
-            st.tau_local = self._adaptive_bump(t, st.tau_local, k)
-        idx, val = ops.compact_gt(t, st.tau_local)
+            taus = [st.tau_local * ok.bump_scale ** i
+                    for i in range(ok.bump_max_loops + 1)]
+            idx, val, chosen, _cnt = ops.compact_adaptive(t, taus, 4 * k // 3)
+            st.tau_local = taus[chosen]
         sel = idx.numel()
         if sel < ok.local_lo_num * k // ok.local_lo_den:
             st.tau_local /= ok.scale_local
@@ -518,16 +527,20 @@ class AllReducer:
                 else:
                     break
             st.tau_local = tau
+            idx, val = ops.compact_gt(t, tau)
         else:
             if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
                 st.tau_local = ops.kth_abs_value(t, k)
+                idx, val = ops.compact_gt(t, st.tau_local)
             else:
-                st.tau_local = self._adaptive_bump(t, st.tau_local, k)
-        tau = st.tau_local
+                taus = [st.tau_local * ok.bump_scale ** i
+                        for i in range(ok.bump_max_loops + 1)]
+                idx, val, chosen, _cnt = ops.compact_adaptive(t, taus, 4 * k // 3)
+                st.tau_local = taus[chosen]
+            tau = st.tau_local
 
         bounds = self._uniform_boundaries(n)
         lo, hi = int(bounds[rank]), int(bounds[rank + 1])
-        idx, val = ops.compact_gt(t, tau)
         sel = idx.numel()
         if not gaussian:
             if sel < ok.local_lo_num * k // ok.local_lo_den:

@@ -69,6 +69,12 @@ def count_multi_gt(t: torch.Tensor, taus) -> list:
     return [int(x) for x in _backend(t).count_multi_gt(t, [float(x) for x in taus])]
 
 
+def compact_adaptive(t: torch.Tensor, taus, hi_limit: int):
+    out = _backend(t).compact_adaptive(t, [float(x) for x in taus], int(hi_limit))
+    idx, val, chosen, count = out
+    return idx, val, int(chosen), int(count)
+
+
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
     return _backend(dest).scatter_add_(dest, idx, val)
 

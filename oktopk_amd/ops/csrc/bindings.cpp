@@ -14,7 +14,8 @@ extern "C" {
 void launch_count_gt(const float*, int64_t, float, unsigned long long*, hipStream_t);
 void launch_count_multi_gt(const float*, int64_t, const float*, int, unsigned long long*,
                            hipStream_t);
-void launch_compact_count(const float*, int64_t, float, int64_t, int, int*, hipStream_t);
+void launch_compact_count_multi(const float*, int64_t, const float*, int, int64_t,
+                                int, int*, hipStream_t);
 void launch_compact_write(const float*, int64_t, float, int64_t, int, const int*,
                           int32_t*, float*, hipStream_t);
 void launch_hist(const float*, int64_t, uint32_t, uint32_t, int, int, unsigned int*,
@@ -79,34 +80,96 @@ static std::vector<int64_t> count_multi_gt(torch::Tensor t, std::vector<double> 
     return res;
 }
 
-static std::vector<torch::Tensor> compact_gt(torch::Tensor t, double tau) {
-    check_f32_1d(t, "t");
-    const at::cuda::CUDAGuard guard(t.device());
-    int64_t n = t.numel();
-    // chunk per block: multiple of BLOCK*COMPACT_VEC (1024) covering n
-    const int64_t unit = 1024;
+struct CompactGeom {
+    int64_t chunk;
+    int nblocks;
+};
+
+static CompactGeom compact_geom(int64_t n) {
+    // chunk per block: multiple of BLOCK*COMPACT_VEC (2048) covering n,
+    // keeping <= 2048 blocks (G11 grid sizing)
+    const int64_t unit = 2048;
     int64_t nchunks = (n + unit - 1) / unit;
     if (nchunks < 1) nchunks = 1;
     if (nchunks > 2048) nchunks = 2048;
     int64_t chunk = ((n + nchunks - 1) / nchunks + unit - 1) / unit * unit;
     int nblocks = (int)((n + chunk - 1) / chunk);
     if (nblocks < 1) nblocks = 1;
+    return {chunk, nblocks};
+}
 
-    auto counts = torch::empty({nblocks}, t.options().dtype(torch::kInt32));
-    launch_compact_count(t.data_ptr<float>(), n, (float)tau, chunk, nblocks,
-                         counts.data_ptr<int>(), cur_stream());
-    auto csum = counts.cumsum(0, torch::kInt32);
-    auto offsets = csum - counts;  // exclusive scan
-    int64_t total = csum[nblocks - 1].cpu().item<int>();
+// shared tail: given per-candidate block counts (ntau x nblocks, already on
+// CPU) and the chosen candidate row, launch the write pass.
+static std::vector<torch::Tensor> compact_finish(
+    torch::Tensor t, double tau, const CompactGeom& g,
+    const int* row_counts /* nblocks ints for the chosen tau */) {
+    int64_t total = 0;
+    std::vector<int> offs(g.nblocks);
+    for (int b = 0; b < g.nblocks; ++b) {
+        offs[b] = (int)total;
+        total += row_counts[b];
+    }
     auto idx = torch::empty({total}, t.options().dtype(torch::kInt32));
     auto val = torch::empty({total}, t.options());
     if (total > 0) {
-        launch_compact_write(t.data_ptr<float>(), n, (float)tau, chunk, nblocks,
-                             offsets.contiguous().data_ptr<int>(),
+        auto offsets = torch::from_blob(offs.data(), {g.nblocks}, torch::kInt32)
+                           .to(t.device(), /*non_blocking=*/false);
+        launch_compact_write(t.data_ptr<float>(), t.numel(), (float)tau, g.chunk,
+                             g.nblocks, offsets.data_ptr<int>(),
                              idx.data_ptr<int32_t>(), val.data_ptr<float>(),
                              cur_stream());
     }
     return {idx, val};
+}
+
+static std::vector<torch::Tensor> compact_gt(torch::Tensor t, double tau) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    int64_t n = t.numel();
+    auto g = compact_geom(n);
+    auto counts = torch::empty({g.nblocks}, t.options().dtype(torch::kInt32));
+    float tf = (float)tau;
+    launch_compact_count_multi(t.data_ptr<float>(), n, &tf, 1, g.chunk, g.nblocks,
+                               counts.data_ptr<int>(), cur_stream());
+    auto h = counts.cpu();
+    return compact_finish(t, tau, g, h.data_ptr<int>());
+}
+
+// Fused adaptive-threshold compaction: ONE pass counts all candidate taus
+// per block; host applies the bump rule (smallest i with count <= hi_limit,
+// reference add2residual VGG/compression.py:384-404); write pass extracts at
+// the chosen tau.  Returns {idx, val, chosen_index, chosen_count}.
+static std::vector<torch::Tensor> compact_adaptive(
+    torch::Tensor t, std::vector<double> taus, int64_t hi_limit) {
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    TORCH_CHECK(taus.size() >= 1 && taus.size() <= 8, "1..8 thresholds");
+    int ntau = (int)taus.size();
+    int64_t n = t.numel();
+    auto g = compact_geom(n);
+    float tf[8];
+    for (int j = 0; j < ntau; ++j) tf[j] = (float)taus[j];
+    auto counts = torch::empty({(int64_t)ntau * g.nblocks},
+                               t.options().dtype(torch::kInt32));
+    launch_compact_count_multi(t.data_ptr<float>(), n, tf, ntau, g.chunk, g.nblocks,
+                               counts.data_ptr<int>(), cur_stream());
+    auto h = counts.cpu();
+    const int* hp = h.data_ptr<int>();
+    int chosen = ntau - 1;
+    int64_t chosen_total = 0;
+    for (int c = 0; c < ntau; ++c) {
+        int64_t tot = 0;
+        for (int b = 0; b < g.nblocks; ++b) tot += hp[(int64_t)c * g.nblocks + b];
+        if (c == ntau - 1 || tot <= hi_limit) {
+            chosen = c;
+            chosen_total = tot;
+            break;
+        }
+    }
+    auto out = compact_finish(t, taus[chosen], g, hp + (int64_t)chosen * g.nblocks);
+    out.push_back(torch::tensor((int64_t)chosen));
+    out.push_back(torch::tensor(chosen_total));
+    return out;
 }
 
 static double kth_abs_value(torch::Tensor t, int64_t k) {
@@ -238,6 +301,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("count_gt", &count_gt, "count |t| > tau");
     m.def("count_multi_gt", &count_multi_gt, "counts for up to 8 thresholds, one pass");
     m.def("compact_gt", &compact_gt, "ascending (idx,val) where |t| > tau");
+    m.def("compact_adaptive", &compact_adaptive,
+          "one-pass adaptive-threshold compaction: counts all candidate taus, "
+          "applies the bump rule, extracts at the chosen tau");
     m.def("kth_abs_value", &kth_abs_value, "exact k-th largest |t| via radix select");
     m.def("scatter_add_", &scatter_add_, "dest[idx] += val");
     m.def("zero_at_", &zero_at_, "t[idx] = 0");

@@ -135,25 +135,56 @@ extern "C" void launch_count_multi_gt(const float* t, int64_t n, const float* ta
 // Pass B: stable intra-block scan (wave ballots + LDS) writing idx+val at
 //         exclusive-scanned block offsets -> output ascending by index.
 // ---------------------------------------------------------------------------
-#define COMPACT_VEC 4  // elements per thread per iteration
+#define COMPACT_VEC 8  // elements per thread per iteration (2x float4)
 
-__global__ void compact_count_kernel(const float* __restrict__ t, int64_t n,
-                                     uint32_t tau_bits, int64_t chunk,
-                                     int* __restrict__ block_counts) {
+// Pass A: per-block counts for up to 8 candidate thresholds in ONE pass.
+// Feeding the adaptive-bump choice (add2residual, VGG/compression.py:384-404)
+// from the same pass that compaction needs anyway: one read of the tensor
+// replaces the reference's count-per-candidate loop AND the separate
+// compact-count pass.  block_counts layout: [cand][block].
+__global__ void compact_count_multi_kernel(const float* __restrict__ t, int64_t n,
+                                           TauSet taus, int64_t chunk,
+                                           int* __restrict__ block_counts,
+                                           int nblocks) {
     int64_t start = (int64_t)blockIdx.x * chunk;
     int64_t end = (start + chunk < n) ? start + chunk : n;
-    int cnt = 0;
-    for (int64_t i = start + threadIdx.x; i < end; i += BLOCK)
-        cnt += sel_gt(abs_bits(t[i]), tau_bits);
-    for (int off = 32; off > 0; off >>= 1) cnt += __shfl_down(cnt, off, 64);
+    int cnt[8];
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) cnt[c] = 0;
+    // chunk starts are 1024-element aligned -> float4-aligned; only the very
+    // last block can have a ragged tail.
+    int64_t vend = start + ((end - start) & ~3LL);
+    const float4* t4 = reinterpret_cast<const float4*>(t + start);
+    int64_t n4 = (vend - start) >> 2;
+    for (int64_t i = threadIdx.x; i < n4; i += BLOCK) {
+        float4 x = t4[i];
+        uint32_t a0 = abs_bits(x.x), a1 = abs_bits(x.y), a2 = abs_bits(x.z),
+                 a3 = abs_bits(x.w);
+        #pragma unroll
+        for (int c = 0; c < 8; ++c)
+            if (c < taus.n)
+                cnt[c] += sel_gt(a0, taus.tb[c]) + sel_gt(a1, taus.tb[c]) +
+                          sel_gt(a2, taus.tb[c]) + sel_gt(a3, taus.tb[c]);
+    }
+    for (int64_t i = vend + threadIdx.x; i < end; i += BLOCK) {
+        uint32_t a = abs_bits(t[i]);
+        #pragma unroll
+        for (int c = 0; c < 8; ++c)
+            if (c < taus.n) cnt[c] += sel_gt(a, taus.tb[c]);
+    }
     __shared__ int ws[WAVES_PER_BLOCK];
     int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
-    if (lane == 0) ws[wave] = cnt;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-        int s = 0;
-        for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
-        block_counts[blockIdx.x] = s;
+    for (int c = 0; c < taus.n; ++c) {
+        int v = cnt[c];
+        for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+        if (lane == 0) ws[wave] = v;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int s = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
+            block_counts[(int64_t)c * nblocks + blockIdx.x] = s;
+        }
+        __syncthreads();
     }
 }
 
@@ -166,26 +197,33 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
     int64_t end = (start + chunk < n) ? start + chunk : n;
     int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
     uint64_t lt_mask = ((uint64_t)1 << lane) - 1;
-    __shared__ int wsum[WAVES_PER_BLOCK];
-    __shared__ int running;
-    if (threadIdx.x == 0) running = block_offsets[blockIdx.x];
-    __syncthreads();
+    __shared__ int wsum[2][WAVES_PER_BLOCK];  // double-buffered: 1 barrier/iter
+    int run = block_offsets[blockIdx.x];  // wave-uniform broadcast load
+    int parity = 0;
 
     // iteration covers BLOCK*COMPACT_VEC consecutive elements; thread tid owns
-    // elements base+tid*4 .. +3, so (wave, lane, j) order == index order.
+    // elements base+tid*8 .. +7, so (wave, lane, j) order == index order.
     for (int64_t base = start; base < end; base += (int64_t)BLOCK * COMPACT_VEC) {
         int64_t my = base + (int64_t)threadIdx.x * COMPACT_VEC;
         bool p[COMPACT_VEC];
         float v[COMPACT_VEC];
-        int own = 0;
-        #pragma unroll
-        for (int j = 0; j < COMPACT_VEC; ++j) {
-            int64_t i = my + j;
-            bool ok = i < end;
-            float x = ok ? t[i] : 0.f;
-            p[j] = ok && sel_gt(abs_bits(x), tau_bits);
-            v[j] = x;
-            own += p[j];
+        if (my + COMPACT_VEC <= end) {
+            const float4* src = reinterpret_cast<const float4*>(t + my);
+            float4 x0 = src[0], x1 = src[1];
+            v[0] = x0.x; v[1] = x0.y; v[2] = x0.z; v[3] = x0.w;
+            v[4] = x1.x; v[5] = x1.y; v[6] = x1.z; v[7] = x1.w;
+            #pragma unroll
+            for (int j = 0; j < COMPACT_VEC; ++j)
+                p[j] = sel_gt(abs_bits(v[j]), tau_bits);
+        } else {
+            #pragma unroll
+            for (int j = 0; j < COMPACT_VEC; ++j) {
+                int64_t i = my + j;
+                bool ok = i < end;
+                float x = ok ? t[i] : 0.f;
+                p[j] = ok && sel_gt(abs_bits(x), tau_bits);
+                v[j] = x;
+            }
         }
         int lane_prefix = 0, wave_total = 0;
         #pragma unroll
@@ -194,14 +232,15 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
             lane_prefix += __popcll(b & lt_mask);
             wave_total += __popcll(b);
         }
-        if (lane == 0) wsum[wave] = wave_total;
+        if (lane == 0) wsum[parity][wave] = wave_total;
         __syncthreads();
         int wave_prefix = 0, iter_total = 0;
         for (int w = 0; w < WAVES_PER_BLOCK; ++w) {
-            if (w < wave) wave_prefix += wsum[w];
-            iter_total += wsum[w];
+            int s = wsum[parity][w];
+            if (w < wave) wave_prefix += s;
+            iter_total += s;
         }
-        int pos = running + wave_prefix + lane_prefix;
+        int pos = run + wave_prefix + lane_prefix;
         #pragma unroll
         for (int j = 0; j < COMPACT_VEC; ++j) {
             if (p[j]) {
@@ -210,18 +249,20 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
                 ++pos;
             }
         }
-        __syncthreads();
-        if (threadIdx.x == 0) running += iter_total;
-        __syncthreads();
+        run += iter_total;  // every thread derives the same running offset
+        parity ^= 1;
     }
 }
 
-extern "C" void launch_compact_count(const float* t, int64_t n, float tau,
-                                     int64_t chunk, int nblocks,
-                                     int* block_counts, hipStream_t stream) {
-    uint32_t tb = tau_to_bits(tau);
-    hipLaunchKernelGGL(compact_count_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
-                       t, n, tb, chunk, block_counts);
+extern "C" void launch_compact_count_multi(const float* t, int64_t n,
+                                           const float* taus, int ntau,
+                                           int64_t chunk, int nblocks,
+                                           int* block_counts, hipStream_t stream) {
+    TauSet ts;
+    ts.n = ntau;
+    for (int j = 0; j < 8; ++j) ts.tb[j] = j < ntau ? tau_to_bits(taus[j]) : 0;
+    hipLaunchKernelGGL(compact_count_multi_kernel, dim3(nblocks), dim3(BLOCK), 0,
+                       stream, t, n, ts, chunk, block_counts, nblocks);
 }
 
 extern "C" void launch_compact_write(const float* t, int64_t n, float tau,

@@ -20,6 +20,7 @@ __all__ = [
     "compact_gt",
     "count_gt",
     "count_multi_gt",
+    "compact_adaptive",
     "scatter_add_",
     "zero_at_",
     "fill_sparse_scaled_",
@@ -56,6 +57,21 @@ def count_gt(t: torch.Tensor, tau: float) -> int:
 def count_multi_gt(t: torch.Tensor, taus) -> list:
     a = t.reshape(-1).abs()
     return [int((a > float(x)).sum().item()) for x in taus]
+
+
+def compact_adaptive(t: torch.Tensor, taus, hi_limit: int):
+    """Adaptive-threshold compaction (fused in HIP): choose the first tau in
+    `taus` whose selected count <= hi_limit (else the last), extract there.
+    Returns (idx, val, chosen_index, chosen_count).  Mirrors the bump loop of
+    add2residual, VGG/compression.py:384-404."""
+    counts = count_multi_gt(t, taus)
+    chosen = len(taus) - 1
+    for c, tot in enumerate(counts):
+        if c == len(taus) - 1 or tot <= hi_limit:
+            chosen = c
+            break
+    idx, val = compact_gt(t, taus[chosen])
+    return idx, val, chosen, counts[chosen]
 
 
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
