@@ -98,21 +98,22 @@ static CompactGeom compact_geom(int64_t n) {
     return {chunk, nblocks};
 }
 
-// shared tail: given per-candidate block counts (ntau x nblocks, already on
-// CPU) and the chosen candidate row, launch the write pass.
+// shared tail: given per-candidate per-wave counts (ntau x nblocks*4 on CPU)
+// and the chosen candidate row, launch the write pass.
 static std::vector<torch::Tensor> compact_finish(
     torch::Tensor t, double tau, const CompactGeom& g,
-    const int* row_counts /* nblocks ints for the chosen tau */) {
+    const int* row_counts /* nblocks*4 ints for the chosen tau */) {
+    const int nw = g.nblocks * 4;  // WAVES_PER_BLOCK
     int64_t total = 0;
-    std::vector<int> offs(g.nblocks);
-    for (int b = 0; b < g.nblocks; ++b) {
+    std::vector<int> offs(nw);
+    for (int b = 0; b < nw; ++b) {
         offs[b] = (int)total;
         total += row_counts[b];
     }
     auto idx = torch::empty({total}, t.options().dtype(torch::kInt32));
     auto val = torch::empty({total}, t.options());
     if (total > 0) {
-        auto offsets = torch::from_blob(offs.data(), {g.nblocks}, torch::kInt32)
+        auto offsets = torch::from_blob(offs.data(), {nw}, torch::kInt32)
                            .to(t.device(), /*non_blocking=*/false);
         launch_compact_write(t.data_ptr<float>(), t.numel(), (float)tau, g.chunk,
                              g.nblocks, offsets.data_ptr<int>(),
@@ -127,7 +128,7 @@ static std::vector<torch::Tensor> compact_gt(torch::Tensor t, double tau) {
     const at::cuda::CUDAGuard guard(t.device());
     int64_t n = t.numel();
     auto g = compact_geom(n);
-    auto counts = torch::empty({g.nblocks}, t.options().dtype(torch::kInt32));
+    auto counts = torch::empty({g.nblocks * 4}, t.options().dtype(torch::kInt32));
     float tf = (float)tau;
     launch_compact_count_multi(t.data_ptr<float>(), n, &tf, 1, g.chunk, g.nblocks,
                                counts.data_ptr<int>(), cur_stream());
@@ -149,7 +150,8 @@ static std::vector<torch::Tensor> compact_adaptive(
     auto g = compact_geom(n);
     float tf[8];
     for (int j = 0; j < ntau; ++j) tf[j] = (float)taus[j];
-    auto counts = torch::empty({(int64_t)ntau * g.nblocks},
+    const int nw = g.nblocks * 4;
+    auto counts = torch::empty({(int64_t)ntau * nw},
                                t.options().dtype(torch::kInt32));
     launch_compact_count_multi(t.data_ptr<float>(), n, tf, ntau, g.chunk, g.nblocks,
                                counts.data_ptr<int>(), cur_stream());
@@ -159,14 +161,14 @@ static std::vector<torch::Tensor> compact_adaptive(
     int64_t chosen_total = 0;
     for (int c = 0; c < ntau; ++c) {
         int64_t tot = 0;
-        for (int b = 0; b < g.nblocks; ++b) tot += hp[(int64_t)c * g.nblocks + b];
+        for (int b = 0; b < nw; ++b) tot += hp[(int64_t)c * nw + b];
         if (c == ntau - 1 || tot <= hi_limit) {
             chosen = c;
             chosen_total = tot;
             break;
         }
     }
-    auto out = compact_finish(t, taus[chosen], g, hp + (int64_t)chosen * g.nblocks);
+    auto out = compact_finish(t, taus[chosen], g, hp + (int64_t)chosen * nw);
     out.push_back(torch::tensor((int64_t)chosen));
     out.push_back(torch::tensor(chosen_total));
     return out;

@@ -142,21 +142,29 @@ extern "C" void launch_count_multi_gt(const float* t, int64_t n, const float* ta
 // from the same pass that compaction needs anyway: one read of the tensor
 // replaces the reference's count-per-candidate loop AND the separate
 // compact-count pass.  block_counts layout: [cand][block].
+// Both passes are WAVE-autonomous: each wave owns a contiguous subchunk
+// (chunk / WAVES_PER_BLOCK elements), so pass B needs no LDS and no
+// __syncthreads — a per-iteration block barrier was measured at 0.8 TB/s
+// (each iteration's loads could not overlap the previous barrier); the
+// barrier-free wave loop software-pipelines to the streaming rate.
+// Count layout: [cand][block][wave].
 __global__ void compact_count_multi_kernel(const float* __restrict__ t, int64_t n,
                                            TauSet taus, int64_t chunk,
-                                           int* __restrict__ block_counts,
+                                           int* __restrict__ wave_counts,
                                            int nblocks) {
-    int64_t start = (int64_t)blockIdx.x * chunk;
-    int64_t end = (start + chunk < n) ? start + chunk : n;
+    int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int64_t sub = chunk / WAVES_PER_BLOCK;  // multiple of 512
+    int64_t start = (int64_t)blockIdx.x * chunk + (int64_t)wave * sub;
+    int64_t end = start + sub;
+    if (start > n) start = n;
+    if (end > n) end = n;
     int cnt[8];
     #pragma unroll
     for (int c = 0; c < 8; ++c) cnt[c] = 0;
-    // chunk starts are 1024-element aligned -> float4-aligned; only the very
-    // last block can have a ragged tail.
     int64_t vend = start + ((end - start) & ~3LL);
     const float4* t4 = reinterpret_cast<const float4*>(t + start);
     int64_t n4 = (vend - start) >> 2;
-    for (int64_t i = threadIdx.x; i < n4; i += BLOCK) {
+    for (int64_t i = lane; i < n4; i += 64) {
         float4 x = t4[i];
         uint32_t a0 = abs_bits(x.x), a1 = abs_bits(x.y), a2 = abs_bits(x.z),
                  a3 = abs_bits(x.w);
@@ -166,45 +174,38 @@ __global__ void compact_count_multi_kernel(const float* __restrict__ t, int64_t 
                 cnt[c] += sel_gt(a0, taus.tb[c]) + sel_gt(a1, taus.tb[c]) +
                           sel_gt(a2, taus.tb[c]) + sel_gt(a3, taus.tb[c]);
     }
-    for (int64_t i = vend + threadIdx.x; i < end; i += BLOCK) {
+    for (int64_t i = vend + lane; i < end; i += 64) {
         uint32_t a = abs_bits(t[i]);
         #pragma unroll
         for (int c = 0; c < 8; ++c)
             if (c < taus.n) cnt[c] += sel_gt(a, taus.tb[c]);
     }
-    __shared__ int ws[WAVES_PER_BLOCK];
-    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
     for (int c = 0; c < taus.n; ++c) {
         int v = cnt[c];
         for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
-        if (lane == 0) ws[wave] = v;
-        __syncthreads();
-        if (threadIdx.x == 0) {
-            int s = 0;
-            for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
-            block_counts[(int64_t)c * nblocks + blockIdx.x] = s;
-        }
-        __syncthreads();
+        if (lane == 0)
+            wave_counts[((int64_t)c * nblocks + blockIdx.x) * WAVES_PER_BLOCK + wave] = v;
     }
 }
 
 __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
                                      uint32_t tau_bits, int64_t chunk,
-                                     const int* __restrict__ block_offsets,
+                                     const int* __restrict__ wave_offsets,
                                      int32_t* __restrict__ out_idx,
                                      float* __restrict__ out_val) {
-    int64_t start = (int64_t)blockIdx.x * chunk;
-    int64_t end = (start + chunk < n) ? start + chunk : n;
-    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int64_t sub = chunk / WAVES_PER_BLOCK;
+    int64_t start = (int64_t)blockIdx.x * chunk + (int64_t)wave * sub;
+    int64_t end = start + sub;
+    if (start > n) start = n;
+    if (end > n) end = n;
     uint64_t lt_mask = ((uint64_t)1 << lane) - 1;
-    __shared__ int wsum[2][WAVES_PER_BLOCK];  // double-buffered: 1 barrier/iter
-    int run = block_offsets[blockIdx.x];  // wave-uniform broadcast load
-    int parity = 0;
+    int run = wave_offsets[blockIdx.x * WAVES_PER_BLOCK + wave];
 
-    // iteration covers BLOCK*COMPACT_VEC consecutive elements; thread tid owns
-    // elements base+tid*8 .. +7, so (wave, lane, j) order == index order.
-    for (int64_t base = start; base < end; base += (int64_t)BLOCK * COMPACT_VEC) {
-        int64_t my = base + (int64_t)threadIdx.x * COMPACT_VEC;
+    // iteration covers 64*COMPACT_VEC consecutive elements per wave; lane l
+    // owns elements base+l*8 .. +7, so (lane, j) order == index order.
+    for (int64_t base = start; base < end; base += 64 * COMPACT_VEC) {
+        int64_t my = base + (int64_t)lane * COMPACT_VEC;
         bool p[COMPACT_VEC];
         float v[COMPACT_VEC];
         if (my + COMPACT_VEC <= end) {
@@ -232,15 +233,7 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
             lane_prefix += __popcll(b & lt_mask);
             wave_total += __popcll(b);
         }
-        if (lane == 0) wsum[parity][wave] = wave_total;
-        __syncthreads();
-        int wave_prefix = 0, iter_total = 0;
-        for (int w = 0; w < WAVES_PER_BLOCK; ++w) {
-            int s = wsum[parity][w];
-            if (w < wave) wave_prefix += s;
-            iter_total += s;
-        }
-        int pos = run + wave_prefix + lane_prefix;
+        int pos = run + lane_prefix;
         #pragma unroll
         for (int j = 0; j < COMPACT_VEC; ++j) {
             if (p[j]) {
@@ -249,8 +242,7 @@ __global__ void compact_write_kernel(const float* __restrict__ t, int64_t n,
                 ++pos;
             }
         }
-        run += iter_total;  // every thread derives the same running offset
-        parity ^= 1;
+        run += wave_total;
     }
 }
 
