@@ -34,8 +34,10 @@ def main():
     args = ap.parse_args()
     torch.cuda.set_device(0)
     g = torch.Generator().manual_seed(0)
-    t = torch.randn(N, generator=g).cuda()
-    r = torch.randn(N, generator=g).cuda()
+    t0 = torch.randn(N, generator=g).cuda()
+    r0 = torch.randn(N, generator=g).cuda()
+    t = t0.clone()
+    r = r0.clone()
     dest = torch.zeros(N, device="cuda")
     m = torch.zeros(N, device="cuda")
     v = torch.zeros(N, device="cuda")
@@ -47,6 +49,12 @@ def main():
     rows = []
 
     def bench(name, fn, nbytes):
+        # reset inputs: in-place ops (ef_restore) must not poison later rows
+        # (an earlier version measured compact at 100% density because
+        # ef_restore had blown t up to inf — §5.4 rule 25 of the HIP guide:
+        # check the DATA FILL your bench actually runs on)
+        t.copy_(t0)
+        r.copy_(r0)
         ms = timeit(fn, args.iters) * 1000
         rows.append((name, ms, nbytes / GB / (ms / 1000)))
 
