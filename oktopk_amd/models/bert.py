@@ -92,8 +92,18 @@ class BertLayer(nn.Module):
 
     def forward(self, x, attn_mask=None):
         x = self.ln1(x + self.drop(self.attn(x, attn_mask)))
-        x = self.ln2(x + self.drop(self.fc2(F.gelu(self.fc1(x)))))
+        h = self._mlp_act(x)
+        x = self.ln2(x + self.drop(self.fc2(h)))
         return x
+
+    def _mlp_act(self, x):
+        # hand-written MFMA fused GEMM+bias+GELU on GPU/bf16 (reference
+        # LinearActivation fused-gelu, modeling.py:75); torch path otherwise
+        from ..ops.fused_linear import fused_available, fused_linear_gelu
+
+        if fused_available(x, self.fc1.weight):
+            return fused_linear_gelu(x, self.fc1.weight, self.fc1.bias)
+        return F.gelu(self.fc1(x))
 
 
 class BertModel(nn.Module):

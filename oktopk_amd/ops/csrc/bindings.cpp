@@ -32,6 +32,8 @@ void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
 void launch_adam(float*, const float*, float*, float*, int64_t, float, float, float,
                  float, float, hipStream_t);
 void launch_sumsq(const float*, int64_t, double*, hipStream_t);
+void launch_linear_gelu(const void*, const void*, const float*, void*, void*, int,
+                        int, int, int, hipStream_t);
 }
 
 namespace {
@@ -298,6 +300,37 @@ static double l2norm(torch::Tensor t) {
     return std::sqrt(out.cpu().item<double>());
 }
 
+static std::vector<torch::Tensor> linear_gelu(torch::Tensor x, torch::Tensor w,
+                                              torch::Tensor bias, bool apply_gelu,
+                                              bool want_pre) {
+    TORCH_CHECK(x.is_cuda() && w.is_cuda(), "GPU tensors required");
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && w.scalar_type() == torch::kBFloat16,
+                "bf16 inputs required");
+    TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+                "x [M,K], w [N,K]");
+    TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+    TORCH_CHECK(x.size(1) % 32 == 0, "K must be a multiple of 32");
+    const at::cuda::CUDAGuard guard(x.device());
+    int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+    auto y = torch::empty({M, N}, x.options());
+    torch::Tensor z;
+    const float* bptr = nullptr;
+    torch::Tensor bias_f;
+    if (bias.defined() && bias.numel()) {
+        bias_f = bias.to(torch::kFloat32).contiguous();
+        bptr = bias_f.data_ptr<float>();
+    }
+    void* zptr = nullptr;
+    if (want_pre) {
+        z = torch::empty({M, N}, x.options());
+        zptr = z.data_ptr();
+    }
+    launch_linear_gelu(x.data_ptr(), w.data_ptr(), bptr, y.data_ptr(), zptr,
+                       (int)M, (int)N, (int)K, apply_gelu ? 1 : 0, cur_stream());
+    if (want_pre) return {y, z};
+    return {y};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "oktopk_amd CDNA4 HIP kernels (gfx950)";
     m.def("count_gt", &count_gt, "count |t| > tau");
@@ -315,4 +348,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
     m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
+    m.def("linear_gelu", &linear_gelu,
+          "fused y=gelu(x@W^T+b) bf16 MFMA kernel (optionally returns pre-act)");
 }

@@ -19,6 +19,7 @@ setup(
             sources=[
                 "oktopk_amd/ops/csrc/bindings.cpp",
                 "oktopk_amd/ops/csrc/kernels.hip",
+                "oktopk_amd/ops/csrc/linear_gelu.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
