@@ -20,7 +20,10 @@ _INV_SQRT_2PI = 0.3989422804014327
 
 
 def fused_available(x: torch.Tensor, weight: torch.Tensor) -> bool:
-    if os.environ.get("OKTOPK_NO_FUSED_MLP", "0") == "1":
+    # opt-in: at BERT-base shapes (M=1024, 192 blocks) the correctness-tier
+    # kernel underfills 256 CUs and loses ~1ms/step to hipBLASLt+gelu
+    # (measured); enable for A/B with OKTOPK_FUSED_MLP=1
+    if os.environ.get("OKTOPK_FUSED_MLP", "0") != "1":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16):
         return False

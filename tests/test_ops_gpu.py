@@ -230,12 +230,12 @@ def test_bert_layer_uses_fused_mlp():
                      hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
     lyr = BertLayer(cfg).bfloat16().cuda()
     x = torch.randn(2, 16, 128).bfloat16().cuda()
-    out_fused = lyr(x)
-    os.environ["OKTOPK_NO_FUSED_MLP"] = "1"
+    os.environ["OKTOPK_FUSED_MLP"] = "1"
     try:
-        out_ref = lyr(x)
+        out_fused = lyr(x)
     finally:
-        os.environ["OKTOPK_NO_FUSED_MLP"] = "0"
+        os.environ["OKTOPK_FUSED_MLP"] = "0"
+    out_ref = lyr(x)
     assert torch.allclose(out_fused.float(), out_ref.float(), atol=0.1), (
         (out_fused - out_ref).abs().max()
     )
