@@ -220,22 +220,6 @@ class AllReducer:
             torch.cat(vals) if vals else buf[:0].view(torch.float32)
         )
 
-    def _adaptive_bump(self, t: torch.Tensor, tau: float, k: int) -> float:
-        """Raise tau while too many elements are selected (reference
-        add2residual, VGG/compression.py:384-404).  All candidate thresholds
-        tau * scale^i are counted in ONE kernel pass (count_multi_gt) — the
-        sequential 5-iteration loop of the reference costs one full tensor
-        read plus a device sync per iteration."""
-        ok = self.cfg.oktopk
-        m = ok.bump_max_loops
-        taus = [tau * ok.bump_scale ** i for i in range(m + 1)]
-        counts = ops.count_multi_gt(t, taus)
-        # loop semantics: bump while count(current) > 4k/3, at most m times
-        i = 0
-        while i < m and counts[i] > 4 * k // 3:
-            i += 1
-        return taus[i]
-
     # -- Ok-Topk (SURVEY.md section 2.5) --------------------------------
     def _oktopk(self, name: str, t: torch.Tensor, st: TensorState) -> torch.Tensor:
         cfg = self.cfg
@@ -268,19 +252,27 @@ class AllReducer:
             st.boundaries = self._uniform_boundaries(n)
         if it % ok.region_repartition_interval == 0 and P > 1:
             m = idx.numel()
+            # The allreduce below is COLLECTIVE: every rank must reach it on
+            # repartition iterations regardless of its local selection size
+            # (a rank-local `if m >= P: allreduce` would deadlock RCCL when
+            # one rank's selection degenerates). Degenerate ranks contribute
+            # uniform-split quantiles instead.
             if m >= P:
                 step = m // P
                 pos = torch.arange(1, P, dtype=torch.int64, device=idx.device) * step
-                q = comm.to_comm(idx.long()[pos])
-                comm.allreduce_(q)
-                q = (q // P).cpu()
-                b = torch.empty(P + 1, dtype=torch.int64)
-                b[0] = 0
-                b[1:P] = q
-                b[P] = n
-                # guard against degenerate (non-monotone) boundaries
-                if bool((b[1:] >= b[:-1]).all()):
-                    st.boundaries = b
+                q_local = idx.long()[pos]
+            else:
+                q_local = torch.arange(1, P, dtype=torch.int64, device=idx.device) * (n // P)
+            q = comm.to_comm(q_local)
+            comm.allreduce_(q)
+            q = (q // P).cpu()
+            b = torch.empty(P + 1, dtype=torch.int64)
+            b[0] = 0
+            b[1:P] = q
+            b[P] = n
+            # guard against degenerate (non-monotone) boundaries
+            if bool((b[1:] >= b[:-1]).all()):
+                st.boundaries = b
         bounds = st.boundaries
         lo, hi = int(bounds[rank]), int(bounds[rank + 1])
 
