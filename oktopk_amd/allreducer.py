@@ -67,6 +67,7 @@ class TensorState:
     boundaries: Optional[torch.Tensor] = None
     # scratch dense byte-mask for residual credit (allocated lazily)
     mask: Optional[torch.Tensor] = None
+    grad_src: Optional[torch.Tensor] = None  # transient per-run fused-EF source
     values_snapshot: Optional[torch.Tensor] = None  # topkA family residual credit
     indexes_snapshot: Optional[torch.Tensor] = None
 
@@ -123,8 +124,13 @@ class AllReducer:
         self.timers[name][phase] += dt
 
     # ------------------------------------------------------------------
-    def run(self, name: str, tensor: torch.Tensor) -> torch.Tensor:
+    def run(self, name: str, tensor: torch.Tensor,
+            grad_src: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Sparse-allreduce `tensor` (1-D fp32 grad) in place; returns it.
+
+        `grad_src` (optional, bf16): raw model-dtype gradient; the upcast
+        into `tensor` is fused with the error-feedback restore where the
+        compressor path supports it (one streaming pass saved).
 
         The result equals (approximately, by global-top-k truncation) the
         dense mean gradient over all ranks.
@@ -133,6 +139,13 @@ class AllReducer:
         st = self.state(name, t)
         comp = self.cfg.compressor
         ok = self.cfg.oktopk
+        fused_ef = comp in ("oktopk", "topkAopt", "topkSA", "gaussiankSA")
+        if grad_src is not None and not (
+            fused_ef and st.counter >= ok.dense_warmup_iters
+        ):
+            t.copy_(grad_src.reshape(-1).to(t.dtype))
+            grad_src = None
+        st.grad_src = grad_src
 
         eps_ref = None
         if self.cfg.profiling_norm and comp not in ("none", "dense"):
@@ -190,6 +203,15 @@ class AllReducer:
         return t
 
     # -- helpers --------------------------------------------------------
+    def _ef_restore(self, t: torch.Tensor, st: TensorState) -> None:
+        """EF restore+snapshot, fused with the bf16 grad upcast when the
+        caller provided grad_src (see run())."""
+        if st.grad_src is not None:
+            ops.ef_restore_upcast_(t, st.residual, st.grad_src.reshape(-1))
+            st.grad_src = None
+        else:
+            ops.ef_restore_snapshot_(t, st.residual)
+
     def _k(self, n: int) -> int:
         return max(1, int(n * self.cfg.density))
 
@@ -236,7 +258,7 @@ class AllReducer:
         # reference's count-per-candidate loop + separate compact-count pass
         # collapse into one tensor read (ops.compact_adaptive).
         s0 = time.perf_counter()
-        ops.ef_restore_snapshot_(t, st.residual)
+        self._ef_restore(t, st)
         if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
             st.tau_local = ops.kth_abs_value(t, k)
             idx, val = ops.compact_gt(t, st.tau_local)
@@ -421,7 +443,7 @@ class AllReducer:
         it = st.counter
 
         s0 = time.perf_counter()
-        ops.ef_restore_snapshot_(t, st.residual)
+        self._ef_restore(t, st)
         if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
             st.tau_local = ops.kth_abs_value(t, k)
             idx, val = ops.compact_gt(t, st.tau_local)
@@ -516,7 +538,7 @@ class AllReducer:
         it = st.counter
 
         s0 = time.perf_counter()
-        ops.ef_restore_snapshot_(t, st.residual)
+        self._ef_restore(t, st)
         if gaussian:
             tau = _gaussian_threshold(t, self.cfg.density)
             for _ in range(3):

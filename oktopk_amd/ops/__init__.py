@@ -97,6 +97,10 @@ def ef_restore_snapshot_(t: torch.Tensor, residual: torch.Tensor) -> torch.Tenso
     return _backend(t).ef_restore_snapshot_(t, residual)
 
 
+def ef_restore_upcast_(t: torch.Tensor, residual: torch.Tensor, g: torch.Tensor) -> torch.Tensor:
+    return _backend(t).ef_restore_upcast_(t, residual, g)
+
+
 def fused_sgd_(param, grad, momentum_buf, lr, momentum, weight_decay, nesterov):
     return _backend(param).fused_sgd_(
         param, grad, momentum_buf, float(lr), float(momentum), float(weight_decay), bool(nesterov)

@@ -27,6 +27,7 @@ void launch_zero_at(float*, const int32_t*, int64_t, hipStream_t);
 void launch_isin_sorted(const int32_t*, int64_t, const int32_t*, int64_t, bool*,
                         hipStream_t);
 void launch_ef_restore(float*, float*, int64_t, hipStream_t);
+void launch_ef_upcast(float*, float*, const void*, int64_t, hipStream_t);
 void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
                 hipStream_t);
 void launch_adam(float*, const float*, float*, float*, int64_t, float, float, float,
@@ -268,6 +269,19 @@ static torch::Tensor ef_restore_snapshot_(torch::Tensor t, torch::Tensor r) {
     return t;
 }
 
+static torch::Tensor ef_restore_upcast_(torch::Tensor t, torch::Tensor r,
+                                        torch::Tensor g) {
+    check_f32_1d(t, "t");
+    check_f32_1d(r, "r");
+    TORCH_CHECK(g.is_cuda() && g.scalar_type() == torch::kBFloat16 && g.is_contiguous(),
+                "g must be contiguous bf16 on GPU");
+    TORCH_CHECK(t.numel() == r.numel() && t.numel() == g.numel());
+    const at::cuda::CUDAGuard guard(t.device());
+    launch_ef_upcast(t.data_ptr<float>(), r.data_ptr<float>(), g.data_ptr(),
+                     t.numel(), cur_stream());
+    return t;
+}
+
 static void fused_sgd_(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
                        double momentum, double wd, bool nesterov) {
     check_f32_1d(p, "p");
@@ -345,6 +359,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fill_sparse_scaled_", &fill_sparse_scaled_, "out=0; out[idx]=val*scale");
     m.def("isin_sorted", &isin_sorted, "membership of a in sorted b");
     m.def("ef_restore_snapshot_", &ef_restore_snapshot_, "t+=r; r=t (fused)");
+    m.def("ef_restore_upcast_", &ef_restore_upcast_,
+          "t = float(g_bf16) + r; r = t (fused upcast + EF restore)");
     m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
     m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");

@@ -26,6 +26,7 @@ __all__ = [
     "fill_sparse_scaled_",
     "isin_sorted",
     "ef_restore_snapshot_",
+    "ef_restore_upcast_",
     "fused_sgd_",
     "fused_adam_",
     "l2norm",
@@ -111,6 +112,15 @@ def isin_sorted(a: torch.Tensor, b_sorted: torch.Tensor) -> torch.Tensor:
     b64 = b_sorted.long()
     pos = torch.searchsorted(b64, a64).clamp_(max=b64.numel() - 1)
     return b64[pos] == a64
+
+
+def ef_restore_upcast_(t: torch.Tensor, residual: torch.Tensor,
+                       g: torch.Tensor) -> torch.Tensor:
+    """t = float(g) + residual; residual = t (fused upcast + EF restore)."""
+    t.copy_(g.to(t.dtype))
+    t.add_(residual)
+    residual.copy_(t)
+    return t
 
 
 def ef_restore_snapshot_(t: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:

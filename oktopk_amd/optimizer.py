@@ -327,9 +327,10 @@ class FlatBertAdam:
             # volume — the right dense baseline on MI355X), then upcast
             self.reducer.run("flat", self.flat_grad_model)
             self.flat_grad.copy_(self.flat_grad_model)
+        elif self.flat_grad_model is not self.flat_grad:
+            # bf16 grads: the upcast is fused into the engine's EF restore
+            self.reducer.run("flat", self.flat_grad, grad_src=self.flat_grad_model)
         else:
-            if self.flat_grad_model is not self.flat_grad:
-                self.flat_grad.copy_(self.flat_grad_model)  # one bulk upcast
             self.reducer.run("flat", self.flat_grad)
         # 2. grad clip on the reduced gradient (reference optimization.py:197)
         if self.max_grad_norm and self.max_grad_norm > 0:
