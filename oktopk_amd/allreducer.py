@@ -103,6 +103,7 @@ class AllReducer:
         self.states: Dict[str, TensorState] = {}
         self.timers: Dict[str, Dict[str, float]] = {}
         self.eps_log: List[Tuple[int, float]] = []
+        self.train_epoch = 0  # drives the dynamic density schedule
 
     def set_comm(self, comm: Comm) -> None:
         """Swap the communicator after an elastic shrink; per-world state
@@ -212,8 +213,16 @@ class AllReducer:
         else:
             ops.ef_restore_snapshot_(t, st.residual)
 
+    def get_current_density(self) -> float:
+        """Per-epoch density schedule (reference get_current_density,
+        VGG/allreducer.py:265-270)."""
+        sched = self.cfg.dynamic_densities
+        if sched:
+            return float(sched[min(self.train_epoch, len(sched) - 1)])
+        return self.cfg.density
+
     def _k(self, n: int) -> int:
-        return max(1, int(n * self.cfg.density))
+        return max(1, int(n * self.get_current_density()))
 
     def _uniform_boundaries(self, n: int) -> torch.Tensor:
         P = self.comm.size

@@ -58,6 +58,10 @@ class EngineConfig:
 
     compressor: str = "oktopk"
     density: float = 0.001
+    # per-epoch density schedule (reference _dynamic_densities,
+    # VGG/allreducer.py:265-270): density used = schedule[min(epoch, len-1)]
+    # when non-empty, else `density`
+    dynamic_densities: tuple = ()
     # Gradient-bucket merge threshold in bytes (reference groups at 640 MB,
     # /root/reference/VGG/allreducer.py:27; we default far smaller because
     # xGMI-chunked RCCL likes a handful of large-but-not-huge buckets).

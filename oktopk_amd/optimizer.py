@@ -91,8 +91,12 @@ class _DistributedOptimizer:
         named_parameters: Iterable[Tuple[str, torch.Tensor]],
         comm: Optional[Comm] = None,
         cfg: Optional[EngineConfig] = None,
+        max_grad_norm: float = 0.0,
     ):
         self.optimizer = optimizer
+        # post-reduce gradient clipping (reference clips for the LSTM recipe,
+        # VGG/main_trainer.py:96-99); 0 disables
+        self.max_grad_norm = max_grad_norm
         self.comm = comm or Comm(None)
         self.cfg = cfg or EngineConfig()
         self.reducer = AllReducer(self.comm, self.cfg)
@@ -177,6 +181,16 @@ class _DistributedOptimizer:
     def step(self, closure=None):
         if not self.local:
             self.synchronize()
+        if self.max_grad_norm and self.max_grad_norm > 0:
+            sq = 0.0
+            for b in self.buckets:
+                gn = ops.l2norm(b.flat)
+                sq += gn * gn
+            total = sq ** 0.5
+            if total > self.max_grad_norm:
+                scale = self.max_grad_norm / (total + 1e-6)
+                for b in self.buckets:
+                    b.flat.mul_(scale)
         out = self.optimizer.step(closure)
         return out
 
@@ -210,6 +224,7 @@ def DistributedOptimizer(
     compression: Optional[str] = None,
     is_sparse: Optional[bool] = None,
     density: Optional[float] = None,
+    norm_clip: Optional[float] = None,
     **_ignored,
 ) -> _DistributedOptimizer:
     """Factory with the reference's calling convention
@@ -222,7 +237,8 @@ def DistributedOptimizer(
         cfg.compressor = "dense"
     if density is not None:
         cfg.density = density
-    return _DistributedOptimizer(optimizer, named_parameters, comm, cfg)
+    return _DistributedOptimizer(optimizer, named_parameters, comm, cfg,
+                                 max_grad_norm=norm_clip or 0.0)
 
 
 class FlatBertAdam:

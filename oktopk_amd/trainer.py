@@ -240,6 +240,44 @@ class Trainer:
             self._graph_loss = None
             return False
 
+    def set_epoch(self, epoch: int) -> None:
+        """Advance the engine's dynamic density schedule (reference
+        train_epoch plumbing, VGG/allreducer.py:207-208)."""
+        red = getattr(self.opt, "reducer", None)
+        if red is not None:
+            red.train_epoch = epoch
+
+    @torch.no_grad()
+    def evaluate(self, n_batches: int = 4) -> dict:
+        """Synthetic-shape eval (reference DLTrainer.test,
+        VGG/dl_trainer.py:709-784): top-1 for classifiers, greedy-decode WER
+        for CTC, perplexity for LMs, loss for BERT."""
+        from .utils import GreedyDecoder, accuracy_topk, perplexity, wer
+
+        self.model.eval()
+        try:
+            b = self.batches
+            if b.family in ("cifar", "imagenet"):
+                out = self.model(b.x)
+                top1 = accuracy_topk(out.float(), b.y, (1,))[0]
+                return {"top1": top1}
+            if b.family == "ptb":
+                logits, _ = self.model(b.tokens)
+                nll = torch.nn.functional.cross_entropy(
+                    logits.view(-1, logits.size(-1)).float(), b.targets.view(-1)
+                )
+                return {"perplexity": perplexity(float(nll))}
+            if self.model_name == "lstman4":
+                logits = self.model(b.x)
+                dec = GreedyDecoder("_'abcdefghijklmnopqrstuvwxyz ")
+                hyps = dec.decode(logits.float())
+                refs = ["" for _ in hyps]  # synthetic targets have no text
+                return {"wer": sum(wer(h, r) for h, r in zip(hyps, refs)) / len(hyps)}
+            loss = self._forward_loss()
+            return {"loss": float(loss.float())}
+        finally:
+            self.model.train()
+
     def step(self) -> float:
         """One optimizer step (with nsteps_update grad-accumulation substeps)."""
         if self._graph is not None and self.nsteps_update == 1:

@@ -94,3 +94,49 @@ def test_timing_table_populated():
     eng.run("w", t)
     tbl = eng.timing_table("w")
     assert "compress" in tbl and "allgather" in tbl
+
+
+def test_dynamic_density_schedule():
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    cfg = EngineConfig(compressor="oktopk", density=0.5,
+                       dynamic_densities=(0.25, 0.1, 0.05),
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(None), cfg)
+    assert eng.get_current_density() == 0.25
+    eng.train_epoch = 1
+    assert eng.get_current_density() == 0.1
+    eng.train_epoch = 99
+    assert eng.get_current_density() == 0.05
+
+
+def test_trainer_evaluate_cifar():
+    from oktopk_amd.config import EngineConfig, OkTopkConfig
+    from oktopk_amd.trainer import Trainer
+
+    cfg = EngineConfig(compressor="dense")
+    tr = Trainer("caffe_cifar", batch_size=4, cfg=cfg, dtype="fp32")
+    m = tr.evaluate()
+    assert "top1" in m and 0.0 <= m["top1"] <= 100.0
+
+
+def test_sgd_path_grad_clip():
+    import torch
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.optimizer import DistributedOptimizer
+
+    torch.manual_seed(0)
+    model = torch.nn.Linear(10, 10)
+    inner = torch.optim.SGD(model.parameters(), lr=1.0)
+    opt = DistributedOptimizer(inner, model.named_parameters(),
+                               cfg=EngineConfig(compressor="dense"),
+                               norm_clip=0.1)
+    p0 = model.weight.detach().clone()
+    opt.zero_grad()
+    (model(torch.randn(4, 10) * 100).sum()).backward()
+    opt.step()
+    delta = (model.weight.detach() - p0).norm()
+    # lr=1, clipped grad norm <= 0.1 -> total update norm <= ~0.1
+    assert delta <= 0.11, delta
