@@ -210,12 +210,15 @@ class Trainer:
             return False
         if getattr(self.opt, "local", False):
             return False
-        # hook-driven optimizers run comm inside backward - not capturable
+        # hook-driven optimizers: capture with hooks muted (local=True, the
+        # hooks are python callbacks and never fire during graph REPLAY
+        # anyway); bucket reduces then run post-replay from synchronize()
         from .optimizer import _DistributedOptimizer
 
-        if isinstance(self.opt, _DistributedOptimizer):
-            return False
+        hook_driven = isinstance(self.opt, _DistributedOptimizer)
         try:
+            if hook_driven:
+                self.opt.local = True
             torch.cuda.synchronize()
             self.opt.zero_grad()
             side = torch.cuda.Stream()
@@ -239,6 +242,9 @@ class Trainer:
             self._graph = None
             self._graph_loss = None
             return False
+        finally:
+            if hook_driven:
+                self.opt.local = False
 
     def set_epoch(self, epoch: int) -> None:
         """Advance the engine's dynamic density schedule (reference
