@@ -393,10 +393,7 @@ class AllReducer:
             if st.mask is None or st.mask.numel() != n:
                 st.mask = torch.zeros(n, dtype=torch.bool, device=t.device)
             st.mask[g_sel_idx.long()] = True
-            member = st.mask[idx.long()]
-            involved = idx[member]
-            if involved.numel():
-                ops.zero_at_(st.residual, involved)
+            ops.zero_at_masked_(st.residual, idx, st.mask)
             st.mask[g_sel_idx.long()] = False  # cheap sparse reset
         self._time(name, "merge", time.perf_counter() - s4)
         return result

@@ -83,6 +83,10 @@ def zero_at_(t: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
     return _backend(t).zero_at_(t, idx)
 
 
+def zero_at_masked_(t: torch.Tensor, idx: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
+    return _backend(t).zero_at_masked_(t, idx, mask)
+
+
 def fill_sparse_scaled_(
     out: torch.Tensor, idx: torch.Tensor, val: torch.Tensor, scale: float
 ) -> torch.Tensor:

@@ -24,6 +24,7 @@ void launch_scatter_add(float*, const int32_t*, const float*, int64_t, hipStream
 void launch_scatter_set_scaled(float*, const int32_t*, const float*, float, int64_t,
                                hipStream_t);
 void launch_zero_at(float*, const int32_t*, int64_t, hipStream_t);
+void launch_zero_at_masked(float*, const int32_t*, const bool*, int64_t, hipStream_t);
 void launch_isin_sorted(const int32_t*, int64_t, const int32_t*, int64_t, bool*,
                         hipStream_t);
 void launch_ef_restore(float*, float*, int64_t, hipStream_t);
@@ -234,6 +235,19 @@ static torch::Tensor zero_at_(torch::Tensor t, torch::Tensor idx) {
     return t;
 }
 
+static torch::Tensor zero_at_masked_(torch::Tensor t, torch::Tensor idx,
+                                     torch::Tensor mask) {
+    check_f32_1d(t, "t");
+    check_i32_1d(idx, "idx");
+    TORCH_CHECK(mask.is_cuda() && mask.scalar_type() == torch::kBool &&
+                mask.is_contiguous(), "mask must be contiguous bool on GPU");
+    const at::cuda::CUDAGuard guard(t.device());
+    if (idx.numel())
+        launch_zero_at_masked(t.data_ptr<float>(), idx.data_ptr<int32_t>(),
+                              mask.data_ptr<bool>(), idx.numel(), cur_stream());
+    return t;
+}
+
 static torch::Tensor fill_sparse_scaled_(torch::Tensor out, torch::Tensor idx,
                                          torch::Tensor val, double scale) {
     check_f32_1d(out, "out");
@@ -356,6 +370,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("kth_abs_value", &kth_abs_value, "exact k-th largest |t| via radix select");
     m.def("scatter_add_", &scatter_add_, "dest[idx] += val");
     m.def("zero_at_", &zero_at_, "t[idx] = 0");
+    m.def("zero_at_masked_", &zero_at_masked_, "t[idx[i]] = 0 where mask[idx[i]]");
     m.def("fill_sparse_scaled_", &fill_sparse_scaled_, "out=0; out[idx]=val*scale");
     m.def("isin_sorted", &isin_sorted, "membership of a in sorted b");
     m.def("ef_restore_snapshot_", &ef_restore_snapshot_, "t+=r; r=t (fused)");

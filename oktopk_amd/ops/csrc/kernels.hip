@@ -376,6 +376,27 @@ extern "C" void launch_scatter_set_scaled(float* dest, const int32_t* idx,
                        stream, dest, idx, val, scale, m);
 }
 
+// residual credit (reference intersect1d + update_residuals,
+// VGG/allreducer.py:844-845): zero residual[idx[i]] where mask[idx[i]] —
+// one kernel, no masked-select round trip
+__global__ void zero_at_masked_kernel(float* __restrict__ dest,
+                                      const int32_t* __restrict__ idx,
+                                      const bool* __restrict__ mask, int64_t m) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    for (; i < m; i += stride) {
+        int32_t j = idx[i];
+        if (mask[j]) dest[j] = 0.f;
+    }
+}
+
+extern "C" void launch_zero_at_masked(float* dest, const int32_t* idx,
+                                      const bool* mask, int64_t m,
+                                      hipStream_t stream) {
+    hipLaunchKernelGGL(zero_at_masked_kernel, dim3(n_blocks(m, 4)), dim3(BLOCK), 0,
+                       stream, dest, idx, mask, m);
+}
+
 __global__ void zero_at_kernel(float* __restrict__ dest,
                                const int32_t* __restrict__ idx, int64_t m) {
     int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;

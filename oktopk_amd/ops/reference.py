@@ -23,6 +23,7 @@ __all__ = [
     "compact_adaptive",
     "scatter_add_",
     "zero_at_",
+    "zero_at_masked_",
     "fill_sparse_scaled_",
     "isin_sorted",
     "ef_restore_snapshot_",
@@ -86,6 +87,14 @@ def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> to
 def zero_at_(t: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
     """t[idx] = 0. Reference: update_residuals, VGG/compression.py:467-471."""
     t.reshape(-1)[idx.long()] = 0
+    return t
+
+
+def zero_at_masked_(t: torch.Tensor, idx: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
+    """t[idx[i]] = 0 wherever mask[idx[i]] (fused residual credit)."""
+    j = idx.long()
+    sel = j[mask[j]]
+    t.reshape(-1)[sel] = 0
     return t
 
 

@@ -114,6 +114,7 @@ def main(argv=None) -> int:
 
     samples_per_iter = args.batch_size * comm.size
     for epoch in range(start_epoch, args.max_epochs):
+        trainer.set_epoch(epoch)
         t0 = time.time()
         for it in range(args.iters_per_epoch):
             loss = trainer.step()

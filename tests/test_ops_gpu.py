@@ -239,3 +239,16 @@ def test_bert_layer_uses_fused_mlp():
     assert torch.allclose(out_fused.float(), out_ref.float(), atol=0.1), (
         (out_fused - out_ref).abs().max()
     )
+
+
+def test_zero_at_masked():
+    t = randn_gpu(1000, seed=50)
+    ref = t.cpu().clone()
+    idx = torch.tensor([5, 100, 200, 999], dtype=torch.int32).cuda()
+    mask = torch.zeros(1000, dtype=torch.bool).cuda()
+    mask[100] = True
+    mask[999] = True
+    hip().zero_at_masked_(t, idx, mask)
+    R.zero_at_masked_(ref, idx.cpu(), mask.cpu())
+    assert torch.equal(t.cpu(), ref)
+    assert t[100].item() == 0 and t[999].item() == 0 and t[5].item() != 0
