@@ -1,0 +1,99 @@
+"""Engine edge cases that only bite at scale: all-zero gradients (empty
+selections everywhere), tiny tensors (n < world), density=1.0, empty
+alltoallv legs."""
+import pytest
+import torch
+
+from conftest import run_dist
+
+from oktopk_amd import AllReducer, Comm, EngineConfig
+from oktopk_amd.config import OkTopkConfig
+
+
+def make(comp="oktopk", density=0.05, **okkw):
+    okkw.setdefault("dense_warmup_iters", 0)
+    return AllReducer(Comm(None), EngineConfig(compressor=comp, density=density,
+                                               oktopk=OkTopkConfig(**okkw)))
+
+
+def test_all_zero_gradient_world1():
+    eng = make()
+    t = torch.zeros(4096)
+    out = eng.run("w", t)
+    assert out.abs().sum() == 0
+    # and again (thresholds now zero/degenerate)
+    out = eng.run("w", torch.zeros(4096))
+    assert torch.isfinite(out).all()
+
+
+def test_tiny_tensor_world1():
+    eng = make(density=1.0)
+    for it in range(3):
+        t = torch.randn(3, generator=torch.Generator().manual_seed(it))
+        out = eng.run("w", t)
+        assert torch.isfinite(out).all()
+
+
+def test_density_one_world1():
+    eng = make(density=1.0)
+    t = torch.randn(1000, generator=torch.Generator().manual_seed(0))
+    ref = t.clone()
+    out = eng.run("w", t)
+    # everything above-threshold except ties at the minimum |value|
+    assert (out != 0).sum() >= 990
+    nz = out.nonzero().view(-1)
+    assert torch.allclose(out[nz], ref[nz])
+
+
+def _zero_grads_dist(rank):
+    import torch.distributed as dist
+
+    eng = AllReducer(Comm(dist.group.WORLD),
+                     EngineConfig(compressor="oktopk", density=0.05,
+                                  oktopk=OkTopkConfig(dense_warmup_iters=0)))
+    for it in range(3):
+        out = eng.run("w", torch.zeros(2048))
+        assert out.abs().sum() == 0
+
+
+def test_all_zero_gradient_world2():
+    run_dist(_zero_grads_dist, 2)
+
+
+def _one_rank_zero(rank):
+    """One rank has zero grads (empty selection), the other doesn't —
+    exercises empty alltoallv legs and empty allgather segments."""
+    import torch.distributed as dist
+
+    eng = AllReducer(Comm(dist.group.WORLD),
+                     EngineConfig(compressor="oktopk", density=0.05,
+                                  oktopk=OkTopkConfig(dense_warmup_iters=0)))
+    for it in range(4):
+        if rank == 0:
+            t = torch.zeros(2048)
+        else:
+            t = torch.randn(2048, generator=torch.Generator().manual_seed(it))
+        out = eng.run("w", t)
+        assert torch.isfinite(out).all()
+
+
+def test_one_rank_all_zero_world2():
+    run_dist(_one_rank_zero, 2)
+
+
+def _tiny_dist(rank):
+    import torch.distributed as dist
+
+    eng = AllReducer(Comm(dist.group.WORLD),
+                     EngineConfig(compressor="oktopk", density=1.0,
+                                  oktopk=OkTopkConfig(
+                                      dense_warmup_iters=0,
+                                      region_repartition_interval=1)))
+    for it in range(3):
+        t = torch.randn(3, generator=torch.Generator().manual_seed(100 * rank + it))
+        out = eng.run("w", t)
+        assert torch.isfinite(out).all()
+
+
+def test_tiny_tensor_world2_repartition_every_step():
+    run_dist(_tiny_dist, 2)
