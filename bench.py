@@ -28,6 +28,8 @@ def parse_args():
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--density", type=float, default=0.001)
     p.add_argument("--compressor", type=str, default="oktopk")
+    p.add_argument("--wire-dtype", type=str, default="bf16", choices=["bf16", "fp32"],
+                   help="dtype of sparse values on the wire (indices stay int32)")
     p.add_argument("--dense-baseline-steps", type=int, default=-1,
                    help="steps for the in-run dense baseline (-1: min(steps,10); 0: skip)")
     return p.parse_args()
@@ -58,7 +60,7 @@ def build_trainer(args, comm, compressor):
         "lstm" if args.model == "lstman4" else "vgg"
     )
     cfg = EngineConfig.preset(preset, compressor=compressor, density=args.density,
-                              dense_warmup_iters=0)
+                              wire_dtype=args.wire_dtype, dense_warmup_iters=0)
     dtype = "bf16" if torch.cuda.is_available() else "fp32"
     return Trainer(
         model_name=args.model,
@@ -153,6 +155,7 @@ def main():
                 "parallelism": f"dp{n_gpus}",
                 "compressor": args.compressor,
                 "density": args.density,
+                "wire_dtype": args.wire_dtype,
                 "allreduce_ms_per_step": round(allreduce_ms, 3),
                 "comm_ms_per_step": round(comm_ms, 3),
                 "dense_ms_per_step": round(dense_ms, 3) if dense_ms else None,

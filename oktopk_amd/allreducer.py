@@ -233,20 +233,40 @@ class AllReducer:
         b[P] = n
         return b
 
+    @property
+    def _wire_bf16(self) -> bool:
+        return self.cfg.wire_dtype == "bf16"
+
+    def _pack_ints(self, n: int) -> int:
+        """int32 words used by a packed segment of n (idx, val) pairs."""
+        if self._wire_bf16:
+            return n + (n + 1) // 2  # idx i32 + val bf16 padded to even
+        return 2 * n
+
     def _pack(self, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
-        """Pack (int32 idx, fp32 val) into one int32 buffer [idx | val bits]."""
+        """Pack (int32 idx, values) into one int32 buffer [idx | val bits].
+        Wire values are fp32 or bf16 per cfg.wire_dtype (bf16 halves the
+        sparse message volume; local EF state stays fp32)."""
+        if self._wire_bf16:
+            v = val.to(torch.bfloat16)
+            if v.numel() % 2:
+                v = torch.cat([v, v.new_zeros(1)])
+            return torch.cat([idx.view(torch.int32), v.view(torch.int32)])
         return torch.cat([idx.view(torch.int32), val.view(torch.int32)])
 
-    @staticmethod
-    def _unpack(buf: torch.Tensor, sizes: List[int]) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Unpack rank-ordered concatenation of packed segments."""
+    def _unpack(self, buf: torch.Tensor, counts: List[int]) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Unpack rank-ordered packed segments; `counts` are ELEMENT counts
+        per segment.  Values come back fp32."""
         idxs, vals = [], []
         off = 0
-        for s in sizes:
-            half = s // 2
-            idxs.append(buf[off : off + half])
-            vals.append(buf[off + half : off + s].view(torch.float32))
-            off += s
+        for n in counts:
+            ints = self._pack_ints(n)
+            idxs.append(buf[off : off + n])
+            if self._wire_bf16:
+                vals.append(buf[off + n : off + ints].view(torch.bfloat16)[:n].float())
+            else:
+                vals.append(buf[off + n : off + ints].view(torch.float32))
+            off += ints
         return torch.cat(idxs) if idxs else buf[:0], (
             torch.cat(vals) if vals else buf[:0].view(torch.float32)
         )
@@ -322,17 +342,18 @@ class AllReducer:
                 idx.long(), bounds[1:P].to(idx.device)
             ).cpu()
             cuts = [0] + [int(x) for x in split_pts] + [sel]
-            send_counts = [2 * (cuts[i + 1] - cuts[i]) for i in range(P)]
-            # pack per destination: [idx|valbits] per region segment
-            segs = []
-            for i in range(P):
-                a, b_ = cuts[i], cuts[i + 1]
-                segs.append(idx[a:b_].view(torch.int32))
-                segs.append(val[a:b_].view(torch.int32))
+            elem_counts = [cuts[i + 1] - cuts[i] for i in range(P)]
+            # pack per destination: [idx | val bits] per region segment
+            segs = [self._pack(idx[cuts[i] : cuts[i + 1]], val[cuts[i] : cuts[i + 1]])
+                    for i in range(P)]
             send = torch.cat(segs) if segs else idx.view(torch.int32)[:0]
-            recv_counts = comm.alltoall_sizes(send_counts, comm.device)
-            recv = comm.alltoallv(comm.to_comm(send), send_counts, recv_counts)
-            r_idx, r_val = self._unpack(recv, recv_counts)
+            recv_elems = comm.alltoall_sizes(elem_counts, comm.device)
+            recv = comm.alltoallv(
+                comm.to_comm(send),
+                [self._pack_ints(c) for c in elem_counts],
+                [self._pack_ints(c) for c in recv_elems],
+            )
+            r_idx, r_val = self._unpack(recv, recv_elems)
         else:
             r_idx, r_val = idx, val
         self._time(name, "alltoall", time.perf_counter() - s1)
@@ -354,8 +375,9 @@ class AllReducer:
         gidx = gidx + lo  # absolute indices (int32 + int offset)
 
         pack = comm.to_comm(self._pack(gidx, gval))
-        buf, sizes = comm.allgatherv(pack)
-        all_idx, all_val = self._unpack(buf, sizes)
+        elem_counts = [int(x) for x in comm.allgather_sizes(gidx.numel(), comm.device)]
+        buf, _ = comm.allgatherv(pack, sizes=[self._pack_ints(c) for c in elem_counts])
+        all_idx, all_val = self._unpack(buf, elem_counts)
         all_idx = all_idx.to(t.device)
         all_val = all_val.to(t.device)
         self._time(name, "allgather", time.perf_counter() - s3)
@@ -424,7 +446,7 @@ class AllReducer:
         self._time(name, "allgather", time.perf_counter() - s1)
 
         s2 = time.perf_counter()
-        all_idx, all_val = self._unpack(buf, [2 * k] * P)
+        all_idx, all_val = self._unpack(buf, [k] * P)
         result = t
         result.zero_()
         ops.scatter_add_(result, all_idx.to(t.device), all_val.to(t.device))
@@ -467,8 +489,9 @@ class AllReducer:
 
         s1 = time.perf_counter()
         pack = comm.to_comm(self._pack(idx, val))
-        buf, sizes = comm.allgatherv(pack)
-        all_idx, all_val = self._unpack(buf, sizes)
+        elem_counts = [int(x) for x in comm.allgather_sizes(idx.numel(), comm.device)]
+        buf, _ = comm.allgatherv(pack, sizes=[self._pack_ints(c) for c in elem_counts])
+        all_idx, all_val = self._unpack(buf, elem_counts)
         self._time(name, "allgather", time.perf_counter() - s1)
 
         s2 = time.perf_counter()
@@ -507,8 +530,9 @@ class AllReducer:
 
         s1 = time.perf_counter()
         pack = comm.to_comm(self._pack(idx, val))
-        buf, sizes = comm.allgatherv(pack)
-        all_idx, all_val = self._unpack(buf, sizes)
+        elem_counts = [int(x) for x in comm.allgather_sizes(idx.numel(), comm.device)]
+        buf, _ = comm.allgatherv(pack, sizes=[self._pack_ints(c) for c in elem_counts])
+        all_idx, all_val = self._unpack(buf, elem_counts)
         self._time(name, "allgather", time.perf_counter() - s1)
 
         s2 = time.perf_counter()
@@ -582,16 +606,17 @@ class AllReducer:
         if P > 1:
             split_pts = torch.searchsorted(idx.long(), bounds[1:P].to(idx.device)).cpu()
             cuts = [0] + [int(x) for x in split_pts] + [sel]
-            send_counts = [2 * (cuts[i + 1] - cuts[i]) for i in range(P)]
-            segs = []
-            for i in range(P):
-                a, b_ = cuts[i], cuts[i + 1]
-                segs.append(idx[a:b_].view(torch.int32))
-                segs.append(val[a:b_].view(torch.int32))
+            elem_counts = [cuts[i + 1] - cuts[i] for i in range(P)]
+            segs = [self._pack(idx[cuts[i] : cuts[i + 1]], val[cuts[i] : cuts[i + 1]])
+                    for i in range(P)]
             send = torch.cat(segs)
-            recv_counts = comm.alltoall_sizes(send_counts, comm.device)
-            recv = comm.alltoallv(comm.to_comm(send), send_counts, recv_counts)
-            r_idx, r_val = self._unpack(recv, recv_counts)
+            recv_elems = comm.alltoall_sizes(elem_counts, comm.device)
+            recv = comm.alltoallv(
+                comm.to_comm(send),
+                [self._pack_ints(c) for c in elem_counts],
+                [self._pack_ints(c) for c in recv_elems],
+            )
+            r_idx, r_val = self._unpack(recv, recv_elems)
         else:
             r_idx, r_val = idx, val
         self._time(name, "alltoall", time.perf_counter() - s1)
@@ -622,8 +647,9 @@ class AllReducer:
         else:
             gidx = gidx + lo
             pack = comm.to_comm(self._pack(gidx, gval))
-            buf, sizes = comm.allgatherv(pack)
-            all_idx, all_val = self._unpack(buf, sizes)
+            elem_counts = [int(x) for x in comm.allgather_sizes(gidx.numel(), comm.device)]
+            buf, _ = comm.allgatherv(pack, sizes=[self._pack_ints(c) for c in elem_counts])
+            all_idx, all_val = self._unpack(buf, elem_counts)
             ops.fill_sparse_scaled_(
                 result, all_idx.to(t.device), all_val.to(t.device), 1.0 / P
             )
@@ -667,7 +693,7 @@ class AllReducer:
                         other = torch.empty_like(pack)
                         rr = comm.irecv(other, src=peer, tag=7)
                         rr.wait()
-                        o_idx, o_val = self._unpack(other, [other.numel()])
+                        o_idx, o_val = self._unpack(other, [k])
                         merged = torch.zeros(n, dtype=t.dtype, device=t.device)
                         ops.scatter_add_(merged, cur_idx.to(t.device), cur_val.to(t.device))
                         ops.scatter_add_(merged, o_idx.to(t.device), o_val.to(t.device))
@@ -681,10 +707,10 @@ class AllReducer:
                 peer_dist *= 2
             # root (rank 0) holds the winner; broadcast 2k packet
             final = self._pack(cur_idx, cur_val) if rank == 0 else torch.empty(
-                2 * k, dtype=torch.int32, device=comm.device
+                self._pack_ints(k), dtype=torch.int32, device=comm.device
             )
             comm.broadcast_(final, src=0)
-            g_idx, g_val = self._unpack(final, [2 * k])
+            g_idx, g_val = self._unpack(final, [k])
         else:
             g_idx, g_val = idx, val
         self._time(name, "allreduce", time.perf_counter() - s1)

@@ -174,3 +174,53 @@ def test_oktopk_values_exact_on_full_support():
 
 def test_world4_oktopk():
     run_dist(_mass_conservation, 4, args=("oktopk",))
+
+
+def _bf16_wire(rank):
+    """bf16 wire values: result must match the fp32-wire engine within bf16
+    rounding on the selected entries."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    def build(wire):
+        return AllReducer(
+            Comm(dist.group.WORLD),
+            EngineConfig(compressor="oktopk", density=DENSITY, wire_dtype=wire,
+                         oktopk=OkTopkConfig(dense_warmup_iters=0)),
+        )
+
+    e32, e16 = build("fp32"), build("bf16")
+    for it in range(4):
+        t = _grad(rank, it)
+        o32 = e32.run("w", t.clone())
+        o16 = e16.run("w", t.clone())
+        nz = o32.nonzero().view(-1)
+        # selected support should largely agree; values within bf16 rounding
+        assert torch.allclose(o16[nz], o32[nz], rtol=0.02, atol=1e-3), (
+            (o16[nz] - o32[nz]).abs().max()
+        )
+
+
+def test_bf16_wire_world2():
+    run_dist(_bf16_wire, 2)
+
+
+def _bf16_wire_all_compressors(rank):
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    for comp in ("topkA", "gtopk", "topkSA", "gaussiank"):
+        eng = AllReducer(
+            Comm(dist.group.WORLD),
+            EngineConfig(compressor=comp, density=0.05, wire_dtype="bf16",
+                         oktopk=OkTopkConfig(dense_warmup_iters=0)),
+        )
+        for it in range(3):
+            out = eng.run("w", _grad(rank, it, 4096))
+            assert torch.isfinite(out).all()
+
+
+def test_bf16_wire_topkA_gtopk_world2():
+    run_dist(_bf16_wire_all_compressors, 2)
