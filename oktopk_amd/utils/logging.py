@@ -56,6 +56,34 @@ class MetricWriter:
                 + "\n"
             )
 
+    def add_histogram(self, tag: str, values, step: int, bins: int = 16) -> None:
+        """Weight-histogram parity with tensorboardX (reference
+        dl_trainer.py:572-577): stores quantiles + moments as JSON."""
+        if not self._f:
+            return
+        import torch
+
+        v = values.detach().float().reshape(-1) if hasattr(values, "detach") else None
+        if v is None or v.numel() == 0:
+            return
+        qs = torch.quantile(
+            v, torch.linspace(0, 1, bins + 1, device=v.device)
+        ).cpu().tolist()
+        self._f.write(
+            json.dumps(
+                {
+                    "wall": time.time(),
+                    "step": step,
+                    "tag": tag,
+                    "hist_quantiles": [round(float(x), 6) for x in qs],
+                    "mean": float(v.mean()),
+                    "std": float(v.std()),
+                }
+            )
+            + "
+"
+        )
+
     def add_dict(self, scalars: dict, step: int) -> None:
         for k, v in scalars.items():
             if isinstance(v, (int, float)):

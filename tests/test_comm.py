@@ -74,3 +74,27 @@ def test_world1_noop_comm():
     assert torch.equal(buf, ref) and sizes == [4]
     out = c.alltoallv(t, [4], [4])
     assert torch.equal(out, ref)
+
+
+def _ddp_worker(rank):
+    import torch
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.ddp_trainer import ssgd_with_ddp
+    import torch.distributed as dist
+
+    model, opt = ssgd_with_ddp("caffe_cifar", comm=Comm(dist.group.WORLD))
+    x = torch.randn(2, 3, 32, 32, generator=torch.Generator().manual_seed(rank))
+    y = torch.randint(0, 10, (2,), generator=torch.Generator().manual_seed(rank))
+    loss = torch.nn.functional.cross_entropy(model(x), y)
+    opt.zero_grad()
+    loss.backward()
+    opt.step()
+    # DDP keeps replicas in sync: parameters identical across ranks
+    p = next(model.parameters()).detach().clone()
+    p0 = p.clone()
+    dist.broadcast(p0, src=0)
+    assert torch.allclose(p, p0)
+
+
+def test_ddp_alternative_trainer():
+    run_dist(_ddp_worker, 2)
