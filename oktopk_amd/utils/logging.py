@@ -80,8 +80,7 @@ class MetricWriter:
                     "std": float(v.std()),
                 }
             )
-            + "
-"
+            + "\n"
         )
 
     def add_dict(self, scalars: dict, step: int) -> None:
