@@ -194,6 +194,8 @@ class PipelineRuntime:
         device: Optional[torch.device] = None,
         act_dtype: torch.dtype = torch.float32,
         comm_device: Optional[torch.device] = None,
+        prev_rank: Optional[int] = None,
+        next_rank: Optional[int] = None,
     ):
         self.stage = stage
         self.stage_id = stage_id
@@ -203,6 +205,10 @@ class PipelineRuntime:
         self.comm = StageComm(device=comm_device or torch.device("cpu"))
         self.is_first = stage_id == 0
         self.is_last = stage_id == num_stages - 1
+        # global ranks of the pipeline neighbours; defaults assume pure PP
+        # (rank == stage). Hybrid DPxPP passes the replica-chain ranks.
+        self.prev_rank = prev_rank if prev_rank is not None else stage_id - 1
+        self.next_rank = next_rank if next_rank is not None else stage_id + 1
         self.stats = RuntimeStats()
 
     # -- GPipe flush schedule (reference runtime.py:842) -----------------
@@ -215,6 +221,13 @@ class PipelineRuntime:
         step.  Each microbatch dict carries the stage's locally-needed fields:
         first stage: input_ids/token_type/attn; last stage: attn + labels;
         intermediate: attn."""
+        # with a hook-driven DistributedOptimizer, gradients accumulate over
+        # ALL microbatches before the (sparse) allreduce fires once from
+        # step()->synchronize(); per-microbatch hook reduces would desync
+        # the DP replicas mid-accumulation
+        has_local = hasattr(optimizer, "local")
+        if has_local:
+            optimizer.local = True
         if self.num_stages == 1:
             total = 0.0
             optimizer.zero_grad()
@@ -222,6 +235,8 @@ class PipelineRuntime:
                 loss = self.stage(**mb)
                 loss.backward()
                 total += float(loss.detach().float())
+            if has_local:
+                optimizer.local = False
             optimizer.step()
             return total / max(len(microbatches), 1)
 
@@ -232,16 +247,16 @@ class PipelineRuntime:
         for mb in microbatches:
             if self.is_first:
                 out = self.stage(**mb)
-                self.comm.send(out.detach(), dst=self.stage_id + 1)
+                self.comm.send(out.detach(), dst=self.next_rank)
                 fwd_acts.append((None, out))
             else:
                 hidden = self.comm.recv(
-                    src=self.stage_id - 1, dtype=self.act_dtype, device=self.device
+                    src=self.prev_rank, dtype=self.act_dtype, device=self.device
                 )
                 hidden.requires_grad_(True)
                 out = self.stage(hidden, **mb)
                 if not self.is_last:
-                    self.comm.send(out.detach(), dst=self.stage_id + 1)
+                    self.comm.send(out.detach(), dst=self.next_rank)
                 else:
                     losses.append(out)
                 fwd_acts.append((hidden, out))
@@ -255,12 +270,14 @@ class PipelineRuntime:
                 total_loss += float(loss.detach().float())
             else:
                 grad_out = self.comm.recv(
-                    src=self.stage_id + 1, dtype=self.act_dtype, device=self.device
+                    src=self.next_rank, dtype=self.act_dtype, device=self.device
                 )
                 out.backward(grad_out)
             if not self.is_first:
-                self.comm.send(hidden_in.grad, dst=self.stage_id - 1)
+                self.comm.send(hidden_in.grad, dst=self.prev_rank)
         self.comm.flush()
+        if has_local:
+            optimizer.local = False
         optimizer.step()
         return total_loss / max(len(microbatches), 1)
 
@@ -273,6 +290,9 @@ class PipelineRuntime:
             return self.run_step_with_flushes(microbatches, optimizer)
         n = len(microbatches)
         warmup = min(self.num_stages - 1 - self.stage_id, n)
+        has_local = hasattr(optimizer, "local")
+        if has_local:
+            optimizer.local = True
         optimizer.zero_grad()
         fwd_q: deque = deque()
         total_loss = 0.0
@@ -282,16 +302,16 @@ class PipelineRuntime:
             nonlocal fwd_i
             if self.is_first:
                 out = self.stage(**mb)
-                self.comm.send(out.detach(), dst=self.stage_id + 1)
+                self.comm.send(out.detach(), dst=self.next_rank)
                 fwd_q.append((None, out))
             else:
                 hidden = self.comm.recv(
-                    src=self.stage_id - 1, dtype=self.act_dtype, device=self.device
+                    src=self.prev_rank, dtype=self.act_dtype, device=self.device
                 )
                 hidden.requires_grad_(True)
                 out = self.stage(hidden, **mb)
                 if not self.is_last:
-                    self.comm.send(out.detach(), dst=self.stage_id + 1)
+                    self.comm.send(out.detach(), dst=self.next_rank)
                 fwd_q.append((hidden, out))
             fwd_i += 1
 
@@ -303,11 +323,11 @@ class PipelineRuntime:
                 total_loss += float(out.detach().float())
             else:
                 grad_out = self.comm.recv(
-                    src=self.stage_id + 1, dtype=self.act_dtype, device=self.device
+                    src=self.next_rank, dtype=self.act_dtype, device=self.device
                 )
                 out.backward(grad_out)
             if not self.is_first:
-                self.comm.send(hidden_in.grad, dst=self.stage_id - 1)
+                self.comm.send(hidden_in.grad, dst=self.prev_rank)
 
         for _ in range(warmup):
             do_forward(microbatches[fwd_i])
@@ -317,6 +337,8 @@ class PipelineRuntime:
         while fwd_q:
             do_backward()
         self.comm.flush()
+        if has_local:
+            optimizer.local = False
         optimizer.step()
         return total_loss / max(n, 1)
 
@@ -388,3 +410,33 @@ class OptimizerWithWeightStashing:
         if self.num_versions > 1:
             self.queue.popleft()
             self.queue.append(self._clone())
+
+
+# ---------------------------------------------------------------------------
+# hybrid DP x PP (reference StageRuntime's DDP-group code paths,
+# BERT/runtime.py:278-322, which ship commented out — functional here)
+# ---------------------------------------------------------------------------
+
+def make_hybrid_groups(num_stages: int, dp_degree: int):
+    """Partition WORLD = num_stages * dp_degree ranks into a stage grid.
+
+    Rank layout: ranks [s*dp_degree, (s+1)*dp_degree) form stage s; replica d
+    is the pipeline chain {d, dp_degree+d, 2*dp_degree+d, ...}.  Returns
+    (stage_id, replica_id, dp_comm, prev_rank, next_rank); dp_comm wraps the
+    same-stage process group (use it as the Comm of a DistributedOptimizer to
+    run Ok-Topk sparse allreduce WITHIN each stage's replicas).
+    Every rank must call this collectively (new_group semantics)."""
+    world = dist.get_world_size()
+    assert world == num_stages * dp_degree, (world, num_stages, dp_degree)
+    rank = dist.get_rank()
+    stage_id = rank // dp_degree
+    replica_id = rank % dp_degree
+    dp_group = None
+    for s in range(num_stages):
+        ranks = list(range(s * dp_degree, (s + 1) * dp_degree))
+        g = dist.new_group(ranks=ranks)
+        if s == stage_id:
+            dp_group = g
+    prev_rank = rank - dp_degree if stage_id > 0 else None
+    next_rank = rank + dp_degree if stage_id < num_stages - 1 else None
+    return stage_id, replica_id, Comm(dp_group), prev_rank, next_rank

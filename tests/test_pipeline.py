@@ -128,3 +128,54 @@ def test_weight_stashing_semantics():
     )
     stash.load_new_params()
     assert torch.allclose(m.weight.detach(), w_v1)
+
+
+def _hybrid_worker(rank):
+    """World 4 = 2 stages x 2 DP replicas; each stage's gradients are
+    allreduced across its replicas by the sparse engine (dense mode here for
+    exact equivalence).  Check: dp peers stay in sync, and stage-0 grads
+    equal the unsplit DP-2 reference."""
+    import torch.distributed as dist
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.optimizer import DistributedOptimizer
+    from oktopk_amd.pipeline import PipelineRuntime, make_hybrid_groups, partition_bert
+
+    stage_id, replica_id, dp_comm, prev_rank, next_rank = make_hybrid_groups(2, 2)
+    model = _model()
+    stages = partition_bert(model, 2)
+    stage = stages[stage_id]
+    rt = PipelineRuntime(stage, stage_id=stage_id, num_stages=2,
+                         prev_rank=prev_rank, next_rank=next_rank)
+    inner = torch.optim.SGD(stage.parameters(), lr=0.0)
+    opt = DistributedOptimizer(inner, stage.named_parameters(), comm=dp_comm,
+                               cfg=EngineConfig(compressor="dense"))
+    # replica d trains on shard d (2 microbatches each)
+    mbs = _microbatches(4, seed=1)[2 * replica_id : 2 * replica_id + 2]
+    if stage_id == 0:
+        my = [{k: m[k] for k in ("input_ids", "token_type_ids", "attention_mask")}
+              for m in mbs]
+    else:
+        my = [{k: m[k] for k in ("attention_mask", "masked_lm_labels",
+                                 "next_sentence_label")} for m in mbs]
+    rt.run_step_with_flushes(my, opt)
+
+    # dp peers in sync: grads identical across the stage group
+    g = stage.parameters().__next__().grad.clone()
+    g0 = g.clone()
+    dist.broadcast(g0, src=stage_id * 2, group=dp_comm.group)
+    assert torch.allclose(g, g0, atol=1e-6)
+
+    if stage_id == 0:
+        # compare against unsplit model running all 4 microbatches (DP mean)
+        ref = _model()
+        ref.zero_grad()
+        for m in _microbatches(4, seed=1):
+            ref(**m).backward()
+        want = ref.bert.layer[0].fc1.weight.grad / 2  # dense allreduce /P
+        got = stage.layers[0].fc1.weight.grad
+        assert torch.allclose(got, want, atol=1e-5), (got - want).abs().max()
+
+
+def test_hybrid_dp_pp_world4():
+    run_dist(_hybrid_worker, 4)
