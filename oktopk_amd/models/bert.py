@@ -91,10 +91,21 @@ class BertLayer(nn.Module):
         self.drop = nn.Dropout(cfg.hidden_dropout_prob)
 
     def forward(self, x, attn_mask=None):
-        x = self.ln1(x + self.drop(self.attn(x, attn_mask)))
+        a = self.drop(self.attn(x, attn_mask))
+        x = self._add_ln(x, a, self.ln1)
         h = self._mlp_act(x)
-        x = self.ln2(x + self.drop(self.fc2(h)))
+        o = self.drop(self.fc2(h))
+        x = self._add_ln(x, o, self.ln2)
         return x
+
+    def _add_ln(self, x, r, ln):
+        # fused residual+LayerNorm CDNA4 kernel on GPU/bf16
+        # (OKTOPK_NO_FUSED_LN=1 opts out); torch path otherwise
+        from ..ops.fused_ln import fused_add_layernorm, fused_ln_available
+
+        if fused_ln_available(x):
+            return fused_add_layernorm(x, r, ln)
+        return ln(x + r)
 
     def _mlp_act(self, x):
         # hand-written MFMA fused GEMM+bias+GELU on GPU/bf16 (reference

@@ -36,6 +36,11 @@ void launch_adam(float*, const float*, float*, float*, int64_t, float, float, fl
 void launch_sumsq(const float*, int64_t, double*, hipStream_t);
 void launch_linear_gelu(const void*, const void*, const float*, void*, void*, int,
                         int, int, int, hipStream_t);
+void launch_add_ln_fwd(const void*, const void*, const void*, const void*, void*,
+                       void*, float*, float*, int64_t, int, float, hipStream_t);
+void launch_add_ln_bwd(const void*, const void*, const float*, const float*,
+                       const void*, void*, float*, float*, int64_t, int,
+                       hipStream_t);
 }
 
 namespace {
@@ -359,6 +364,44 @@ static std::vector<torch::Tensor> linear_gelu(torch::Tensor x, torch::Tensor w,
     return {y};
 }
 
+static std::vector<torch::Tensor> add_ln_fwd(torch::Tensor x, torch::Tensor r,
+                                             torch::Tensor gamma, torch::Tensor beta,
+                                             double eps) {
+    TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+    TORCH_CHECK(r.sizes() == x.sizes() && r.is_contiguous());
+    int H = (int)x.size(-1);
+    TORCH_CHECK(H % 128 == 0 && H <= 2048, "H must be a multiple of 128, <= 2048");
+    const at::cuda::CUDAGuard guard(x.device());
+    int64_t rows = x.numel() / H;
+    auto y = torch::empty_like(x);
+    auto s = torch::empty_like(x);
+    auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+    auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+    launch_add_ln_fwd(x.data_ptr(), r.data_ptr(), gamma.contiguous().data_ptr(),
+                      beta.contiguous().data_ptr(), y.data_ptr(), s.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(), rows, H,
+                      (float)eps, cur_stream());
+    return {y, s, mean, rstd};
+}
+
+static std::vector<torch::Tensor> add_ln_bwd(torch::Tensor gy, torch::Tensor s,
+                                             torch::Tensor mean, torch::Tensor rstd,
+                                             torch::Tensor gamma) {
+    TORCH_CHECK(gy.is_cuda() && gy.scalar_type() == torch::kBFloat16);
+    int H = (int)gy.size(-1);
+    const at::cuda::CUDAGuard guard(gy.device());
+    int64_t rows = gy.numel() / H;
+    auto gyc = gy.contiguous();
+    auto gx = torch::empty_like(gyc);
+    auto dgamma = torch::zeros({H}, gy.options().dtype(torch::kFloat32));
+    auto dbeta = torch::zeros({H}, gy.options().dtype(torch::kFloat32));
+    launch_add_ln_bwd(gyc.data_ptr(), s.data_ptr(), mean.data_ptr<float>(),
+                      rstd.data_ptr<float>(), gamma.contiguous().data_ptr(),
+                      gx.data_ptr(), dgamma.data_ptr<float>(),
+                      dbeta.data_ptr<float>(), rows, H, cur_stream());
+    return {gx, dgamma, dbeta};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "oktopk_amd CDNA4 HIP kernels (gfx950)";
     m.def("count_gt", &count_gt, "count |t| > tau");
@@ -379,6 +422,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
     m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
+    m.def("add_ln_fwd", &add_ln_fwd, "fused y=LN(x+r) forward (bf16)");
+    m.def("add_ln_bwd", &add_ln_bwd, "fused add+LN backward (bf16, fp32 col sums)");
     m.def("linear_gelu", &linear_gelu,
           "fused y=gelu(x@W^T+b) bf16 MFMA kernel (optionally returns pre-act)");
 }
