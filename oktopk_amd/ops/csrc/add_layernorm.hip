@@ -92,6 +92,12 @@ __global__ void add_ln_fwd_kernel(const short* __restrict__ x,
 
 // bwd: gx = rstd * (gyh - mean(gyh) - h * mean(gyh*h)), h = (s-mean)*rstd,
 // gyh = gy*gamma; dgamma += gy*h, dbeta += gy (fp32 atomics)
+// Per-lane caches of the whole row would be runtime-indexed arrays ->
+// scratch (measured 272 B/lane, bench regression); instead each row is read
+// twice (pass 1 sums, pass 2 recomputes — L1/L2-resident the second time).
+// dgamma/dbeta accumulate into LDS column partials; ONE global atomic per
+// column per block (a per-element global atomicAdd was ~1.5M atomics per
+// call).  Grid is small + row-strided so flush contention stays low.
 __global__ void add_ln_bwd_kernel(const short* __restrict__ gy,
                                   const short* __restrict__ s,
                                   const float* __restrict__ mean_in,
@@ -101,46 +107,59 @@ __global__ void add_ln_bwd_kernel(const short* __restrict__ gy,
                                   float* __restrict__ dgamma,
                                   float* __restrict__ dbeta,
                                   int64_t rows, int H) {
-    int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
-    int64_t row = (int64_t)blockIdx.x * (LN_BLOCK / 64) + wave;
-    if (row >= rows) return;
-    const int npairs = H / 128;
-    const bf16x2_t* gyp = reinterpret_cast<const bf16x2_t*>(gy + row * H);
-    const bf16x2_t* sp = reinterpret_cast<const bf16x2_t*>(s + row * H);
-    const bf16x2_t* gp = reinterpret_cast<const bf16x2_t*>(gamma);
-    bf16x2_t* gxp = reinterpret_cast<bf16x2_t*>(gx + row * H);
-    float mean = mean_in[row], rstd = rstd_in[row];
+    __shared__ float col_g[2048];
+    __shared__ float col_b[2048];
+    for (int c = threadIdx.x; c < H; c += LN_BLOCK) {
+        col_g[c] = 0.f;
+        col_b[c] = 0.f;
+    }
+    __syncthreads();
 
-    float gyv[16][2], hv[16][2], gyhv[16][2];
-    float sum_gyh = 0.f, sum_gyh_h = 0.f;
-    for (int j = 0; j < npairs; ++j) {
-        int c = lane + 64 * j;
-        bf16x2_t g = gyp[c], sv = sp[c], gm = gp[c];
-        #pragma unroll
-        for (int q = 0; q < 2; ++q) {
-            float gf = b2f(g[q]);
-            float h = (b2f(sv[q]) - mean) * rstd;
-            float gyh = gf * b2f(gm[q]);
-            gyv[j][q] = gf;
-            hv[j][q] = h;
-            gyhv[j][q] = gyh;
-            sum_gyh += gyh;
-            sum_gyh_h += gyh * h;
+    int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int npairs = H / 128;
+    const int rpb = LN_BLOCK / 64;
+    const bf16x2_t* gp = reinterpret_cast<const bf16x2_t*>(gamma);
+    for (int64_t row = (int64_t)blockIdx.x * rpb + wave; row < rows;
+         row += (int64_t)gridDim.x * rpb) {
+        const bf16x2_t* gyp = reinterpret_cast<const bf16x2_t*>(gy + row * H);
+        const bf16x2_t* sp = reinterpret_cast<const bf16x2_t*>(s + row * H);
+        bf16x2_t* gxp = reinterpret_cast<bf16x2_t*>(gx + row * H);
+        float mean = mean_in[row], rstd = rstd_in[row];
+
+        float sum_gyh = 0.f, sum_gyh_h = 0.f;
+        for (int j = 0; j < npairs; ++j) {
+            int c = lane + 64 * j;
+            bf16x2_t g = gyp[c], sv = sp[c], gm = gp[c];
+            #pragma unroll
+            for (int q = 0; q < 2; ++q) {
+                float h = (b2f(sv[q]) - mean) * rstd;
+                float gyh = b2f(g[q]) * b2f(gm[q]);
+                sum_gyh += gyh;
+                sum_gyh_h += gyh * h;
+            }
+        }
+        float m1 = wave_sum(sum_gyh) / H;
+        float m2 = wave_sum(sum_gyh_h) / H;
+        for (int j = 0; j < npairs; ++j) {
+            int c = lane + 64 * j;
+            bf16x2_t g = gyp[c], sv = sp[c], gm = gp[c];
+            bf16x2_t o;
+            #pragma unroll
+            for (int q = 0; q < 2; ++q) {
+                float gf = b2f(g[q]);
+                float h = (b2f(sv[q]) - mean) * rstd;
+                float gyh = gf * b2f(gm[q]);
+                o[q] = f2b(rstd * (gyh - m1 - h * m2));
+                atomicAdd(&col_g[2 * c + q], gf * h);
+                atomicAdd(&col_b[2 * c + q], gf);
+            }
+            gxp[c] = o;
         }
     }
-    float m1 = wave_sum(sum_gyh) / H;
-    float m2 = wave_sum(sum_gyh_h) / H;
-    for (int j = 0; j < npairs; ++j) {
-        int c = lane + 64 * j;
-        bf16x2_t o;
-        #pragma unroll
-        for (int q = 0; q < 2; ++q) {
-            float v = rstd * (gyhv[j][q] - m1 - hv[j][q] * m2);
-            o[q] = f2b(v);
-            atomicAdd(&dgamma[2 * c + q], gyv[j][q] * hv[j][q]);
-            atomicAdd(&dbeta[2 * c + q], gyv[j][q]);
-        }
-        gxp[c] = o;
+    __syncthreads();
+    for (int c = threadIdx.x; c < H; c += LN_BLOCK) {
+        atomicAdd(&dgamma[c], col_g[c]);
+        atomicAdd(&dbeta[c], col_b[c]);
     }
 }
 
@@ -162,6 +181,7 @@ extern "C" void launch_add_ln_bwd(const void* gy, const void* s, const float* me
                                   hipStream_t stream) {
     int rows_per_block = LN_BLOCK / 64;
     int64_t grid = (rows + rows_per_block - 1) / rows_per_block;
+    if (grid > 128) grid = 128;  // row-strided; bounds the column-flush atomics
     hipLaunchKernelGGL(add_ln_bwd_kernel, dim3((uint32_t)grid), dim3(LN_BLOCK), 0,
                        stream, (const short*)gy, (const short*)s, mean, rstd,
                        (const short*)gamma, (short*)gx, dgamma, dbeta, rows, H);
