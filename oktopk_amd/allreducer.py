@@ -28,14 +28,14 @@ from __future__ import annotations
 import logging
 import math
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 import torch
 
 from . import ops
 from .comm import Comm
-from .config import EngineConfig, OkTopkConfig
+from .config import EngineConfig
 
 logger = logging.getLogger("oktopk_amd")
 
@@ -185,6 +185,17 @@ class AllReducer:
             eps_topk, eps_den = eps_ref
             num = ops.l2norm(out - eps_topk)
             self.eps_log.append((st.counter, num / eps_den))
+
+        # per-phase timing table (reference prints every 50 iterations,
+        # VGG/allreducer.py:379-439)
+        if (
+            self.cfg.profiling
+            and self.comm.rank == 0
+            and st.counter > 0
+            and st.counter % self.cfg.profiling_interval == 0
+        ):
+            logger.info("timing[%s] iter %d: %s", name, st.counter,
+                        {k: round(v, 4) for k, v in self.timers.get(name, {}).items()})
 
         st.counter += 1
         return tensor

@@ -6,8 +6,6 @@ offline — torch.stft computes the identical feature.
 """
 from __future__ import annotations
 
-import math
-import os
 from typing import List, Optional, Tuple
 
 import torch

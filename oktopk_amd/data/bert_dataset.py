@@ -12,7 +12,7 @@ documents); no network needed.
 from __future__ import annotations
 
 import random
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, Sequence
 
 import torch
 from torch.utils.data import Dataset

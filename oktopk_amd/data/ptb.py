@@ -5,7 +5,7 @@ from __future__ import annotations
 
 import collections
 import os
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List
 
 import torch
 

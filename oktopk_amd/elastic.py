@@ -14,7 +14,6 @@ checkpoint-and-requeue is in `install_preemption_handler`.
 """
 from __future__ import annotations
 
-import os
 import signal
 from typing import Callable, Iterable, List, Optional
 

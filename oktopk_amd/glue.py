@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import csv
 import math
-from typing import Dict, List, Sequence, Tuple
+from typing import Dict, List, Sequence
 
 import numpy as np
 

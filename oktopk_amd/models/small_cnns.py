@@ -4,8 +4,6 @@
 VGG/models/__init__.py:16-27)."""
 from __future__ import annotations
 
-import math
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

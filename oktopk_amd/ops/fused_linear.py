@@ -10,7 +10,6 @@ Parity: the reference's LinearActivation fused-gelu module
 """
 from __future__ import annotations
 
-import math
 import os
 
 import torch

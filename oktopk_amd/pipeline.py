@@ -26,7 +26,7 @@ thread-per-tensor-per-peer because gloo CPU sends block
 from __future__ import annotations
 
 from collections import deque
-from typing import Callable, Dict, List, Optional, Sequence, Tuple
+from typing import List, Optional, Sequence, Tuple
 
 import torch
 import torch.distributed as dist

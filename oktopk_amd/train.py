@@ -13,11 +13,8 @@ Launch (mirrors the reference's srun/mpirun scripts — see launch/):
 from __future__ import annotations
 
 import argparse
-import json
 import os
 import time
-
-import torch
 
 from .comm import init_from_env
 from .config import EngineConfig

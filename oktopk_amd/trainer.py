@@ -8,7 +8,6 @@ shapes of the reference configs.
 """
 from __future__ import annotations
 
-import time
 from typing import Optional
 
 import torch

@@ -6,8 +6,6 @@ of the reference's Cray Aries numbers.
 """
 from __future__ import annotations
 
-import math
-
 # Per-link xGMI: ~153 GB/s peak, assume ~80% achievable; latency per hop.
 XGMI_LINK_BW = 153e9 * 0.8  # B/s
 XGMI_ALPHA = 8e-6  # s per message

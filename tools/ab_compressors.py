@@ -19,8 +19,8 @@ from oktopk_amd.comm import init_from_env
 from oktopk_amd.config import EngineConfig
 from oktopk_amd.trainer import Trainer
 
-COMPRESSORS = ["dense", "oktopk", "topkA", "topkAopt", "topkSA", "gtopk",
-               "gaussiank", "gaussiankSA"]
+COMPRESSORS = ["dense", "oktopk", "topkA", "topkA2", "topkAopt", "topkSA",
+               "gtopk", "gaussiank", "gaussiankSA"]
 
 
 def main():
