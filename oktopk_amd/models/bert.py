@@ -69,7 +69,16 @@ class BertSelfAttention(nn.Module):
 
     def forward(self, x, attn_mask=None):
         b, s, h = x.shape
-        qkv = self.qkv(x).view(b, s, 3, self.nh, self.hd).permute(2, 0, 3, 1, 4)
+        qkv_flat = self.qkv(x)
+        # hand-written CDNA4 fused attention (QK^T+softmax+dropout+PV in one
+        # MFMA kernel) on GPU/bf16 at the reference shape; eager otherwise
+        from ..ops.fused_attn import fused_attention, fused_attn_available
+
+        if fused_attn_available(qkv_flat, self.nh, s, self.attn_drop.p):
+            ctx = fused_attention(qkv_flat, attn_mask, self.nh,
+                                  self.attn_drop.p, self.training)
+            return self.out(ctx)
+        qkv = qkv_flat.view(b, s, 3, self.nh, self.hd).permute(2, 0, 3, 1, 4)
         q, k, v = qkv[0], qkv[1], qkv[2]  # (b, nh, s, hd)
         scores = torch.matmul(q, k.transpose(-1, -2)) / math.sqrt(self.hd)
         if attn_mask is not None:

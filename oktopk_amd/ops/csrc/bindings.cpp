@@ -4,6 +4,7 @@
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+#include <ATen/cuda/CUDAGeneratorImpl.h>
 #include <c10/cuda/CUDAGuard.h>
 
 #include <cstdint>
@@ -41,6 +42,10 @@ void launch_add_ln_fwd(const void*, const void*, const void*, const void*, void*
 void launch_add_ln_bwd(const void*, const void*, const float*, const float*,
                        const void*, void*, float*, float*, int64_t, int,
                        hipStream_t);
+void launch_attn_fwd(const void*, const void*, void*, void*, void*, int, int,
+                     float, float, unsigned long long, unsigned long long,
+                     const void*, const void*, unsigned int, int, int,
+                     hipStream_t);
 }
 
 namespace {
@@ -402,6 +407,68 @@ static std::vector<torch::Tensor> add_ln_bwd(torch::Tensor gy, torch::Tensor s,
     return {gx, dgamma, dbeta};
 }
 
+static std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, torch::Tensor mask,
+                                           int64_t num_heads, double dropout_p,
+                                           bool training, bool want_saved) {
+    TORCH_CHECK(qkv.is_cuda() && qkv.scalar_type() == torch::kBFloat16 &&
+                qkv.is_contiguous());
+    TORCH_CHECK(qkv.dim() == 3, "qkv must be [b, s, 3*h]");
+    int64_t B = qkv.size(0), S = qkv.size(1);
+    int64_t H3 = qkv.size(2);
+    int64_t H = H3 / 3;
+    int64_t HD = H / num_heads;
+    TORCH_CHECK(S == 128 && HD == 64, "fused attention specialised for s=128, hd=64");
+    const at::cuda::CUDAGuard guard(qkv.device());
+    auto out = torch::empty({B, S, H}, qkv.options());
+    torch::Tensor p_save, a_save;
+    void* pptr = nullptr;
+    void* aptr = nullptr;
+    if (want_saved) {
+        p_save = torch::empty({B * num_heads, S, S}, qkv.options());
+        a_save = torch::empty({B * num_heads, S, S}, qkv.options());
+        pptr = p_save.data_ptr();
+        aptr = a_save.data_ptr();
+    }
+    const void* mptr = nullptr;
+    torch::Tensor mask_c;
+    if (mask.defined() && mask.numel()) {
+        mask_c = mask.reshape({B, S}).to(torch::kBFloat16).contiguous();
+        mptr = mask_c.data_ptr();
+    }
+    int apply_dropout = (training && dropout_p > 0.0) ? 1 : 0;
+    unsigned long long seed = 0, offset = 0;
+    const void* seed_ptr = nullptr;
+    const void* offset_ptr = nullptr;
+    unsigned int intragraph = 0;
+    int captured = 0;
+    if (apply_dropout) {
+        auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+            c10::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+        at::PhiloxCudaState state;
+        {
+            std::lock_guard<std::mutex> lock(gen->mutex_);
+            // one philox4x32 block covers 4 elements
+            state = gen->philox_cuda_state((B * num_heads * S * S + 3) / 4 + 1);
+        }
+        if (state.captured_) {
+            captured = 1;
+            seed_ptr = state.seed_.ptr;
+            offset_ptr = state.offset_.ptr;
+            intragraph = state.offset_intragraph_;
+        } else {
+            seed = state.seed_.val;
+            offset = state.offset_.val;
+        }
+    }
+    float scale = 1.0f / std::sqrt((float)HD);
+    launch_attn_fwd(qkv.data_ptr(), mptr, out.data_ptr(), pptr, aptr, (int)B,
+                    (int)num_heads, scale, (float)(1.0 - dropout_p), seed, offset,
+                    seed_ptr, offset_ptr, intragraph, captured, apply_dropout,
+                    cur_stream());
+    if (want_saved) return {out, p_save, a_save};
+    return {out};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "oktopk_amd CDNA4 HIP kernels (gfx950)";
     m.def("count_gt", &count_gt, "count |t| > tau");
@@ -422,6 +489,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
     m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
+    m.def("attn_fwd", &attn_fwd,
+          "fused self-attention forward: softmax(QK^T*scale+mask) dropout @ V "
+          "(bf16 MFMA, seq=128/hd=64; returns ctx [+P, A for backward])");
     m.def("add_ln_fwd", &add_ln_fwd, "fused y=LN(x+r) forward (bf16)");
     m.def("add_ln_bwd", &add_ln_bwd, "fused add+LN backward (bf16, fp32 col sums)");
     m.def("linear_gelu", &linear_gelu,

@@ -1,0 +1,96 @@
+"""Fused self-attention with autograd.
+
+Forward: one CDNA4 kernel (ops/csrc/attention.hip) — QK^T, masked softmax,
+capture-safe philox dropout, PV — consuming the qkv projection buffer
+directly and emitting the context in (b, s, h) layout (the permute/copy
+chain of the eager path disappears).
+
+Backward (hand-written, torch bmms): with P (post-softmax) and A (post-
+dropout) saved by the forward,
+    dV = A^T gO,   dA = gO V^T,   dP = dA * mask/keep   (mask = [A != 0]),
+    dS = P * (dP - rowsum(dP * P)),  dQ = dS K * scale,  dK = dS^T Q * scale.
+The [A != 0] mask reconstruction is exact except at entries where P itself
+is bf16-zero — which carry no gradient anyway.
+"""
+from __future__ import annotations
+
+import math
+import os
+
+import torch
+
+
+def fused_attn_available(x: torch.Tensor, num_heads: int, seq: int, dropout_p: float) -> bool:
+    if os.environ.get("OKTOPK_NO_FUSED_ATTN", "0") == "1":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    hd = x.shape[-1] // (3 * num_heads)
+    if seq != 128 or hd != 64:
+        return False
+    from . import hip_available
+
+    return hip_available()
+
+
+class _FusedAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, mask, num_heads, dropout_p, training):
+        from oktopk_amd import _hip_ops
+
+        need_grad = qkv.requires_grad
+        outs = _hip_ops.attn_fwd(
+            qkv,
+            mask if mask is not None else torch.empty(0, device=qkv.device),
+            num_heads,
+            float(dropout_p),
+            bool(training),
+            need_grad,
+        )
+        if need_grad:
+            out, p, a = outs
+            ctx.save_for_backward(qkv, p, a)
+            ctx.meta = (num_heads, dropout_p, training)
+        else:
+            out = outs[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        qkv, p, a = ctx.saved_tensors
+        num_heads, dropout_p, training = ctx.meta
+        b, s, h3 = qkv.shape
+        h = h3 // 3
+        hd = h // num_heads
+        scale = 1.0 / math.sqrt(hd)
+        # views of the projection buffer: (b*nh, s, hd)
+        qkv5 = qkv.view(b, s, 3, num_heads, hd)
+        q = qkv5[:, :, 0].permute(0, 2, 1, 3).reshape(b * num_heads, s, hd)
+        k = qkv5[:, :, 1].permute(0, 2, 1, 3).reshape(b * num_heads, s, hd)
+        v = qkv5[:, :, 2].permute(0, 2, 1, 3).reshape(b * num_heads, s, hd)
+        go_h = go.view(b, s, num_heads, hd).permute(0, 2, 1, 3).reshape(
+            b * num_heads, s, hd
+        )
+        dv = torch.bmm(a.transpose(1, 2), go_h)
+        da = torch.bmm(go_h, v.transpose(1, 2))
+        if training and dropout_p > 0:
+            keep = 1.0 - dropout_p
+            dp = da * (a != 0).to(da.dtype) / keep
+        else:
+            dp = da
+        pf = p.float()
+        dpf = dp.float()
+        ds = (pf * (dpf - (dpf * pf).sum(dim=-1, keepdim=True))).to(qkv.dtype)
+        dq = torch.bmm(ds, k) * scale
+        dk = torch.bmm(ds.transpose(1, 2), q) * scale
+        dqkv = torch.empty_like(qkv).view(b, s, 3, num_heads, hd)
+        dqkv[:, :, 0] = dq.view(b, num_heads, s, hd).permute(0, 2, 1, 3)
+        dqkv[:, :, 1] = dk.view(b, num_heads, s, hd).permute(0, 2, 1, 3)
+        dqkv[:, :, 2] = dv.view(b, num_heads, s, hd).permute(0, 2, 1, 3)
+        return dqkv.view(b, s, h3), None, None, None, None
+
+
+def fused_attention(qkv, mask, num_heads, dropout_p, training):
+    """ctx = dropout(softmax(QK^T*scale + mask)) @ V from the fused qkv
+    projection buffer [b, s, 3h]; returns [b, s, h]."""
+    return _FusedAttention.apply(qkv, mask, num_heads, dropout_p, training)

@@ -21,6 +21,7 @@ setup(
                 "oktopk_amd/ops/csrc/kernels.hip",
                 "oktopk_amd/ops/csrc/linear_gelu.hip",
                 "oktopk_amd/ops/csrc/add_layernorm.hip",
+                "oktopk_amd/ops/csrc/attention.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

@@ -305,3 +305,97 @@ def test_fused_add_layernorm_autograd_module():
     y.sum().backward()
     assert torch.equal(x.grad, r.grad)  # add passes gradient through
     assert ln.weight.grad is not None and torch.isfinite(ln.weight.grad).all()
+
+
+def _attn_ref(qkv, mask, nh, dropout_p=0.0):
+    b, s, h3 = qkv.shape
+    h = h3 // 3
+    hd = h // nh
+    q, k, v = (
+        qkv.view(b, s, 3, nh, hd).permute(2, 0, 3, 1, 4).float().unbind(0)
+    )
+    scores = q @ k.transpose(-1, -2) / (hd ** 0.5)
+    if mask is not None:
+        scores = scores + mask.view(b, 1, 1, s).float()
+    p = torch.softmax(scores, dim=-1)
+    ctx = p @ v
+    return ctx.transpose(1, 2).reshape(b, s, h), p
+
+
+def test_fused_attention_forward_no_dropout():
+    from oktopk_amd import _hip_ops
+
+    torch.manual_seed(5)
+    b, s, nh, hd = 3, 128, 4, 64
+    qkv = (torch.randn(b, s, 3 * nh * hd) * 0.5).bfloat16().cuda()
+    mask = torch.zeros(b, s).bfloat16().cuda()
+    mask[:, 100:] = -10000.0  # padded keys
+    out, p, a = _hip_ops.attn_fwd(qkv, mask, nh, 0.0, True, True)
+    ref, p_ref = _attn_ref(qkv, mask, nh)
+    assert torch.allclose(out.float(), ref, atol=0.03, rtol=0.02), (
+        (out.float() - ref).abs().max()
+    )
+    assert torch.allclose(
+        p.view(b, nh, s, s).float(), p_ref, atol=0.01
+    ), (p.view(b, nh, s, s).float() - p_ref).abs().max()
+    assert torch.equal(p, a)  # no dropout -> identical
+
+
+def test_fused_attention_dropout_stats():
+    from oktopk_amd import _hip_ops
+
+    torch.manual_seed(6)
+    b, s, nh = 4, 128, 8
+    qkv = (torch.randn(b, s, 3 * nh * 64) * 0.5).bfloat16().cuda()
+    drop_p = 0.25
+    out, p, a = _hip_ops.attn_fwd(qkv, torch.empty(0).cuda(), nh, drop_p, True, True)
+    dropped = (a == 0) & (p != 0)
+    frac = dropped.float().mean().item()
+    assert abs(frac - drop_p) < 0.01, frac
+    # kept entries are scaled by 1/keep
+    kept = a != 0
+    ratio = (a[kept].float() / p[kept].float()).mean().item()
+    assert abs(ratio - 1 / (1 - drop_p)) < 0.02, ratio
+    # two calls give different masks (generator offset advances)
+    _, _, a2 = _hip_ops.attn_fwd(qkv, torch.empty(0).cuda(), nh, drop_p, True, True)
+    assert not torch.equal(a, a2)
+
+
+def test_fused_attention_autograd():
+    from oktopk_amd.ops.fused_attn import fused_attention
+
+    torch.manual_seed(7)
+    b, s, nh, hd = 2, 128, 4, 64
+    qkv = (torch.randn(b, s, 3 * nh * hd) * 0.5).bfloat16().cuda().requires_grad_(True)
+    mask = torch.zeros(b, s).bfloat16().cuda()
+    out = fused_attention(qkv, mask, nh, 0.0, True)
+    gy = torch.randn_like(out)
+    out.backward(gy)
+
+    q2 = qkv.detach().float().requires_grad_(True)
+    ctxr, _ = _attn_ref(q2, mask, nh)
+    ctxr.backward(gy.float())
+    assert torch.allclose(qkv.grad.float(), q2.grad, atol=0.1, rtol=0.05), (
+        (qkv.grad.float() - q2.grad).abs().max()
+    )
+
+
+def test_bert_layer_fused_attention_matches_eager():
+    import os
+    from oktopk_amd.models.bert import BertConfig, BertLayer
+
+    torch.manual_seed(8)
+    cfg = BertConfig(hidden_size=768, num_attention_heads=12,
+                     intermediate_size=1024, hidden_dropout_prob=0.0,
+                     attention_probs_dropout_prob=0.0)
+    lyr = BertLayer(cfg).bfloat16().cuda()
+    x = torch.randn(2, 128, 768).bfloat16().cuda()
+    out_fused = lyr(x)
+    os.environ["OKTOPK_NO_FUSED_ATTN"] = "1"
+    try:
+        out_eager = lyr(x)
+    finally:
+        os.environ["OKTOPK_NO_FUSED_ATTN"] = "0"
+    assert torch.allclose(out_fused.float(), out_eager.float(), atol=0.12), (
+        (out_fused.float() - out_eager.float()).abs().max()
+    )
