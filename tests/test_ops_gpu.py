@@ -390,12 +390,12 @@ def test_bert_layer_fused_attention_matches_eager():
                      attention_probs_dropout_prob=0.0)
     lyr = BertLayer(cfg).bfloat16().cuda()
     x = torch.randn(2, 128, 768).bfloat16().cuda()
-    out_fused = lyr(x)
-    os.environ["OKTOPK_NO_FUSED_ATTN"] = "1"
+    os.environ["OKTOPK_FUSED_ATTN"] = "1"
     try:
-        out_eager = lyr(x)
+        out_fused = lyr(x)
     finally:
-        os.environ["OKTOPK_NO_FUSED_ATTN"] = "0"
+        os.environ["OKTOPK_FUSED_ATTN"] = "0"
+    out_eager = lyr(x)
     assert torch.allclose(out_fused.float(), out_eager.float(), atol=0.12), (
         (out_fused.float() - out_eager.float()).abs().max()
     )
