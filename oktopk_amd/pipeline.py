@@ -343,6 +343,83 @@ class PipelineRuntime:
         return total_loss / max(n, 1)
 
 
+    # -- continuous PipeDream (per-microbatch updates + weight stashing) --
+    def run_pipedream(self, microbatches: Sequence[dict],
+                      stash: "OptimizerWithWeightStashing") -> float:
+        """The reference's actual PipeDream mode (BERT/runtime.py:902
+        run_training_loop + OptimizerWithWeightStashing): steady-state 1F1B
+        with an optimizer STEP after every microbatch's backward; each
+        forward uses the oldest stashed weight version so forward/backward
+        of one microbatch see consistent weights despite in-flight updates.
+        """
+        if self.num_stages == 1:
+            total = 0.0
+            for mb in microbatches:
+                stash.load_old_params()
+                out = self.stage(**mb)
+                stash.zero_grad()
+                out.backward()
+                stash.step()
+                total += float(out.detach().float())
+            return total / max(len(microbatches), 1)
+        # (multi-stage below)
+
+        n = len(microbatches)
+        warmup = min(self.num_stages - 1 - self.stage_id, n)
+        fwd_q: deque = deque()
+        total_loss = 0.0
+        fwd_i = 0
+
+        def do_forward(mb):
+            nonlocal fwd_i
+            version = stash.oldest_version()
+            stash.load_version(version)
+            if self.is_first:
+                out = self.stage(**mb)
+                self.comm.send(out.detach(), dst=self.next_rank)
+                fwd_q.append((None, out, version))
+            else:
+                hidden = self.comm.recv(
+                    src=self.prev_rank, dtype=self.act_dtype, device=self.device
+                )
+                hidden.requires_grad_(True)
+                out = self.stage(hidden, **mb)
+                if not self.is_last:
+                    self.comm.send(out.detach(), dst=self.next_rank)
+                fwd_q.append((hidden, out, version))
+            fwd_i += 1
+
+        def do_backward_and_step():
+            nonlocal total_loss
+            hidden_in, out, version = fwd_q.popleft()
+            # backward runs against the SAME weight version its forward used
+            # (reference load_backward_params,
+            # optimizer_with_stashing_and_aggregation.py:117-147)
+            stash.load_version(version)
+            stash.zero_grad()
+            if self.is_last:
+                out.backward()
+                total_loss += float(out.detach().float())
+            else:
+                grad_out = self.comm.recv(
+                    src=self.next_rank, dtype=self.act_dtype, device=self.device
+                )
+                out.backward(grad_out)
+            if not self.is_first:
+                self.comm.send(hidden_in.grad, dst=self.prev_rank)
+            stash.step()
+
+        for _ in range(warmup):
+            do_forward(microbatches[fwd_i])
+        while fwd_i < n:
+            do_forward(microbatches[fwd_i])
+            do_backward_and_step()
+        while fwd_q:
+            do_backward_and_step()
+        self.comm.flush()
+        return total_loss / max(n, 1)
+
+
 class RuntimeStats:
     """fwd/bwd compute+comm counters (reference BERT/runtime_utilities.py:4-28)."""
 
@@ -369,47 +446,65 @@ class RuntimeStats:
 # ---------------------------------------------------------------------------
 
 class OptimizerWithWeightStashing:
-    """Keeps `num_versions` cloned weight versions; forward uses the oldest,
-    the update applies to the latest (PipeDream weight semantics)."""
+    """PipeDream weight stashing (reference BERT/optimizer.py:19,
+    optimizer_with_stashing.py): keeps `num_versions` weight versions;
+    each microbatch's forward AND backward run under the version that was
+    newest at its forward time, while updates apply to a master copy.
+
+    Implementation note for modern torch: the base optimizer steps on MASTER
+    clones (never part of an autograd graph), and live module weights are
+    only written through `.data.copy_` — a plain in-place `optimizer.step()`
+    on live params bumps their autograd version counters and poisons every
+    in-flight microbatch's backward."""
 
     def __init__(self, modules: Sequence[nn.Module], base_optimizer, num_versions: int):
         self.modules = list(modules)
         self.base = base_optimizer
         self.num_versions = max(1, num_versions)
+        # live params in a stable order + master clones
+        self.live = [p for m in self.modules for p in m.parameters()]
+        self.masters = [p.detach().clone() for p in self.live]
+        # re-point the base optimizer at the masters (positionally)
+        it = iter(self.masters)
+        for group in self.base.param_groups:
+            group["params"] = [next(it) for _ in group["params"]]
         self.queue: deque = deque()
-        self.latest_version = 0
         for _ in range(self.num_versions):
-            self.queue.append(self._clone())
+            self.queue.append([mp.detach().clone() for mp in self.masters])
+        self.latest_version = 0
 
-    def _clone(self):
-        return [
-            {k: v.detach().clone() for k, v in m.state_dict().items()}
-            for m in self.modules
-        ]
+    def _load(self, version) -> None:
+        with torch.no_grad():
+            for p, v in zip(self.live, version):
+                p.data.copy_(v)
 
-    def _load(self, versions):
-        for m, sd in zip(self.modules, versions):
-            m.load_state_dict(sd, strict=True)
+    def oldest_version(self):
+        return self.queue[0]
+
+    def load_version(self, version) -> None:
+        self._load(version)
 
     def load_old_params(self):
-        if self.num_versions > 1:
-            self._load(self.queue[0])
+        self._load(self.queue[0])
 
     def load_new_params(self):
-        if self.num_versions > 1:
-            self._load(self.queue[-1])
+        self._load(self.queue[-1])
 
     def zero_grad(self):
-        self.base.zero_grad()
+        for p in self.live:
+            if p.grad is not None:
+                p.grad = None
 
     def step(self):
-        # gradients were computed against the OLD weights; apply to latest
-        self.load_new_params()
+        # gradients were computed on a stashed version of the LIVE params;
+        # apply them to the masters (always the newest weights)
+        for mp, p in zip(self.masters, self.live):
+            mp.grad = None if p.grad is None else p.grad.detach()
         self.base.step()
         self.latest_version += 1
-        if self.num_versions > 1:
+        self.queue.append([mp.detach().clone() for mp in self.masters])
+        while len(self.queue) > self.num_versions:
             self.queue.popleft()
-            self.queue.append(self._clone())
 
 
 # ---------------------------------------------------------------------------
