@@ -309,6 +309,9 @@ class AllReducer:
             st.tau_local = taus[chosen]
         tau = st.tau_local
 
+        if it in self.cfg.profiling_grad_iters and self.comm.rank == 0:
+            self._dump_grad_stats(name, it, t, tau)
+
         # --- 2. balanced region repartition (reuses the selection) -------
         if st.boundaries is None:
             st.boundaries = self._uniform_boundaries(n)
@@ -743,6 +746,33 @@ class AllReducer:
     # ------------------------------------------------------------------
     def timing_table(self, name: str) -> Dict[str, float]:
         return dict(self.timers.get(name, {}))
+
+    def _dump_grad_stats(self, name: str, it: int, t: torch.Tensor, tau: float) -> None:
+        """Gradient-distribution research dump (reference PROFILING_GRAD
+        np.save of local grads + thresholds, VGG/allreducer.py:608-623) —
+        saves |grad| quantiles + the active threshold instead of the raw
+        14-110M-element tensor."""
+        import json as _json
+        import os as _os
+
+        d = self.cfg.profiling_grad_dir
+        _os.makedirs(d, exist_ok=True)
+        a = t.detach().abs().float()
+        qs = torch.quantile(
+            a[:: max(1, a.numel() // 100_000)],
+            torch.linspace(0, 1, 21, device=a.device),
+        )
+        with open(_os.path.join(d, f"{name}_iter{it}.json"), "w") as f:
+            _json.dump(
+                {
+                    "iter": it,
+                    "tau_local": tau,
+                    "tau_global": self.states[name].tau_global,
+                    "abs_quantiles": [float(x) for x in qs],
+                    "numel": t.numel(),
+                },
+                f,
+            )
 
 
 def _gaussian_threshold(t: torch.Tensor, density: float) -> float:

@@ -75,6 +75,11 @@ class EngineConfig:
     # Per-phase timing table cadence (reference prints every 50 iterations).
     profiling: bool = False
     profiling_interval: int = 50
+    # gradient-distribution research dumps (reference PROFILING_GRAD,
+    # VGG/allreducer.py:608-623): save |grad| histograms + thresholds for
+    # the named iterations into profiling_grad_dir
+    profiling_grad_iters: tuple = ()
+    profiling_grad_dir: str = "grad_dumps"
 
     @classmethod
     def preset(cls, name: str, **overrides) -> "EngineConfig":

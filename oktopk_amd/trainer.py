@@ -82,6 +82,30 @@ class SyntheticBatches:
         for k, v in list(self.__dict__.items()):
             if isinstance(v, torch.Tensor):
                 setattr(self, k, v.to(device))
+        self._roll = 0
+
+    def randomize_(self) -> None:
+        """Re-roll batch CONTENT in place (buffers keep identity, so a
+        captured hipGraph that reads them stays valid)."""
+        self._roll += 1
+        g = torch.Generator().manual_seed(99991 * self._roll + 7)
+        if self.family in ("cifar", "imagenet"):
+            self.x.copy_(torch.randn(self.x.shape, generator=g).to(self.x.device))
+            self.y.copy_(torch.randint(0, int(self.y.max().clamp(min=9)) + 1,
+                                       self.y.shape, generator=g).to(self.y.device))
+        elif self.family == "ptb":
+            self.tokens.copy_(torch.randint(0, 10000, self.tokens.shape, generator=g).to(self.tokens.device))
+            self.targets.copy_(torch.randint(0, 10000, self.targets.shape, generator=g).to(self.targets.device))
+        elif self.family == "an4":
+            self.x.copy_(torch.randn(self.x.shape, generator=g).to(self.x.device))
+        elif self.family == "bert":
+            dev = self.input_ids.device
+            self.input_ids.copy_(torch.randint(0, self.vocab, self.input_ids.shape, generator=g).to(dev))
+            labels = torch.full(self.mlm_labels.shape, -1, dtype=torch.long)
+            mask_pos = torch.rand(*self.mlm_labels.shape, generator=g) < 0.15
+            labels[mask_pos] = torch.randint(0, self.vocab, (int(mask_pos.sum()),), generator=g)
+            self.mlm_labels.copy_(labels.to(dev))
+            self.nsp.copy_(torch.randint(0, 2, self.nsp.shape, generator=g).to(dev))
 
 
 class Trainer:

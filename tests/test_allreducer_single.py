@@ -140,3 +140,34 @@ def test_sgd_path_grad_clip():
     delta = (model.weight.detach() - p0).norm()
     # lr=1, clipped grad norm <= 0.1 -> total update norm <= ~0.1
     assert delta <= 0.11, delta
+
+
+def test_profiling_grad_dump(tmp_path):
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import json, os
+
+    cfg = EngineConfig(compressor="oktopk", density=0.05,
+                       profiling_grad_iters=(1,),
+                       profiling_grad_dir=str(tmp_path),
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(None), cfg)
+    for it in range(2):
+        eng.run("w", torch.randn(2048, generator=torch.Generator().manual_seed(it)))
+    files = os.listdir(tmp_path)
+    assert files, "no dump written"
+    d = json.load(open(os.path.join(tmp_path, files[0])))
+    assert d["iter"] == 1 and len(d["abs_quantiles"]) == 21
+
+
+def test_randomize_batches_in_place():
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.trainer import Trainer
+
+    tr = Trainer("caffe_cifar", batch_size=2, cfg=EngineConfig(compressor="dense"),
+                 dtype="fp32")
+    ptr = tr.batches.x.data_ptr()
+    before = tr.batches.x.clone()
+    tr.batches.randomize_()
+    assert tr.batches.x.data_ptr() == ptr  # in-place (graph-safe)
+    assert not torch.allclose(tr.batches.x, before)

@@ -23,6 +23,7 @@ def run(compressor, steps, profiling_norm=False, dtype="bf16"):
     tr = Trainer("bert_base", batch_size=8, seq_len=128, cfg=cfg, dtype=dtype)
     losses = []
     for i in range(steps):
+        tr.batches.randomize_()
         losses.append(tr.step())
     eps = []
     red = getattr(tr.opt, "reducer", None)
