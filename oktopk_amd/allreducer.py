@@ -150,7 +150,12 @@ class AllReducer:
 
         eps_ref = None
         if self.cfg.profiling_norm and comp not in ("none", "dense"):
-            dense_in = t + st.residual
+            # with the fused grad_src path, t is stale until the EF restore
+            # inside the compressor — the oracle must read the REAL grads
+            if grad_src is not None:
+                dense_in = grad_src.reshape(-1).to(t.dtype) + st.residual
+            else:
+                dense_in = t + st.residual
             eps_ref = self._dense_value(dense_in)
             # reference EPS compares against the GLOBAL TOP-K of the dense
             # mean (VGG/allreducer.py:601-606): zero everything below the

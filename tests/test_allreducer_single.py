@@ -171,3 +171,20 @@ def test_randomize_batches_in_place():
     tr.batches.randomize_()
     assert tr.batches.x.data_ptr() == ptr  # in-place (graph-safe)
     assert not torch.allclose(tr.batches.x, before)
+
+
+def test_eps_oracle_with_grad_src():
+    """The EPS oracle must read grad_src (not the stale fp32 buffer) when
+    the fused-EF path is active; at iteration 0 world-1 the oktopk result IS
+    the dense top-k, so EPS ~ 0."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    cfg = EngineConfig(compressor="oktopk", density=0.05, profiling_norm=True,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(None), cfg)
+    g = torch.randn(4096, generator=torch.Generator().manual_seed(0))
+    t = torch.zeros(4096)  # stale buffer (simulates flat_grad before upcast)
+    eng.run("w", t, grad_src=g.to(torch.bfloat16))
+    eps = eng.eps_log[0][1]
+    assert eps < 0.35, eps  # bf16 wire rounding + strict-> tie loss only
