@@ -653,12 +653,14 @@ class AllReducer:
         ) if P > 1 else gidx.numel()
         result = t
         if P > 1 and nnz_total > (2 * n) // 3:
-            # dense fallback: allgather raw regions (VGG/allreducer.py:1318-1357)
+            # dense fallback: allgather raw regions (VGG/allreducer.py:1318-1357).
+            # Pad to the LARGEST region: with non-divisible n the last
+            # uniform region holds n - (P-1)*(n//P) > ceil(n/P) elements.
+            step = max(int(bounds[i + 1] - bounds[i]) for i in range(P))
             buf = comm.allgather_eq(
-                torch.nn.functional.pad(reduced, (0, (n + P - 1) // P - reduced.numel()))
+                torch.nn.functional.pad(reduced, (0, step - reduced.numel()))
             )
             result.zero_()
-            step = (n + P - 1) // P
             for i in range(P):
                 blo, bhi = int(bounds[i]), int(bounds[i + 1])
                 result[blo:bhi] = buf[i * step : i * step + (bhi - blo)]

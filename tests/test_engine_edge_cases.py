@@ -97,3 +97,31 @@ def _tiny_dist(rank):
 
 def test_tiny_tensor_world2_repartition_every_step():
     run_dist(_tiny_dist, 2)
+
+
+def _dense_fallback_odd_n(rank):
+    """topkSA dense-region fallback with n not divisible by P (the last
+    uniform region is LARGER than ceil(n/P)) must reproduce the dense mean."""
+    import torch.distributed as dist
+
+    eng = AllReducer(Comm(dist.group.WORLD),
+                     EngineConfig(compressor="topkSA", density=1.0,
+                                  oktopk=OkTopkConfig(dense_warmup_iters=0)))
+    n = 1003  # 1003 = 2*501 + 1 -> regions [501, 502]
+    t = torch.randn(n, generator=torch.Generator().manual_seed(rank))
+    ref = sum(
+        torch.randn(n, generator=torch.Generator().manual_seed(r)) for r in range(2)
+    ) / 2
+    out = eng.run("w", t.clone())
+    # density 1.0 selects everything except each rank's strict-> minimum
+    # element (which stays in its residual, faithful EF semantics) -> the
+    # fallback must reproduce the dense mean everywhere but <= P coords
+    bad = ((out - ref).abs() > 1e-5).sum().item()
+    assert bad <= 2, bad
+    # and the region-stitching must be structurally right: large errors
+    # (region offset bugs) would corrupt whole blocks
+    assert (out - ref).abs().max() < 1.0
+
+
+def test_topkSA_dense_fallback_odd_n_world2():
+    run_dist(_dense_fallback_odd_n, 2)
