@@ -372,9 +372,20 @@ class FlatBertAdam:
             self.flat_param_model.copy_(self.flat_param)
         self.step_count += 1
 
+    def sync_master_from_model(self) -> None:
+        """Re-derive the fp32 master weights from the (model-dtype) module
+        params — required after loading a model state_dict WITHOUT optimizer
+        state, else the next step would overwrite the loaded weights with
+        updates of the stale master."""
+        if self.flat_param_model is not self.flat_param:
+            self.flat_param.copy_(self.flat_param_model)
+
     def state_dict(self):
         return {
             "step_count": self.step_count,
+            # fp32 master weights: the model state_dict only carries the
+            # model-dtype (bf16) mirrors
+            "flat_param": self.flat_param,
             "exp_avg": self.exp_avg,
             "exp_avg_sq": self.exp_avg_sq,
             "reducer": {n: s.state_dict() for n, s in self.reducer.states.items()},
@@ -382,6 +393,12 @@ class FlatBertAdam:
 
     def load_state_dict(self, d):
         self.step_count = int(d["step_count"])
+        if "flat_param" in d:
+            self.flat_param.copy_(d["flat_param"])
+            if self.flat_param_model is not self.flat_param:
+                self.flat_param_model.copy_(self.flat_param)
+        else:
+            self.sync_master_from_model()
         self.exp_avg.copy_(d["exp_avg"])
         self.exp_avg_sq.copy_(d["exp_avg_sq"])
         for name, st_d in d.get("reducer", {}).items():

@@ -61,6 +61,10 @@ def load_checkpoint(
     """Returns (iteration, epoch, extra)."""
     state = torch.load(path, map_location=map_location, weights_only=False)
     model.load_state_dict(state["state_dict"])
-    if optimizer is not None and state.get("optimizer") is not None:
-        optimizer.load_state_dict(state["optimizer"])
+    if optimizer is not None:
+        if state.get("optimizer") is not None:
+            optimizer.load_state_dict(state["optimizer"])
+        elif hasattr(optimizer, "sync_master_from_model"):
+            # model-only restore: re-derive fp32 masters from loaded weights
+            optimizer.sync_master_from_model()
     return int(state.get("iter", 0)), int(state.get("epoch", 0)), state.get("extra", {})

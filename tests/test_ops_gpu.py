@@ -399,3 +399,40 @@ def test_bert_layer_fused_attention_matches_eager():
     assert torch.allclose(out_fused.float(), out_eager.float(), atol=0.12), (
         (out_fused.float() - out_eager.float()).abs().max()
     )
+
+
+def test_checkpoint_resume_bf16_masters(tmp_path_factory):
+    """Pure-bf16 path: resume must restore the fp32 MASTER weights, not just
+    the bf16 model mirrors — otherwise the first post-resume step applies an
+    update to the stale (random-init) master and destroys the loaded model."""
+    from oktopk_amd.config import EngineConfig, OkTopkConfig
+    from oktopk_amd.trainer import Trainer
+    from oktopk_amd.utils import load_checkpoint, save_checkpoint
+
+    tmp = tmp_path_factory.mktemp("ck")
+    kw = dict(num_hidden_layers=2, hidden_size=128, num_attention_heads=2,
+              intermediate_size=256, vocab_size=1000)
+    cfg = EngineConfig(compressor="oktopk", density=0.01,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    tr = Trainer("bert_base", batch_size=2, seq_len=128, cfg=cfg,
+                 model_kwargs=kw, dtype="bf16")
+    tr.batches.input_ids.clamp_(max=999)
+    tr.batches.mlm_labels.clamp_(max=999)
+    for _ in range(3):
+        tr.step()
+    path = str(tmp / "ck.pth")
+    save_checkpoint(path, tr.model, tr.opt, iteration=3, epoch=0)
+    w_saved = tr.opt.flat_param.clone()
+
+    tr2 = Trainer("bert_base", batch_size=2, seq_len=128, cfg=cfg,
+                  model_kwargs=kw, dtype="bf16")
+    tr2.batches.input_ids.clamp_(max=999)
+    tr2.batches.mlm_labels.clamp_(max=999)
+    load_checkpoint(path, tr2.model, tr2.opt)
+    assert torch.allclose(tr2.opt.flat_param, w_saved)
+    before = tr2.opt.flat_param.clone()
+    tr2.step()
+    delta = (tr2.opt.flat_param - before).abs().max().item()
+    # one Adam step at lr 2e-4 moves weights by <= ~lr scale, not by the
+    # distance between two random inits (~0.04)
+    assert delta < 5e-3, delta
