@@ -81,7 +81,7 @@ class TensorState:
         }
 
     def load_state_dict(self, d: dict) -> None:
-        self.residual.copy_(d["residual"])
+        self.residual.copy_(d["residual"].to(self.residual.device))
         self.counter = int(d["counter"])
         self.tau_local = float(d["tau_local"])
         self.tau_global = float(d["tau_global"])

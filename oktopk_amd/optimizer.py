@@ -208,10 +208,11 @@ class _DistributedOptimizer:
             if name in self.reducer.states:
                 self.reducer.states[name].load_state_dict(st_d)
             else:
-                # rebuild lazily on first run(); stash tensors directly
+                # rebuild on the bucket device (checkpoint may be CPU-mapped)
                 from .allreducer import TensorState
 
-                st = TensorState(residual=st_d["residual"].clone())
+                dev = self.buckets[0].flat.device if self.buckets else st_d["residual"].device
+                st = TensorState(residual=st_d["residual"].clone().to(dev))
                 st.load_state_dict(st_d)
                 self.reducer.states[name] = st
 
@@ -406,6 +407,9 @@ class FlatBertAdam:
 
             st = self.reducer.states.get(name)
             if st is None:
-                st = TensorState(residual=st_d["residual"].clone())
+                # checkpoint tensors may be on CPU (map_location): place the
+                # restored state on the optimizer's device
+                dev = self.flat_grad.device
+                st = TensorState(residual=st_d["residual"].clone().to(dev))
                 self.reducer.states[name] = st
             st.load_state_dict(st_d)
