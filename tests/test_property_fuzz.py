@@ -1,0 +1,59 @@
+"""Property-based fuzz of the engine invariants (hypothesis).
+
+For any tensor size, density and compressor: results stay finite and the
+error-feedback mass invariant holds at world 1 (result + residual ==
+accumulated input, modulo topkA2's documented truncation)."""
+import pytest
+import torch
+from hypothesis import given, settings, strategies as st
+
+from oktopk_amd import AllReducer, Comm, EngineConfig
+from oktopk_amd.config import OkTopkConfig
+
+CONSERVING = ["oktopk", "topkA", "topkAopt", "topkSA", "gtopk", "gaussiank",
+              "gaussiankSA", "dense"]
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    n=st.integers(min_value=1, max_value=4096),
+    density=st.floats(min_value=0.001, max_value=1.0),
+    comp=st.sampled_from(CONSERVING),
+    iters=st.integers(min_value=1, max_value=4),
+    seed=st.integers(min_value=0, max_value=10_000),
+    scale=st.sampled_from([1.0, 1e-6, 1e4]),
+)
+def test_engine_invariants_world1(n, density, comp, iters, seed, scale):
+    cfg = EngineConfig(compressor=comp, density=density,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(None), cfg)
+    g = torch.Generator().manual_seed(seed)
+    total_in = torch.zeros(n)
+    total_out = torch.zeros(n)
+    for _ in range(iters):
+        t = torch.randn(n, generator=g) * scale
+        total_in += t
+        out = eng.run("w", t.clone())
+        assert torch.isfinite(out).all()
+        total_out += out
+    res = eng.states["w"].residual
+    assert torch.isfinite(res).all()
+    err = (total_out + res - total_in).abs().max().item()
+    tol = max(1e-4 * scale, 1e-6)
+    assert err <= tol, (comp, n, density, err)
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    n=st.integers(min_value=1, max_value=2048),
+    density=st.floats(min_value=0.001, max_value=0.5),
+    seed=st.integers(min_value=0, max_value=1000),
+)
+def test_compact_matches_reference_fuzz(n, density, seed):
+    from oktopk_amd.ops import reference as R
+
+    t = torch.randn(n, generator=torch.Generator().manual_seed(seed))
+    tau = R.kth_abs_value(t, max(1, int(n * density)))
+    idx, val = R.compact_gt(t, tau)
+    assert (t.abs()[idx.long()] > tau).all()
+    assert int((t.abs() > tau).sum()) == idx.numel()
