@@ -61,7 +61,11 @@ def build_trainer(args, comm, compressor):
     )
     cfg = EngineConfig.preset(preset, compressor=compressor, density=args.density,
                               wire_dtype=args.wire_dtype, dense_warmup_iters=0)
+    # LSTM recipes run fp32 (MIOpen fused RNN has no bf16 path — 4x
+    # slower under autocast, measured); conv/transformer recipes bf16
     dtype = "bf16" if torch.cuda.is_available() else "fp32"
+    if args.model.startswith("lstm"):
+        dtype = "fp32"
     return Trainer(
         model_name=args.model,
         batch_size=args.batch_size,
