@@ -150,7 +150,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if torch.cuda.is_available() else "fp32",
+            "dtype": ("fp32" if args.model.startswith("lstm") or not torch.cuda.is_available()
+                      else "bf16"),
             "data": "synthetic",
             "config": {
                 "model": args.model,
