@@ -117,5 +117,20 @@ def fused_adam_(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps, weight_
     )
 
 
+def fused_adam_mirror_(param, grad, exp_avg, exp_avg_sq, param_bf16, lr, beta1,
+                       beta2, eps, weight_decay):
+    """Adam step + bf16 weight-mirror write in one pass (GPU); CPU reference
+    steps then casts."""
+    b = _backend(param)
+    if hasattr(b, "fused_adam_mirror_"):
+        return b.fused_adam_mirror_(
+            param, grad, exp_avg, exp_avg_sq, param_bf16, float(lr), float(beta1),
+            float(beta2), float(eps), float(weight_decay)
+        )
+    b.fused_adam_(param, grad, exp_avg, exp_avg_sq, float(lr), float(beta1),
+                  float(beta2), float(eps), float(weight_decay))
+    param_bf16.copy_(param)
+
+
 def l2norm(t: torch.Tensor) -> float:
     return float(_backend(t).l2norm(t))

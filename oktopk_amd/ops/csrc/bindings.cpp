@@ -32,8 +32,8 @@ void launch_ef_restore(float*, float*, int64_t, hipStream_t);
 void launch_ef_upcast(float*, float*, const void*, int64_t, hipStream_t);
 void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
                 hipStream_t);
-void launch_adam(float*, const float*, float*, float*, int64_t, float, float, float,
-                 float, float, hipStream_t);
+void launch_adam(float*, const float*, float*, float*, void*, int64_t, float, float,
+                 float, float, float, hipStream_t);
 void launch_sumsq(const float*, int64_t, double*, hipStream_t);
 void launch_linear_gelu(const void*, const void*, const float*, void*, void*, int,
                         int, int, int, hipStream_t);
@@ -326,8 +326,23 @@ static void fused_adam_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     check_f32_1d(v, "v");
     const at::cuda::CUDAGuard guard(p.device());
     launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-                v.data_ptr<float>(), p.numel(), (float)lr, (float)b1, (float)b2,
-                (float)eps, (float)wd, cur_stream());
+                v.data_ptr<float>(), nullptr, p.numel(), (float)lr, (float)b1,
+                (float)b2, (float)eps, (float)wd, cur_stream());
+}
+
+static void fused_adam_mirror_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                               torch::Tensor v, torch::Tensor p_bf16, double lr,
+                               double b1, double b2, double eps, double wd) {
+    check_f32_1d(p, "p");
+    check_f32_1d(g, "g");
+    check_f32_1d(m, "m");
+    check_f32_1d(v, "v");
+    TORCH_CHECK(p_bf16.scalar_type() == torch::kBFloat16 && p_bf16.is_contiguous() &&
+                p_bf16.numel() == p.numel());
+    const at::cuda::CUDAGuard guard(p.device());
+    launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                v.data_ptr<float>(), p_bf16.data_ptr(), p.numel(), (float)lr,
+                (float)b1, (float)b2, (float)eps, (float)wd, cur_stream());
 }
 
 static double l2norm(torch::Tensor t) {
@@ -488,6 +503,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "t = float(g_bf16) + r; r = t (fused upcast + EF restore)");
     m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
     m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
+    m.def("fused_adam_mirror_", &fused_adam_mirror_,
+          "fused Adam step + bf16 weight-mirror write");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
     m.def("attn_fwd", &attn_fwd,
           "fused self-attention forward: softmax(QK^T*scale+mask) dropout @ V "

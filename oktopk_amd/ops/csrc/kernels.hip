@@ -551,8 +551,17 @@ extern "C" void launch_sgd(float* p, const float* g, float* buf, int64_t n, floa
                        p, g, buf, n, lr, mom, wd, nesterov, mom != 0.f);
 }
 
+__device__ __forceinline__ short adam_f2b(float f) {
+    union { float f; uint32_t i; } c;
+    c.f = f;
+    uint32_t lsb = (c.i >> 16) & 1;
+    c.i += 0x7fff + lsb;
+    return (short)(c.i >> 16);
+}
+
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
-                            float* __restrict__ m, float* __restrict__ v, int64_t n,
+                            float* __restrict__ m, float* __restrict__ v,
+                            short* __restrict__ p_bf16, int64_t n,
                             float lr, float b1, float b2, float eps, float wd) {
     int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * BLOCK;
@@ -562,15 +571,20 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
         float vi = v[i] * b2 + (1.f - b2) * gi * gi;
         m[i] = mi; v[i] = vi;
         float up = mi / (sqrtf(vi) + eps) + wd * p[i];
-        p[i] -= lr * up;
+        float pn = p[i] - lr * up;
+        p[i] = pn;
+        // fused bf16 weight-mirror write (saves the separate cast pass of
+        // the pure-bf16 model path)
+        if (p_bf16) p_bf16[i] = adam_f2b(pn);
     }
 }
 
-extern "C" void launch_adam(float* p, const float* g, float* m, float* v, int64_t n,
+extern "C" void launch_adam(float* p, const float* g, float* m, float* v,
+                            void* p_bf16, int64_t n,
                             float lr, float b1, float b2, float eps, float wd,
                             hipStream_t stream) {
     hipLaunchKernelGGL(adam_kernel, dim3(n_blocks(n, 4)), dim3(BLOCK), 0, stream,
-                       p, g, m, v, n, lr, b1, b2, eps, wd);
+                       p, g, m, v, (short*)p_bf16, n, lr, b1, b2, eps, wd);
 }
 
 // ---------------------------------------------------------------------------

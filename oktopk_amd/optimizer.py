@@ -354,23 +354,29 @@ class FlatBertAdam:
             gn = ops.l2norm(self.flat_grad)
             if gn > self.max_grad_norm:
                 self.flat_grad.mul_(self.max_grad_norm / (gn + 1e-6))
-        # 3. fused Adam: one launch per weight-decay group (2 total)
+        # 3. fused Adam: one launch per weight-decay group (2 total); on
+        # the bf16 path the model-weight mirror is written BY the Adam
+        # kernel (saves a separate 660 MB cast pass)
         lr = self.current_lr()
         b1, b2 = self.betas
         d = self.decay_numel
+        mirrored = self.flat_param_model is not self.flat_param
+        def _adam(sl, wd):
+            if mirrored:
+                ops.fused_adam_mirror_(
+                    self.flat_param[sl], self.flat_grad[sl], self.exp_avg[sl],
+                    self.exp_avg_sq[sl], self.flat_param_model[sl], lr, b1, b2,
+                    self.eps, wd,
+                )
+            else:
+                ops.fused_adam_(
+                    self.flat_param[sl], self.flat_grad[sl], self.exp_avg[sl],
+                    self.exp_avg_sq[sl], lr, b1, b2, self.eps, wd,
+                )
         if d > 0:
-            ops.fused_adam_(
-                self.flat_param[:d], self.flat_grad[:d], self.exp_avg[:d],
-                self.exp_avg_sq[:d], lr, b1, b2, self.eps, self.weight_decay,
-            )
+            _adam(slice(None, d), self.weight_decay)
         if d < self.numel:
-            ops.fused_adam_(
-                self.flat_param[d:], self.flat_grad[d:], self.exp_avg[d:],
-                self.exp_avg_sq[d:], lr, b1, b2, self.eps, 0.0,
-            )
-        # 4. mirror fp32 master -> model-dtype weights (one bulk cast)
-        if self.flat_param_model is not self.flat_param:
-            self.flat_param_model.copy_(self.flat_param)
+            _adam(slice(d, None), 0.0)
         self.step_count += 1
 
     def sync_master_from_model(self) -> None:

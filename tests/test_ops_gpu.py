@@ -436,3 +436,17 @@ def test_checkpoint_resume_bf16_masters(tmp_path_factory):
     # one Adam step at lr 2e-4 moves weights by <= ~lr scale, not by the
     # distance between two random inits (~0.04)
     assert delta < 5e-3, delta
+
+
+def test_fused_adam_mirror():
+    p = randn_gpu(10_000, seed=60)
+    g = randn_gpu(10_000, seed=61)
+    m = torch.zeros(10_000).cuda()
+    v = torch.zeros(10_000).cuda()
+    pb = torch.zeros(10_000, dtype=torch.bfloat16).cuda()
+    p_ref = p.clone()
+    m_ref, v_ref = m.clone(), v.clone()
+    hip().fused_adam_mirror_(p, g, m, v, pb, 1e-3, 0.9, 0.999, 1e-6, 0.01)
+    hip().fused_adam_(p_ref, g, m_ref, v_ref, 1e-3, 0.9, 0.999, 1e-6, 0.01)
+    assert torch.equal(p, p_ref)
+    assert torch.equal(pb, p.to(torch.bfloat16))
