@@ -1,0 +1,99 @@
+"""Coverage for aux utilities: comm cost models, preemption handler,
+metric histograms, FLOPs counter on RNN/BERT, GLUE TSV loading."""
+import json
+import os
+import signal
+
+import pytest
+import torch
+
+from oktopk_amd.utils import (
+    MetricWriter,
+    alpha_beta_time,
+    get_model_complexity_info,
+    predict_allreduce_time,
+)
+from oktopk_amd.utils.comm_model import predict_oktopk_volume, predict_sparse_allgather_time
+
+
+def test_comm_models_sane():
+    assert alpha_beta_time(0) > 0  # latency floor
+    t1 = predict_allreduce_time(100e6, 8)
+    t2 = predict_allreduce_time(200e6, 8)
+    assert t2 > t1 > 0
+    assert predict_allreduce_time(1e9, 1) == 0.0
+    assert predict_sparse_allgather_time(100_000, 8) < predict_allreduce_time(440e6, 8)
+    assert predict_oktopk_volume(1000, 8) == 6000
+
+
+def test_preemption_handler(tmp_path):
+    from oktopk_amd.elastic import install_preemption_handler
+
+    hits = []
+    old = signal.getsignal(signal.SIGUSR1)
+    try:
+        install_preemption_handler(lambda: hits.append(1), signals=(signal.SIGUSR1,))
+        with pytest.raises(SystemExit) as e:
+            os.kill(os.getpid(), signal.SIGUSR1)
+            signal.sigtimedwait([], 0)  # let the handler run
+        assert hits == [1]
+        assert e.value.code == 128 + signal.SIGUSR1
+    finally:
+        signal.signal(signal.SIGUSR1, old)
+
+
+def test_metric_writer_histogram(tmp_path):
+    path = str(tmp_path / "m.jsonl")
+    w = MetricWriter(path, rank=0)
+    w.add_scalar("loss", 1.5, 3)
+    w.add_histogram("weights", torch.randn(500), 3, bins=4)
+    w.add_dict({"a": 1, "b": "skip-me"}, 4)
+    w.close()
+    lines = [json.loads(l) for l in open(path)]
+    assert lines[0]["tag"] == "loss" and lines[0]["value"] == 1.5
+    hist = lines[1]
+    assert len(hist["hist_quantiles"]) == 5
+    assert hist["hist_quantiles"] == sorted(hist["hist_quantiles"])
+    assert lines[2]["tag"] == "a"
+    # rank != 0 writes nothing
+    w2 = MetricWriter(str(tmp_path / "m2.jsonl"), rank=1)
+    w2.add_scalar("x", 1, 1)
+    w2.close()
+    assert not os.path.exists(str(tmp_path / "m2.jsonl"))
+
+
+def test_flops_counter_lstm_and_bert():
+    from oktopk_amd import models
+
+    m = models.create_net("lstman4", rnn_hidden_size=64, nb_layers=2)
+    flops, params = get_model_complexity_info(
+        m, input_constructor=lambda _: torch.randn(1, 1, 161, 51), input_res=None
+    )
+    assert flops > 1e6 and params == sum(p.numel() for p in m.parameters())
+
+    bert = models.create_net(
+        "bert_base", num_hidden_layers=1, hidden_size=64, num_attention_heads=2,
+        intermediate_size=128, vocab_size=500,
+    )
+    def mk(_):
+        return {"input_ids": torch.randint(0, 500, (1, 16))}
+    flops_b, _ = get_model_complexity_info(bert, input_constructor=mk, input_res=None)
+    assert flops_b > 1e5
+
+
+def test_glue_tsv_roundtrip(tmp_path):
+    from oktopk_amd.glue import score_files
+
+    pred = tmp_path / "pred.tsv"
+    gold = tmp_path / "gold.tsv"
+    pred.write_text("id\tlabel\n0\tyes\n1\tno\n2\tyes\n")
+    gold.write_text("id\tlabel\n0\tyes\n1\tyes\n2\tyes\n")
+    s = score_files("sst-2", str(pred), str(gold))
+    assert s["acc"] == pytest.approx(2 / 3)
+
+
+def test_checkpoint_path_scheme(tmp_path):
+    from oktopk_amd.utils.checkpoint import checkpoint_path
+
+    p = checkpoint_path(str(tmp_path), "bert", epoch=3, stage=1)
+    assert p.endswith("bert/checkpoint.1.pth.tar.epoch.3")
