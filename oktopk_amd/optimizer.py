@@ -332,6 +332,11 @@ class FlatBertAdam:
         for p, (o, n_) in zip(self.params, self.slots):
             p.grad = self.flat_grad_model[o : o + n_].view_as(p)
 
+    @property
+    def param_groups(self):
+        """Minimal torch-optimizer-surface shim (LR tooling interop)."""
+        return [{"lr": self.lr, "params": self.params}]
+
     def current_lr(self) -> float:
         if self.t_total > 0:
             return self.lr * self.schedule(self.step_count / self.t_total, self.warmup)

@@ -149,6 +149,9 @@ class Trainer:
         )
         self.nsteps_update = max(1, nsteps_update)
 
+        if self.device.type == "cuda":
+            # static shapes throughout: let MIOpen find fast conv algos
+            torch.backends.cudnn.benchmark = True
         self.model = models.create_net(model_name, **(model_kwargs or {})).to(self.device)
         # broadcast initial weights (reference comm.bcast(state_dict),
         # VGG/main_trainer.py:52)
@@ -277,10 +280,29 @@ class Trainer:
 
     def set_epoch(self, epoch: int) -> None:
         """Advance the engine's dynamic density schedule (reference
-        train_epoch plumbing, VGG/allreducer.py:207-208)."""
+        train_epoch plumbing, VGG/allreducer.py:207-208) and apply the LR
+        schedule."""
         red = getattr(self.opt, "reducer", None)
         if red is not None:
             red.train_epoch = epoch
+        self.adjust_learning_rate(epoch)
+
+    def adjust_learning_rate(self, epoch: int) -> None:
+        """Step-decay LR schedules for the SGD recipes (reference
+        _adjust_learning_rate_general, VGG/dl_trainer.py:507-570: x0.1 at
+        the milestone epochs of each workload)."""
+        groups = getattr(self.opt, "param_groups", None)
+        if not groups:
+            return
+        base = self._base_lr if hasattr(self, "_base_lr") else None
+        if base is None:
+            self._base_lr = base = groups[0]["lr"]
+        if self.model_name.startswith("bert"):
+            return  # BertAdam has its own warmup schedule
+        milestones = (81, 122) if self.batches.family == "cifar" else (10, 20)
+        factor = 0.1 ** sum(1 for m in milestones if epoch >= m)
+        for g in groups:
+            g["lr"] = base * factor
 
     @torch.no_grad()
     def evaluate(self, n_batches: int = 4) -> dict:

@@ -188,3 +188,19 @@ def test_eps_oracle_with_grad_src():
     eng.run("w", t, grad_src=g.to(torch.bfloat16))
     eps = eng.eps_log[0][1]
     assert eps < 0.35, eps  # bf16 wire rounding + strict-> tie loss only
+
+
+def test_lr_schedule_step_decay():
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.trainer import Trainer
+
+    tr = Trainer("caffe_cifar", batch_size=2, cfg=EngineConfig(compressor="dense"),
+                 dtype="fp32", lr=0.1)
+    tr.set_epoch(0)
+    assert tr.opt.param_groups[0]["lr"] == pytest.approx(0.1)
+    tr.set_epoch(81)
+    assert tr.opt.param_groups[0]["lr"] == pytest.approx(0.01)
+    tr.set_epoch(122)
+    assert tr.opt.param_groups[0]["lr"] == pytest.approx(0.001)
+    tr.set_epoch(5)  # schedules are absolute, not cumulative
+    assert tr.opt.param_groups[0]["lr"] == pytest.approx(0.1)
