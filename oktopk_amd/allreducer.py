@@ -793,3 +793,51 @@ def _gaussian_threshold(t: torch.Tensor, density: float) -> float:
         return abs(mean)
     right = stats.norm.ppf(1 - density / 2, loc=mean, scale=std)
     return float(abs(right))
+
+
+def _benchmark_main():  # pragma: no cover
+    """Standalone engine benchmark (reference benchmark_gtopk_sparse_allreduce,
+    VGG/allreducer.py:1649-1677: random 25M-float tensor, 10 iterations,
+    average time).  Run single-process or under torchrun:
+        python -m oktopk_amd.allreducer --compressor oktopk --numel 25000000
+    """
+    import argparse
+    import time as _time
+
+    from .comm import init_from_env
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--compressor", default="oktopk")
+    ap.add_argument("--numel", type=int, default=25_000_000)
+    ap.add_argument("--density", type=float, default=0.001)
+    ap.add_argument("--iters", type=int, default=10)
+    args = ap.parse_args()
+
+    comm = init_from_env()
+    dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    cfg = EngineConfig(compressor=args.compressor, density=args.density)
+    cfg.oktopk.dense_warmup_iters = 0
+    eng = AllReducer(comm, cfg)
+    g = torch.Generator().manual_seed(comm.rank)
+    t0 = torch.randn(args.numel, generator=g).to(dev)
+    for _ in range(3):
+        eng.run("bench", t0.clone())
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+    comm.barrier()
+    start = _time.perf_counter()
+    for _ in range(args.iters):
+        eng.run("bench", t0.clone())
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+    comm.barrier()
+    dt = (_time.perf_counter() - start) / args.iters
+    if comm.rank == 0:
+        print(f"{args.compressor} numel={args.numel} density={args.density} "
+              f"P={comm.size}: {1000 * dt:.3f} ms/iter")
+        print("phases:", {k: round(1000 * v / (args.iters + 3), 3)
+                          for k, v in eng.timers.get("bench", {}).items()})
+
+
+if __name__ == "__main__":  # pragma: no cover
+    _benchmark_main()
