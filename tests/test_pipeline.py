@@ -247,3 +247,35 @@ def _pipedream_2stage(rank):
 
 def test_pipedream_2stage_runs():
     run_dist(_pipedream_2stage, 2)
+
+
+def _pipeline4_worker(rank):
+    """4-stage pipeline (exercises BertIntermediateStage, untested by the
+    2-stage cases) must match the unsplit model's loss and stage-0 grads."""
+    from oktopk_amd.pipeline import PipelineRuntime, partition_bert
+
+    model = _model()
+    stages = partition_bert(model, 4)
+    stage = stages[rank]
+    rt = PipelineRuntime(stage, stage_id=rank, num_stages=4)
+    opt = torch.optim.SGD(stage.parameters(), lr=0.0)
+    mbs = _microbatches()
+    if rank == 0:
+        my = [{k: m[k] for k in ("input_ids", "token_type_ids", "attention_mask")}
+              for m in mbs]
+    elif rank == 3:
+        my = [{k: m[k] for k in ("attention_mask", "masked_lm_labels",
+                                 "next_sentence_label")} for m in mbs]
+    else:
+        my = [{"attention_mask": m["attention_mask"]} for m in mbs]
+    loss = rt.run_step_1f1b(my, opt)
+    ref_loss, ref_grad = _reference_loss()
+    if rank == 3:
+        assert abs(loss - ref_loss) < 1e-4, (loss, ref_loss)
+    if rank == 0:
+        got = stage.layers[0].fc1.weight.grad
+        assert torch.allclose(got, ref_grad, atol=1e-5), (got - ref_grad).abs().max()
+
+
+def test_pipeline_4stage_1f1b_matches_unsplit():
+    run_dist(_pipeline4_worker, 4)
