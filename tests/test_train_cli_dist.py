@@ -1,0 +1,33 @@
+"""End-to-end CLI under torchrun (2 gloo ranks, CPU): train 1 epoch with
+checkpointing, then resume with --pretrain."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_cli(tmp, extra):
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29549", "-m", "oktopk_amd.train",
+           "--dnn", "resnet20", "--batch-size", "2", "--density", "0.05",
+           "--iters-per-epoch", "3", "--max-epochs", "1", "--dtype", "fp32",
+           "--logdir", os.path.join(tmp, "logs"),
+           "--checkpoint-dir", os.path.join(tmp, "ck")] + extra
+    return subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=420)
+
+
+def test_cli_train_and_resume(tmp_path):
+    tmp = str(tmp_path)
+    out = _run_cli(tmp, [])
+    assert out.returncode == 0, out.stderr[-2000:]
+    ck = os.path.join(tmp, "ck", "checkpoint.epoch.0.pth")
+    assert os.path.exists(ck)
+    metrics = os.path.join(tmp, "logs", "metrics.jsonl")
+    lines = [json.loads(l) for l in open(metrics)]
+    assert any(d["tag"] == "train/loss" for d in lines)
+
+    out2 = _run_cli(tmp, ["--pretrain", ck, "--max-epochs", "1"])
+    assert out2.returncode == 0, out2.stderr[-2000:]
