@@ -104,6 +104,14 @@ class AllReducer:
         self.timers: Dict[str, Dict[str, float]] = {}
         self.eps_log: List[Tuple[int, float]] = []
         self.train_epoch = 0  # drives the dynamic density schedule
+        # Phase timers are host perf_counter spans around async GPU launches;
+        # without a device sync at entry the first host-blocking point inside
+        # compress absorbs whatever GPU work was queued before run() (e.g. a
+        # whole backward graph replay), inflating "compress" with queue depth.
+        # Benchmarks that report phase breakdowns set timing_sync=True; the
+        # hook-driven overlap path leaves it False (a sync per bucket would
+        # destroy backward/compress overlap).
+        self.timing_sync = False
 
     def set_comm(self, comm: Comm) -> None:
         """Swap the communicator after an elastic shrink; per-world state
@@ -137,6 +145,8 @@ class AllReducer:
         dense mean gradient over all ranks.
         """
         t = tensor.reshape(-1)
+        if self.timing_sync and t.is_cuda:
+            torch.cuda.synchronize()
         st = self.state(name, t)
         comp = self.cfg.compressor
         ok = self.cfg.oktopk
