@@ -403,10 +403,13 @@ class AllReducer:
             gidx, gval = ops.compact_gt(reduced, st.tau_global)
         gidx = gidx + lo  # absolute indices (int32 + int offset)
 
-        pack = comm.to_comm(self._pack(gidx, gval))
-        elem_counts = [int(x) for x in comm.allgather_sizes(gidx.numel(), comm.device)]
-        buf, _ = comm.allgatherv(pack, sizes=[self._pack_ints(c) for c in elem_counts])
-        all_idx, all_val = self._unpack(buf, elem_counts)
+        if ok.balanced_allgather and P > 1:
+            all_idx, all_val = self._balanced_round2(gidx, gval)
+        else:
+            pack = comm.to_comm(self._pack(gidx, gval))
+            elem_counts = [int(x) for x in comm.allgather_sizes(gidx.numel(), comm.device)]
+            buf, _ = comm.allgatherv(pack, sizes=[self._pack_ints(c) for c in elem_counts])
+            all_idx, all_val = self._unpack(buf, elem_counts)
         all_idx = all_idx.to(t.device)
         all_val = all_val.to(t.device)
         self._time(name, "allgather", time.perf_counter() - s3)
@@ -448,6 +451,81 @@ class AllReducer:
             st.mask[g_sel_idx.long()] = False  # cheap sparse reset
         self._time(name, "merge", time.perf_counter() - s4)
         return result
+
+    def _balanced_round2(
+        self, gidx: torch.Tensor, gval: torch.Tensor
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Load-balanced second allgather (reference
+        BERT/bert/allreducer.py:615-715, via send/recv metas there).
+
+        Owners' survivor counts s_r are skewed; the plain path pads the
+        AllGather to max(s_r) so wire volume is P*max(s_r).  Here survivors
+        are first re-balanced into equal ceil(S/P) position blocks with one
+        alltoallv (volume bounded by the imbalance — RCCL schedules it over
+        the xGMI p2p links), then exchanged with ONE perfectly-equal
+        AllGather of ceil(S/P) packed elements per rank.  Recv counts are
+        derived locally from the size vector (no extra size exchange).
+        Entries come back in global position order — identical contents and
+        order to the plain path, so both are interchangeable mid-run.
+        """
+        comm = self.comm
+        P, rank = comm.size, comm.rank
+        sizes = [int(x) for x in comm.allgather_sizes(gidx.numel(), comm.device)]
+        S = sum(sizes)
+        if S == 0:
+            return gidx[:0], gval[:0]
+        q = (S + P - 1) // P
+        off = sum(sizes[:rank])
+        s = sizes[rank]
+        # my entries occupy global positions [off, off+s); destination d's
+        # balanced block is [d*q, (d+1)*q)
+        send_counts = [
+            max(0, min(off + s, (d + 1) * q) - max(off, d * q)) for d in range(P)
+        ]
+        recv_counts = []
+        o = 0
+        for r in range(P):
+            recv_counts.append(
+                max(0, min(o + sizes[r], (rank + 1) * q) - max(o, rank * q))
+            )
+            o += sizes[r]
+        segs, c0 = [], 0
+        for d in range(P):
+            c1 = c0 + send_counts[d]
+            segs.append(self._pack(gidx[c0:c1], gval[c0:c1]))
+            c0 = c1
+        send = torch.cat(segs)
+        recv = comm.alltoallv(
+            comm.to_comm(send),
+            [self._pack_ints(c) for c in send_counts],
+            [self._pack_ints(c) for c in recv_counts],
+        )
+        b_idx, b_val = self._unpack(recv, recv_counts)
+        m = b_idx.numel()
+        if m < q:  # only the last block(s) can be short
+            b_idx = torch.cat([b_idx, b_idx.new_zeros(q - m)])
+            b_val = torch.cat([b_val, b_val.new_zeros(q - m)])
+        buf = comm.allgather_eq(comm.to_comm(self._pack(b_idx, b_val)))
+        block_counts = [min(q, max(0, S - r * q)) for r in range(P)]
+        return self._unpack_strided(buf, q, block_counts)
+
+    def _unpack_strided(
+        self, buf: torch.Tensor, q: int, counts: List[int]
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Unpack equal packed blocks of capacity q with true counts
+        `counts` (trailing padding stripped).  Values come back fp32."""
+        stride = self._pack_ints(q)
+        idxs, vals = [], []
+        for r, c in enumerate(counts):
+            base = r * stride
+            idxs.append(buf[base : base + c])
+            if self._wire_bf16:
+                vals.append(
+                    buf[base + q : base + stride].view(torch.bfloat16)[:c].float()
+                )
+            else:
+                vals.append(buf[base + q : base + q + c].view(torch.float32))
+        return torch.cat(idxs), torch.cat(vals)
 
     # -- topkA / topkA2 (VGG/allreducer.py:34-69,481-531) ----------------
     def _topkA(

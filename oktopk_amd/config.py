@@ -49,6 +49,13 @@ class OkTopkConfig:
     global_lo_den: int = 3
     global_hi_num: int = 4
     global_hi_den: int = 3
+    # Round-2 load-balanced redistribution (reference
+    # BERT/bert/allreducer.py:615-715): re-balance the per-owner survivors
+    # into equal ceil(S/P) blocks with one alltoallv, then exchange with a
+    # single perfectly-equal AllGather — wire volume S+pad instead of
+    # P*max(s_r).  Pays when region survivor counts are skewed; the default
+    # pad-to-max path is one collective fewer.
+    balanced_allgather: bool = False
 
 
 @dataclass

@@ -40,6 +40,8 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--compressor", type=str, default="oktopk")
     p.add_argument("--density", type=float, default=0.02)
     p.add_argument("--sigma-scale", type=float, default=2.5)
+    p.add_argument("--balanced-allgather", action="store_true",
+                   help="oktopk round-2 load-balanced redistribution")
     p.add_argument("--dense-warmup", type=int, default=None,
                    help="dense allreduce iterations before sparsifying")
     # optimizer
@@ -72,7 +74,8 @@ def main(argv=None) -> int:
     preset = ("bert" if args.dnn.startswith("bert")
               else ("lstm" if args.dnn.startswith("lstm") else "vgg"))
     overrides = dict(compressor=args.compressor, density=args.density,
-                     profiling=args.profiling, profiling_norm=args.profiling_norm)
+                     profiling=args.profiling, profiling_norm=args.profiling_norm,
+                     balanced_allgather=args.balanced_allgather)
     if args.dense_warmup is not None:
         overrides["dense_warmup_iters"] = args.dense_warmup
     cfg = EngineConfig.preset(preset, **overrides)

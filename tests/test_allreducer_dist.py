@@ -224,3 +224,38 @@ def _bf16_wire_all_compressors(rank):
 
 def test_bf16_wire_topkA_gtopk_world2():
     run_dist(_bf16_wire_all_compressors, 2)
+
+
+def _balanced_equivalence(rank):
+    """Balanced round-2 redistribution (BERT/bert/allreducer.py:615-715)
+    returns the SAME entries in the SAME order as the pad-to-max path, so
+    the full oktopk stream is bit-identical under either setting."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    world = dist.get_world_size()
+    outs = {}
+    for balanced in (False, True):
+        cfg = EngineConfig(
+            compressor="oktopk", density=DENSITY,
+            oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                region_repartition_interval=4,
+                                balanced_allgather=balanced),
+        )
+        eng = AllReducer(Comm(dist.group.WORLD), cfg)
+        acc = []
+        for it in range(ITERS):
+            t = _grad(rank, it)
+            acc.append(eng.run("w", t).clone())
+        outs[balanced] = acc
+    for it, (a, b) in enumerate(zip(outs[False], outs[True])):
+        assert torch.equal(a, b), (it, (a - b).abs().max())
+
+
+def test_balanced_allgather_world2():
+    run_dist(_balanced_equivalence, 2)
+
+
+def test_balanced_allgather_world4():
+    run_dist(_balanced_equivalence, 4)
