@@ -403,7 +403,10 @@ class AllReducer:
             gidx, gval = ops.compact_gt(reduced, st.tau_global)
         gidx = gidx + lo  # absolute indices (int32 + int offset)
 
-        if ok.balanced_allgather and P > 1:
+        if P == 1:
+            # no communication: the packed wire round-trip is pure overhead
+            all_idx, all_val = gidx, gval
+        elif ok.balanced_allgather:
             all_idx, all_val = self._balanced_round2(gidx, gval)
         else:
             pack = comm.to_comm(self._pack(gidx, gval))
