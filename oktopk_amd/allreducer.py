@@ -388,20 +388,32 @@ class AllReducer:
         self._time(name, "alltoall", time.perf_counter() - s1)
 
         s2 = time.perf_counter()
-        reduced = torch.zeros(hi - lo, dtype=t.dtype, device=t.device)
-        if r_idx.numel():
-            # int32 arithmetic keeps the wire dtype end-to-end (n < 2^31)
-            ops.scatter_add_(reduced, r_idx.to(t.device) - lo, r_val.to(t.device))
-        self._time(name, "reduce", time.perf_counter() - s2)
-
-        # --- 4. round 2: global top-k + sparse allgather -----------------
-        s3 = time.perf_counter()
         exact = it % ok.global_threshold_recompute_interval == 0 or st.tau_global <= 0.0
-        if exact:
-            gidx, gval = ops.compact_gt(reduced, 0.0)
+        if P > 1:
+            reduced = torch.zeros(hi - lo, dtype=t.dtype, device=t.device)
+            if r_idx.numel():
+                # int32 arithmetic keeps the wire dtype end-to-end (n < 2^31)
+                ops.scatter_add_(reduced, r_idx.to(t.device) - lo, r_val.to(t.device))
+            self._time(name, "reduce", time.perf_counter() - s2)
+
+            # --- 4. round 2: global top-k + sparse allgather -------------
+            s3 = time.perf_counter()
+            if exact:
+                gidx, gval = ops.compact_gt(reduced, 0.0)
+            else:
+                gidx, gval = ops.compact_gt(reduced, st.tau_global)
+            gidx = gidx + lo  # absolute indices (int32 + int offset)
         else:
-            gidx, gval = ops.compact_gt(reduced, st.tau_global)
-        gidx = gidx + lo  # absolute indices (int32 + int offset)
+            # world-1: round 1 was the identity, so the reduced region IS the
+            # local selection — filter it directly instead of densifying
+            # n floats and re-compacting (saves two full-tensor passes)
+            self._time(name, "reduce", time.perf_counter() - s2)
+            s3 = time.perf_counter()
+            if exact:
+                gidx, gval = idx, val
+            else:
+                keep = val.abs() > st.tau_global
+                gidx, gval = idx[keep], val[keep]
 
         if P == 1:
             # no communication: the packed wire round-trip is pure overhead
