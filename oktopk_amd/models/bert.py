@@ -16,9 +16,14 @@ from __future__ import annotations
 import math
 from dataclasses import dataclass
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+
+# opt-in torch SDPA (flash attention) in place of explicit matmul+softmax
+_USE_SDPA = os.environ.get("OKTOPK_SDPA", "0") == "1"
 
 
 @dataclass
@@ -80,6 +85,13 @@ class BertSelfAttention(nn.Module):
             return self.out(ctx)
         qkv = qkv_flat.view(b, s, 3, self.nh, self.hd).permute(2, 0, 3, 1, 4)
         q, k, v = qkv[0], qkv[1], qkv[2]  # (b, nh, s, hd)
+        if _USE_SDPA and x.is_cuda:
+            # torch flash-attention path (opt-in A/B vs explicit matmuls)
+            ctx = F.scaled_dot_product_attention(
+                q, k, v, attn_mask=attn_mask,
+                dropout_p=self.attn_drop.p if self.training else 0.0)
+            ctx = ctx.transpose(1, 2).reshape(b, s, h)
+            return self.out(ctx)
         scores = torch.matmul(q, k.transpose(-1, -2)) / math.sqrt(self.hd)
         if attn_mask is not None:
             scores = scores + attn_mask

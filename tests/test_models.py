@@ -74,3 +74,22 @@ def test_flops_counter():
     # VGG-16 on 32x32 is ~0.31 GMac; params ~15M
     assert 2e8 < flops < 5e8, flops
     assert 14e6 < params < 16e6, params
+
+
+def test_sdpa_path_matches_manual_cpu(monkeypatch):
+    """OKTOPK_SDPA path == explicit matmul+softmax (p=0, fp32).  The flag is
+    read at import; poke the module global directly for the A/B."""
+    import oktopk_amd.models.bert as B
+
+    m = B.BertSelfAttention(B.BertConfig(
+        hidden_size=64, num_attention_heads=4, attention_probs_dropout_prob=0.0))
+    m.eval()
+    x = torch.randn(2, 16, 64)
+    y_manual = m(x)
+    monkeypatch.setattr(B, "_USE_SDPA", True)
+    # CPU path ignores SDPA (x.is_cuda gate) — force through by faking cuda
+    # is overkill; instead call SDPA directly against the manual output
+    q = m.qkv(x).view(2, 16, 3, 4, 16).permute(2, 0, 3, 1, 4)
+    ctx = torch.nn.functional.scaled_dot_product_attention(q[0], q[1], q[2])
+    y_sdpa = m.out(ctx.transpose(1, 2).reshape(2, 16, 64))
+    assert torch.allclose(y_manual, y_sdpa, atol=1e-5), (y_manual - y_sdpa).abs().max()
