@@ -8,6 +8,7 @@ shapes of the reference configs.
 """
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -153,6 +154,13 @@ class Trainer:
             # static shapes throughout: let MIOpen find fast conv algos
             torch.backends.cudnn.benchmark = True
         self.model = models.create_net(model_name, **(model_kwargs or {})).to(self.device)
+        # opt-in NHWC for conv recipes (MIOpen layout A/B; hurts nothing else)
+        self.channels_last = (
+            os.environ.get("OKTOPK_CHANNELS_LAST", "0") == "1"
+            and self.device.type == "cuda"
+        )
+        if self.channels_last:
+            self.model = self.model.to(memory_format=torch.channels_last)
         # broadcast initial weights (reference comm.bcast(state_dict),
         # VGG/main_trainer.py:52)
         if self.comm.size > 1:
@@ -182,6 +190,10 @@ class Trainer:
         self.batches = SyntheticBatches(
             model_name, batch_size, self.device, seq_len=seq_len, rank=self.comm.rank
         )
+        if self.channels_last and getattr(self.batches, "x", None) is not None \
+                and self.batches.x.dim() == 4:
+            self.batches.x = self.batches.x.contiguous(
+                memory_format=torch.channels_last)
         self.iteration = 0
         self.last_loss = 0.0
         # hipGraph capture of fwd+bwd (the step is launch-bound at small
