@@ -313,14 +313,21 @@ class AllReducer:
         # reference's count-per-candidate loop + separate compact-count pass
         # collapse into one tensor read (ops.compact_adaptive).
         s0 = time.perf_counter()
-        self._ef_restore(t, st)
         if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
+            self._ef_restore(t, st)
             st.tau_local = ops.kth_abs_value(t, k)
             idx, val = ops.compact_gt(t, st.tau_local)
         else:
+            # steady state: EF restore (+bf16 upcast) and candidate counting
+            # fuse into ONE streaming pass — the candidate taus derive from
+            # the previous iteration's threshold, so they are known before
+            # the restore (ops.compact_adaptive_ef)
             taus = [st.tau_local * ok.bump_scale ** i
                     for i in range(ok.bump_max_loops + 1)]
-            idx, val, chosen, _cnt = ops.compact_adaptive(t, taus, 4 * k // 3)
+            grad = st.grad_src.reshape(-1) if st.grad_src is not None else None
+            st.grad_src = None
+            idx, val, chosen, _cnt = ops.compact_adaptive_ef(
+                t, st.residual, grad, taus, 4 * k // 3)
             st.tau_local = taus[chosen]
         tau = st.tau_local
 

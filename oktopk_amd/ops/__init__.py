@@ -75,6 +75,27 @@ def compact_adaptive(t: torch.Tensor, taus, hi_limit: int):
     return idx, val, int(chosen), int(count)
 
 
+def compact_adaptive_ef(t: torch.Tensor, residual: torch.Tensor,
+                        grad, taus, hi_limit: int):
+    """Fused EF restore (+bf16 upcast when `grad` given) + adaptive-threshold
+    compaction: t = float(grad)+residual (or t+=residual), residual = t, and
+    select at the first candidate tau whose count fits hi_limit — one
+    streaming pass instead of EF + count."""
+    aligned = all(
+        x is None or x.data_ptr() % 16 == 0 for x in (t, residual, grad)
+    ) and (grad is None or grad.dtype == torch.bfloat16)
+    if t.is_cuda and not aligned:  # rare: unaligned views fall back unfused
+        if grad is not None:
+            ef_restore_upcast_(t, residual, grad)
+        else:
+            ef_restore_snapshot_(t, residual)
+        return compact_adaptive(t, taus, hi_limit)
+    out = _backend(t).compact_adaptive_ef(
+        t, residual, grad, [float(x) for x in taus], int(hi_limit))
+    idx, val, chosen, count = out
+    return idx, val, int(chosen), int(count)
+
+
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
     return _backend(dest).scatter_add_(dest, idx, val)
 

@@ -30,6 +30,8 @@ void launch_isin_sorted(const int32_t*, int64_t, const int32_t*, int64_t, bool*,
                         hipStream_t);
 void launch_ef_restore(float*, float*, int64_t, hipStream_t);
 void launch_ef_upcast(float*, float*, const void*, int64_t, hipStream_t);
+void launch_ef_count(float*, float*, const void*, int64_t, const float*, int,
+                     int64_t, int, int*, hipStream_t);
 void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
                 hipStream_t);
 void launch_adam(float*, const float*, float*, float*, void*, int64_t, float, float,
@@ -169,6 +171,60 @@ static std::vector<torch::Tensor> compact_adaptive(
                                t.options().dtype(torch::kInt32));
     launch_compact_count_multi(t.data_ptr<float>(), n, tf, ntau, g.chunk, g.nblocks,
                                counts.data_ptr<int>(), cur_stream());
+    auto h = counts.cpu();
+    const int* hp = h.data_ptr<int>();
+    int chosen = ntau - 1;
+    int64_t chosen_total = 0;
+    for (int c = 0; c < ntau; ++c) {
+        int64_t tot = 0;
+        for (int b = 0; b < nw; ++b) tot += hp[(int64_t)c * nw + b];
+        if (c == ntau - 1 || tot <= hi_limit) {
+            chosen = c;
+            chosen_total = tot;
+            break;
+        }
+    }
+    auto out = compact_finish(t, taus[chosen], g, hp + (int64_t)chosen * nw);
+    out.push_back(torch::tensor((int64_t)chosen));
+    out.push_back(torch::tensor(chosen_total));
+    return out;
+}
+
+// fused EF restore + adaptive compaction: one streaming pass performs
+//   t = float(grad_bf16) + residual (or t += residual), residual = t
+// and counts every candidate tau (compact pass A layout); then the usual
+// host bump rule + write pass.  Steady-state oktopk selection drops from
+// three tensor-scale passes (EF, count, write) to two.
+static std::vector<torch::Tensor> compact_adaptive_ef(
+    torch::Tensor t, torch::Tensor residual, c10::optional<torch::Tensor> grad,
+    std::vector<double> taus, int64_t hi_limit) {
+    check_f32_1d(t, "t");
+    check_f32_1d(residual, "residual");
+    TORCH_CHECK(residual.numel() == t.numel(), "residual size mismatch");
+    const at::cuda::CUDAGuard guard(t.device());
+    TORCH_CHECK(taus.size() >= 1 && taus.size() <= 8, "1..8 thresholds");
+    const void* gp = nullptr;
+    if (grad.has_value()) {
+        auto& gt = grad.value();
+        TORCH_CHECK(gt.scalar_type() == torch::kBFloat16 && gt.is_contiguous() &&
+                        gt.numel() == t.numel(),
+                    "grad must be contiguous bf16 of same numel");
+        gp = gt.data_ptr();
+    }
+    TORCH_CHECK((((uintptr_t)t.data_ptr() | (uintptr_t)residual.data_ptr() |
+                  (uintptr_t)gp) & 15) == 0,
+                "compact_adaptive_ef requires 16B-aligned buffers");
+    int ntau = (int)taus.size();
+    int64_t n = t.numel();
+    auto g = compact_geom(n);
+    float tf[8];
+    for (int j = 0; j < ntau; ++j) tf[j] = (float)taus[j];
+    const int nw = g.nblocks * 4;
+    auto counts = torch::empty({(int64_t)ntau * nw},
+                               t.options().dtype(torch::kInt32));
+    launch_ef_count(t.data_ptr<float>(), residual.data_ptr<float>(), gp, n, tf,
+                    ntau, g.chunk, g.nblocks, counts.data_ptr<int>(),
+                    cur_stream());
     auto h = counts.cpu();
     const int* hp = h.data_ptr<int>();
     int chosen = ntau - 1;
@@ -492,6 +548,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("compact_adaptive", &compact_adaptive,
           "one-pass adaptive-threshold compaction: counts all candidate taus, "
           "applies the bump rule, extracts at the chosen tau");
+    m.def("compact_adaptive_ef", &compact_adaptive_ef,
+          "fused EF restore(+bf16 upcast) + adaptive-threshold compaction",
+          py::arg("t"), py::arg("residual"), py::arg("grad"), py::arg("taus"),
+          py::arg("hi_limit"));
     m.def("kth_abs_value", &kth_abs_value, "exact k-th largest |t| via radix select");
     m.def("scatter_add_", &scatter_add_, "dest[idx] += val");
     m.def("zero_at_", &zero_at_, "t[idx] = 0");

@@ -529,6 +529,105 @@ extern "C" void launch_ef_upcast(float* t, float* r, const void* g, int64_t n,
                            dim3(BLOCK), 0, stream, t, r, (const short*)g, n8 * 8, n);
 }
 
+// ---------------------------------------------------------------------------
+// fused EF restore + compact pass A: one streaming pass does
+//   t = (float)g + r   (or t += r when no bf16 grad),  r = t
+// AND counts |t| > tau_c for up to 8 candidate thresholds in the compact
+// count layout [cand][block][wave].  The steady-state (non-exact) oktopk
+// iteration then needs only this pass + the compact write pass: the separate
+// count read of t (440 MB at BERT-base scale) disappears.  Valid because the
+// candidate taus derive from the PREVIOUS iteration's threshold, known
+// before the restore.  Uses the compact wave-subchunk grid so pass B can
+// consume the counts directly.
+// ---------------------------------------------------------------------------
+template <bool HAS_G>
+__global__ void ef_count_kernel(float* __restrict__ t, float* __restrict__ r,
+                                const short* __restrict__ g, int64_t n,
+                                TauSet taus, int64_t chunk,
+                                int* __restrict__ wave_counts, int nblocks) {
+    int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    int64_t sub = chunk / WAVES_PER_BLOCK;  // multiple of 512
+    int64_t start = (int64_t)blockIdx.x * chunk + (int64_t)wave * sub;
+    int64_t end = start + sub;
+    if (start > n) start = n;
+    if (end > n) end = n;
+    int cnt[8];
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) cnt[c] = 0;
+
+    int64_t vend = start + ((end - start) & ~7LL);
+    float4* t4 = reinterpret_cast<float4*>(t + start);
+    float4* r4 = reinterpret_cast<float4*>(r + start);
+    const bf16x8_t* g8 = reinterpret_cast<const bf16x8_t*>(g + start);
+    int64_t ng = (vend - start) >> 3;
+    for (int64_t i = lane; i < ng; i += 64) {
+        float4 ra = r4[2 * i], rb = r4[2 * i + 1];
+        float4 ta, tb;
+        if (HAS_G) {
+            bf16x8_t gv = g8[i];
+            ta.x = bf16bits_to_f32(gv[0]) + ra.x;
+            ta.y = bf16bits_to_f32(gv[1]) + ra.y;
+            ta.z = bf16bits_to_f32(gv[2]) + ra.z;
+            ta.w = bf16bits_to_f32(gv[3]) + ra.w;
+            tb.x = bf16bits_to_f32(gv[4]) + rb.x;
+            tb.y = bf16bits_to_f32(gv[5]) + rb.y;
+            tb.z = bf16bits_to_f32(gv[6]) + rb.z;
+            tb.w = bf16bits_to_f32(gv[7]) + rb.w;
+        } else {
+            float4 ua = t4[2 * i], ub = t4[2 * i + 1];
+            ta.x = ua.x + ra.x; ta.y = ua.y + ra.y;
+            ta.z = ua.z + ra.z; ta.w = ua.w + ra.w;
+            tb.x = ub.x + rb.x; tb.y = ub.y + rb.y;
+            tb.z = ub.z + rb.z; tb.w = ub.w + rb.w;
+        }
+        t4[2 * i] = ta; t4[2 * i + 1] = tb;
+        r4[2 * i] = ta; r4[2 * i + 1] = tb;
+        uint32_t a[8] = {abs_bits(ta.x), abs_bits(ta.y), abs_bits(ta.z),
+                         abs_bits(ta.w), abs_bits(tb.x), abs_bits(tb.y),
+                         abs_bits(tb.z), abs_bits(tb.w)};
+        #pragma unroll
+        for (int c = 0; c < 8; ++c)
+            if (c < taus.n) {
+                int s = 0;
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) s += sel_gt(a[j], taus.tb[c]);
+                cnt[c] += s;
+            }
+    }
+    for (int64_t i = vend + lane; i < end; i += 64) {
+        float v = (HAS_G ? bf16bits_to_f32(g[i]) : t[i]) + r[i];
+        t[i] = v;
+        r[i] = v;
+        uint32_t a = abs_bits(v);
+        #pragma unroll
+        for (int c = 0; c < 8; ++c)
+            if (c < taus.n) cnt[c] += sel_gt(a, taus.tb[c]);
+    }
+    for (int c = 0; c < taus.n; ++c) {
+        int v = cnt[c];
+        for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+        if (lane == 0)
+            wave_counts[((int64_t)c * nblocks + blockIdx.x) * WAVES_PER_BLOCK + wave] = v;
+    }
+}
+
+extern "C" void launch_ef_count(float* t, float* r, const void* g, int64_t n,
+                                const float* taus, int ntau, int64_t chunk,
+                                int nblocks, int* wave_counts,
+                                hipStream_t stream) {
+    TauSet ts;
+    ts.n = ntau;
+    for (int j = 0; j < 8; ++j) ts.tb[j] = j < ntau ? tau_to_bits(taus[j]) : 0;
+    if (g != nullptr)
+        hipLaunchKernelGGL((ef_count_kernel<true>), dim3(nblocks), dim3(BLOCK), 0,
+                           stream, t, r, (const short*)g, n, ts, chunk,
+                           wave_counts, nblocks);
+    else
+        hipLaunchKernelGGL((ef_count_kernel<false>), dim3(nblocks), dim3(BLOCK), 0,
+                           stream, t, r, nullptr, n, ts, chunk, wave_counts,
+                           nblocks);
+}
+
 __global__ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
                            float* __restrict__ buf, int64_t n, float lr, float mom,
                            float wd, int nesterov, int use_mom) {

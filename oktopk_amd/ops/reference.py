@@ -76,6 +76,19 @@ def compact_adaptive(t: torch.Tensor, taus, hi_limit: int):
     return idx, val, chosen, counts[chosen]
 
 
+def compact_adaptive_ef(t: torch.Tensor, residual: torch.Tensor, grad,
+                        taus, hi_limit: int):
+    """Torch oracle of the fused EF restore + adaptive compaction: restore
+    (with optional bf16 grad upcast), snapshot residual, then the bump-rule
+    compaction of compact_adaptive."""
+    if grad is not None:
+        t.copy_(grad.reshape(-1).to(t.dtype) + residual)
+    else:
+        t.add_(residual)
+    residual.copy_(t)
+    return compact_adaptive(t, taus, hi_limit)
+
+
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
     """dest[idx] += val (duplicate indices accumulate).
 

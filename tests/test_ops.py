@@ -97,3 +97,25 @@ def test_fused_adam_sane():
     R.fused_adam_(p, gr, m, v, lr=0.01, beta1=0.9, beta2=0.999, eps=1e-8, weight_decay=0.0)
     assert (p < 0).all()  # moved against the gradient
     assert torch.allclose(m, torch.full((10,), 0.1))
+
+
+def test_compact_adaptive_ef_matches_unfused():
+    g = torch.Generator().manual_seed(11)
+    for grad_dtype in (None, torch.bfloat16):
+        t = torch.randn(5000, generator=g)
+        r = torch.randn(5000, generator=g) * 0.1
+        grad = None
+        if grad_dtype is not None:
+            grad = torch.randn(5000, generator=g).to(grad_dtype)
+        t2, r2 = t.clone(), r.clone()
+        taus = [0.5, 0.6, 0.8]
+        idx, val, chosen, cnt = R.compact_adaptive_ef(t, r, grad, taus, 400)
+        if grad is not None:
+            t2.copy_(grad.float() + r2)
+        else:
+            t2.add_(r2)
+        r2.copy_(t2)
+        idx2, val2, chosen2, cnt2 = R.compact_adaptive(t2, taus, 400)
+        assert torch.equal(idx, idx2) and torch.equal(val, val2)
+        assert chosen == chosen2 and cnt == cnt2
+        assert torch.equal(t, t2) and torch.equal(r, r2)

@@ -450,3 +450,29 @@ def test_fused_adam_mirror():
     hip().fused_adam_(p_ref, g, m_ref, v_ref, 1e-3, 0.9, 0.999, 1e-6, 0.01)
     assert torch.equal(p, p_ref)
     assert torch.equal(pb, p.to(torch.bfloat16))
+
+
+@pytest.mark.parametrize("n", [1000, 109_482_240 // 16, 1_000_001])
+@pytest.mark.parametrize("with_grad", [False, True])
+def test_compact_adaptive_ef_matches_reference(n, with_grad):
+    g = torch.Generator().manual_seed(n % 97)
+    t_cpu = torch.randn(n, generator=g)
+    r_cpu = torch.randn(n, generator=g) * 0.05
+    grad_cpu = torch.randn(n, generator=g).bfloat16() if with_grad else None
+    k = max(1, n // 1000)
+    tau0 = R.kth_abs_value((grad_cpu.float() + r_cpu) if with_grad else (t_cpu + r_cpu), k)
+    taus = [tau0 * 0.97 * 1.03 ** i for i in range(4)]
+
+    t_g = t_cpu.cuda()
+    r_g = r_cpu.cuda()
+    grad_g = grad_cpu.cuda() if with_grad else None
+    idx, val, chosen, cnt = hip().compact_adaptive_ef(
+        t_g, r_g, grad_g, taus, 4 * k // 3)
+
+    idx2, val2, chosen2, cnt2 = R.compact_adaptive_ef(
+        t_cpu, r_cpu, grad_cpu, taus, 4 * k // 3)
+    assert int(chosen) == chosen2 and int(cnt) == cnt2
+    assert torch.equal(idx.cpu(), idx2)
+    assert torch.allclose(val.cpu(), val2, atol=1e-6)
+    assert torch.allclose(t_g.cpu(), t_cpu, atol=1e-6)
+    assert torch.allclose(r_g.cpu(), r_cpu, atol=1e-6)
