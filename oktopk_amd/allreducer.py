@@ -419,8 +419,10 @@ class AllReducer:
             if exact:
                 gidx, gval = idx, val
             else:
-                keep = val.abs() > st.tau_global
-                gidx, gval = idx[keep], val[keep]
+                # one nonzero() for both gathers (boolean indexing would
+                # run it once per indexed tensor, two host round-trips)
+                sel = (val.abs() > st.tau_global).nonzero(as_tuple=False).squeeze(1)
+                gidx, gval = idx[sel], val[sel]
 
         if P == 1:
             # no communication: the packed wire round-trip is pure overhead
