@@ -113,8 +113,12 @@ def main():
         trainer.opt.reducer.timers = {}
         # under graph replay the engine runs once per step on the flat grad
         # buffer, so the sync is one-per-step and harmless; with live hooks
-        # (capture unavailable) it would serialize per-bucket overlap — skip
-        trainer.opt.reducer.timing_sync = bool(captured)
+        # (capture unavailable) it would serialize per-bucket overlap — skip.
+        # OKTOPK_PHASE_SYNC=0 disables it (phase_ms then absorbs queue depth;
+        # ms_per_step is bracket-timed either way and always honest).
+        trainer.opt.reducer.timing_sync = (
+            bool(captured) and os.environ.get("OKTOPK_PHASE_SYNC", "1") != "0"
+        )
     elapsed = timed_steps(trainer, comm, args.steps)
     ms_per_step = 1000.0 * elapsed / args.steps
     phases = reducer_phase_ms(trainer, args.steps)
