@@ -26,3 +26,25 @@ def test_bench_json_contract():
     assert d["value"] > 0
     assert d["config"]["compressor"] == "oktopk"
     assert d["config"]["speedup_vs_dense"] is not None
+
+
+def test_bench_world2_torchrun():
+    """The driver's multi-GPU invocation shape: torch.distributed.run with
+    --nproc-per-node 2 (gloo on CPU here; RCCL on the GPU box).  Rank 0
+    prints the one JSON line; value aggregates over the world."""
+    env = dict(os.environ, OKTOPK_BACKEND="gloo")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "2961", "bench.py", "--gpus", "2", "--model",
+         "vgg16", "--batch-size", "2", "--steps", "2", "--warmup", "1",
+         "--density", "0.05", "--dense-baseline-steps", "1"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-2000:]
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 4
