@@ -67,14 +67,25 @@ def main():
     bench("fill_sparse_scaled", lambda: H.fill_sparse_scaled_(dest, idx, val, 0.5), 4 * N)
     bench("adam (flat)", lambda: H.fused_adam_(dest, t, m, v, 1e-3, 0.9, 0.999, 1e-6, 0.01), 7 * 4 * N)
     bench("l2norm", lambda: H.l2norm(t), 4 * N)
+    # fused EF+count rows: EF is in-place (t,r grow every call), so the
+    # timed fn must RESET inputs per iteration or the selection density
+    # explodes (rule 25 — this microbench already hit that once, r01-c).
+    # Measure (reset + op) and subtract the separately-measured reset.
     gbf = torch.randn(N, generator=g).bfloat16().cuda()
     taus6 = [tau_01pct * 1.03 ** i for i in range(6)]
-    bench("ef_count fused (fp32)",
-          lambda: H.compact_adaptive_ef(t, r, None, taus6, N),
-          (4 * 4 + 4) * N)  # EF (r/w t,r) + write-pass read
-    bench("ef_count fused (bf16 g)",
-          lambda: H.compact_adaptive_ef(t, r, gbf, taus6, N),
-          (2 + 4 * 3 + 4) * N)
+    reset_ms = timeit(lambda: (t.copy_(t0), r.copy_(r0)), args.iters) * 1000
+
+    def bench_inplace(name, fn, nbytes):
+        ms = timeit(lambda: (t.copy_(t0), r.copy_(r0), fn()),
+                    args.iters) * 1000 - reset_ms
+        rows.append((name, ms, nbytes / GB / (ms / 1000)))
+
+    bench_inplace("ef_count fused (fp32)",
+                  lambda: H.compact_adaptive_ef(t, r, None, taus6, N),
+                  (4 * 4 + 4) * N)  # EF (r/w t,r) + write-pass read
+    bench_inplace("ef_count fused (bf16 g)",
+                  lambda: H.compact_adaptive_ef(t, r, gbf, taus6, N),
+                  (2 + 4 * 3 + 4) * N)
     bench("zero_at 110k", lambda: H.zero_at_(r, idx), 4 * idx.numel())
 
     print(f"{'op':28s} {'ms':>9s} {'GB/s':>8s}")
