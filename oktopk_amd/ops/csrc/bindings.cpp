@@ -32,6 +32,8 @@ void launch_ef_restore(float*, float*, int64_t, hipStream_t);
 void launch_ef_upcast(float*, float*, const void*, int64_t, hipStream_t);
 void launch_ef_count(float*, float*, const void*, int64_t, const float*, int,
                      int64_t, int, int*, hipStream_t);
+void launch_count_totals(const int*, int, int, int64_t*, hipStream_t);
+void launch_scan_offsets(const int*, int, int*, hipStream_t);
 void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
                 hipStream_t);
 void launch_adam(float*, const float*, float*, float*, void*, int64_t, float, float,
@@ -116,23 +118,36 @@ static CompactGeom compact_geom(int64_t n) {
 
 // shared tail: given per-candidate per-wave counts (ntau x nblocks*4 on CPU)
 // and the chosen candidate row, launch the write pass.
+// Offsets stay on the GPU: a device reduction yields the per-candidate
+// totals (the ONLY values the host needs — bump rule + output sizing, 8B
+// each), and a single-block scan turns the chosen row into write offsets
+// in place of the former 196 KB readback + CPU scan + 32 KB upload.
+struct BumpResult { int chosen; int64_t total; };
+
+static BumpResult bump_choose(torch::Tensor counts, int ntau, int nw,
+                              int64_t hi_limit) {
+    auto totals = torch::empty({ntau}, counts.options().dtype(torch::kInt64));
+    launch_count_totals(counts.data_ptr<int>(), nw, ntau,
+                        totals.data_ptr<int64_t>(), cur_stream());
+    auto th = totals.cpu();
+    const int64_t* tp = th.data_ptr<int64_t>();
+    for (int c = 0; c < ntau; ++c)
+        if (c == ntau - 1 || tp[c] <= hi_limit) return {c, tp[c]};
+    return {ntau - 1, tp[ntau - 1]};
+}
+
 static std::vector<torch::Tensor> compact_finish(
-    torch::Tensor t, double tau, const CompactGeom& g,
-    const int* row_counts /* nblocks*4 ints for the chosen tau */) {
+    torch::Tensor t, double tau, const CompactGeom& g, torch::Tensor counts,
+    int chosen, int64_t total) {
     const int nw = g.nblocks * 4;  // WAVES_PER_BLOCK
-    int64_t total = 0;
-    std::vector<int> offs(nw);
-    for (int b = 0; b < nw; ++b) {
-        offs[b] = (int)total;
-        total += row_counts[b];
-    }
     auto idx = torch::empty({total}, t.options().dtype(torch::kInt32));
     auto val = torch::empty({total}, t.options());
     if (total > 0) {
-        auto offsets = torch::from_blob(offs.data(), {nw}, torch::kInt32)
-                           .to(t.device(), /*non_blocking=*/false);
+        auto offs = torch::empty({nw}, t.options().dtype(torch::kInt32));
+        launch_scan_offsets(counts.data_ptr<int>() + (int64_t)chosen * nw, nw,
+                            offs.data_ptr<int>(), cur_stream());
         launch_compact_write(t.data_ptr<float>(), t.numel(), (float)tau, g.chunk,
-                             g.nblocks, offsets.data_ptr<int>(),
+                             g.nblocks, offs.data_ptr<int>(),
                              idx.data_ptr<int32_t>(), val.data_ptr<float>(),
                              cur_stream());
     }
@@ -148,8 +163,8 @@ static std::vector<torch::Tensor> compact_gt(torch::Tensor t, double tau) {
     float tf = (float)tau;
     launch_compact_count_multi(t.data_ptr<float>(), n, &tf, 1, g.chunk, g.nblocks,
                                counts.data_ptr<int>(), cur_stream());
-    auto h = counts.cpu();
-    return compact_finish(t, tau, g, h.data_ptr<int>());
+    auto b = bump_choose(counts, 1, g.nblocks * 4, n);
+    return compact_finish(t, tau, g, counts, 0, b.total);
 }
 
 // Fused adaptive-threshold compaction: ONE pass counts all candidate taus
@@ -171,22 +186,10 @@ static std::vector<torch::Tensor> compact_adaptive(
                                t.options().dtype(torch::kInt32));
     launch_compact_count_multi(t.data_ptr<float>(), n, tf, ntau, g.chunk, g.nblocks,
                                counts.data_ptr<int>(), cur_stream());
-    auto h = counts.cpu();
-    const int* hp = h.data_ptr<int>();
-    int chosen = ntau - 1;
-    int64_t chosen_total = 0;
-    for (int c = 0; c < ntau; ++c) {
-        int64_t tot = 0;
-        for (int b = 0; b < nw; ++b) tot += hp[(int64_t)c * nw + b];
-        if (c == ntau - 1 || tot <= hi_limit) {
-            chosen = c;
-            chosen_total = tot;
-            break;
-        }
-    }
-    auto out = compact_finish(t, taus[chosen], g, hp + (int64_t)chosen * nw);
-    out.push_back(torch::tensor((int64_t)chosen));
-    out.push_back(torch::tensor(chosen_total));
+    auto b = bump_choose(counts, ntau, nw, hi_limit);
+    auto out = compact_finish(t, taus[b.chosen], g, counts, b.chosen, b.total);
+    out.push_back(torch::tensor((int64_t)b.chosen));
+    out.push_back(torch::tensor(b.total));
     return out;
 }
 
@@ -225,22 +228,10 @@ static std::vector<torch::Tensor> compact_adaptive_ef(
     launch_ef_count(t.data_ptr<float>(), residual.data_ptr<float>(), gp, n, tf,
                     ntau, g.chunk, g.nblocks, counts.data_ptr<int>(),
                     cur_stream());
-    auto h = counts.cpu();
-    const int* hp = h.data_ptr<int>();
-    int chosen = ntau - 1;
-    int64_t chosen_total = 0;
-    for (int c = 0; c < ntau; ++c) {
-        int64_t tot = 0;
-        for (int b = 0; b < nw; ++b) tot += hp[(int64_t)c * nw + b];
-        if (c == ntau - 1 || tot <= hi_limit) {
-            chosen = c;
-            chosen_total = tot;
-            break;
-        }
-    }
-    auto out = compact_finish(t, taus[chosen], g, hp + (int64_t)chosen * nw);
-    out.push_back(torch::tensor((int64_t)chosen));
-    out.push_back(torch::tensor(chosen_total));
+    auto b = bump_choose(counts, ntau, nw, hi_limit);
+    auto out = compact_finish(t, taus[b.chosen], g, counts, b.chosen, b.total);
+    out.push_back(torch::tensor((int64_t)b.chosen));
+    out.push_back(torch::tensor(b.total));
     return out;
 }
 

@@ -530,6 +530,72 @@ extern "C" void launch_ef_upcast(float* t, float* r, const void* g, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// Device-side reduction + scan of the compact wave counts: the host needs
+// ONLY the per-candidate totals (8 ints) to apply the bump rule and size the
+// output; the nw-entry offset table stays on the GPU (previously a 196 KB
+// readback + CPU scan + 32 KB upload per compaction).
+// ---------------------------------------------------------------------------
+__global__ void count_totals_kernel(const int* __restrict__ wave_counts, int nw,
+                                    int ntau, int64_t* __restrict__ totals) {
+    int c = blockIdx.x;  // one block per candidate
+    if (c >= ntau) return;
+    int64_t s = 0;
+    for (int b = threadIdx.x; b < nw; b += blockDim.x)
+        s += wave_counts[(int64_t)c * nw + b];
+    __shared__ int64_t sh[256];
+    sh[threadIdx.x] = s;
+    __syncthreads();
+    for (int o = 128; o > 0; o >>= 1) {
+        if (threadIdx.x < o) sh[threadIdx.x] += sh[threadIdx.x + o];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) totals[c] = sh[0];
+}
+
+// exclusive scan of one candidate row into offs (single block; nw <= 8192)
+__global__ void scan_offsets_kernel(const int* __restrict__ row, int nw,
+                                    int* __restrict__ offs) {
+    const int PER = 8;  // 1024 threads x 8 = 8192 max entries
+    __shared__ int partial[1024];
+    int tid = threadIdx.x;
+    int base = tid * PER;
+    int loc[PER];
+    int s = 0;
+    #pragma unroll
+    for (int j = 0; j < PER; ++j) {
+        int i = base + j;
+        loc[j] = s;  // exclusive within this thread's run
+        s += i < nw ? row[i] : 0;
+    }
+    partial[tid] = s;
+    __syncthreads();
+    for (int d = 1; d < 1024; d <<= 1) {  // Hillis-Steele over partials
+        int v = tid >= d ? partial[tid - d] : 0;
+        __syncthreads();
+        partial[tid] += v;
+        __syncthreads();
+    }
+    int prefix = tid == 0 ? 0 : partial[tid - 1];
+    #pragma unroll
+    for (int j = 0; j < PER; ++j) {
+        int i = base + j;
+        if (i < nw) offs[i] = prefix + loc[j];
+    }
+}
+
+extern "C" void launch_count_totals(const int* wave_counts, int nw, int ntau,
+                                    int64_t* totals, hipStream_t stream) {
+    hipLaunchKernelGGL(count_totals_kernel, dim3(ntau), dim3(256), 0, stream,
+                       wave_counts, nw, ntau, totals);
+}
+
+extern "C" void launch_scan_offsets(const int* row, int nw, int* offs,
+                                    hipStream_t stream) {
+    hipLaunchKernelGGL(scan_offsets_kernel, dim3(1), dim3(1024), 0, stream, row,
+                       nw, offs);
+}
+
+// ---------------------------------------------------------------------------
 // fused EF restore + compact pass A: one streaming pass does
 //   t = (float)g + r   (or t += r when no bf16 grad),  r = t
 // AND counts |t| > tau_c for up to 8 candidate thresholds in the compact
