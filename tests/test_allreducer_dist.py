@@ -259,3 +259,59 @@ def test_balanced_allgather_world2():
 
 def test_balanced_allgather_world4():
     run_dist(_balanced_equivalence, 4)
+
+
+def _balanced_fp32_wire(rank):
+    """Balanced redistribution with fp32 wire (packed stride 2n) — the
+    packing geometry differs from bf16; both must agree with pad-to-max."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    outs = {}
+    for balanced in (False, True):
+        cfg = EngineConfig(
+            compressor="oktopk", density=DENSITY, wire_dtype="fp32",
+            oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                region_repartition_interval=3,
+                                balanced_allgather=balanced),
+        )
+        eng = AllReducer(Comm(dist.group.WORLD), cfg)
+        acc = []
+        for it in range(6):
+            acc.append(eng.run("w", _grad(rank, it)).clone())
+        outs[balanced] = acc
+    for a, b in zip(outs[False], outs[True]):
+        assert torch.equal(a, b)
+
+
+def test_balanced_fp32_wire_world3():
+    run_dist(_balanced_fp32_wire, 3)
+
+
+def _elastic_shrink_midstream(rank):
+    """set_comm after a shrink: boundaries reset, stream continues finite
+    (reference err_callback -> update_nworker, VGG/dl_trainer.py:472)."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    world = dist.get_world_size()
+    cfg = EngineConfig(compressor="oktopk", density=DENSITY,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(dist.group.WORLD), cfg)
+    for it in range(4):
+        eng.run("w", _grad(rank, it))
+    # shrink to the first half of ranks (gloo new_group)
+    half = max(1, world // 2)
+    group = dist.new_group(ranks=list(range(half)))
+    if rank < half:
+        eng.set_comm(Comm(group))
+        assert eng.states["w"].boundaries is None  # reset on shrink
+        for it in range(4, 8):
+            out = eng.run("w", _grad(rank, it))
+            assert torch.isfinite(out).all()
+
+
+def test_elastic_shrink_midstream_world4():
+    run_dist(_elastic_shrink_midstream, 4)
