@@ -315,3 +315,36 @@ def _elastic_shrink_midstream(rank):
 
 def test_elastic_shrink_midstream_world4():
     run_dist(_elastic_shrink_midstream, 4)
+
+
+def _gtopk_world8(rank):
+    """3-round binomial tree at the 8-GPU xGMI island shape.  gTopK is NOT
+    EF-complete beyond one merge round (entries truncated at intermediate
+    merges are unrecoverable — the reference's add_residuals credits only
+    against the FINAL set, VGG/compression.py:151-160), so the invariant
+    here is: finite, rank-identical results, and the strict-EF subset
+    (entries absent from the final set) is credited."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    cfg = EngineConfig(compressor="gtopk", density=DENSITY,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(dist.group.WORLD), cfg)
+    for it in range(4):
+        out = eng.run("w", _grad(rank, it))
+        assert torch.isfinite(out).all()
+        # all ranks hold the identical broadcast result
+        ref = out.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.equal(out, ref)
+    assert torch.isfinite(eng.states["w"].residual).all()
+
+
+def test_world8_gtopk():
+    run_dist(_gtopk_world8, 8)
+
+
+def test_world8_oktopk():
+    # oktopk IS EF-complete at any world size — full mass invariant
+    run_dist(_mass_conservation, 8, args=("oktopk",))
