@@ -97,6 +97,11 @@ def main():
     from oktopk_amd.comm import init_from_env
 
     comm = init_from_env()
+    if args.gpus > 1 and comm.size == 1:
+        # --gpus N>1 without a torchrun rendezvous would multiply the
+        # reported aggregate by N with only one process doing work
+        sys.exit("bench.py --gpus N>1 must be launched via torch.distributed.run "
+                 "(one rank per GPU); refusing to report an unmeasured aggregate")
     n_gpus = comm.size if comm.size > 1 else args.gpus
     rank = comm.rank
 
