@@ -23,7 +23,30 @@ from typing import List, Optional, Sequence, Tuple
 import torch
 import torch.distributed as dist
 
-__all__ = ["Comm", "init_from_env", "is_initialized"]
+__all__ = ["AsyncResult", "Comm", "init_from_env", "is_initialized"]
+
+
+class AsyncResult:
+    """Handle for an in-flight collective: .wait() blocks (if needed) and
+    returns the same value the synchronous Comm method would.  Holds refs
+    to the input buffers so they outlive the transfer."""
+
+    def __init__(self, work, out, keep=(), finish=None):
+        self._work = work
+        self._out = out
+        self._keep = keep
+        self._finish = finish
+        self._done = False
+
+    def wait(self):
+        if not self._done:
+            if self._work is not None:
+                self._work.wait()
+            if self._finish is not None:
+                self._out = self._finish(self._out)
+            self._keep = ()
+            self._done = True
+        return self._out
 
 
 def is_initialized() -> bool:
@@ -202,6 +225,72 @@ class Comm:
             group=self.group,
         )
         return out
+
+    # -- async variants (docs/overlap_design.md step 1) -------------------
+    # Each returns an AsyncResult whose .wait() yields the same value the
+    # sync method returns.  Enqueue order must be identical on every rank
+    # (deterministic chunk loops); waits may be deferred so later chunks'
+    # collectives progress on the RCCL streams while the host works.
+
+    def alltoall_sizes_async(self, send_sizes: Sequence[int],
+                             device: torch.device) -> "AsyncResult":
+        if self._size == 1:
+            return AsyncResult(None, [int(send_sizes[0])])
+        inp = torch.tensor(list(send_sizes), dtype=torch.int64, device=device)
+        out = torch.empty(self._size, dtype=torch.int64, device=device)
+        work = dist.all_to_all_single(out, inp, group=self.group, async_op=True)
+        return AsyncResult(work, out, keep=(inp,),
+                           finish=lambda o: [int(x) for x in o.cpu()])
+
+    def alltoallv_async(self, send: torch.Tensor, send_splits: Sequence[int],
+                        recv_splits: Sequence[int]) -> "AsyncResult":
+        send = send.reshape(-1).contiguous()
+        if self._size == 1:
+            assert send.numel() == int(send_splits[0])
+            return AsyncResult(None, send.clone())
+        out = torch.empty(int(sum(recv_splits)), dtype=send.dtype,
+                          device=send.device)
+        work = dist.all_to_all_single(
+            out, send,
+            output_split_sizes=[int(x) for x in recv_splits],
+            input_split_sizes=[int(x) for x in send_splits],
+            group=self.group, async_op=True)
+        return AsyncResult(work, out, keep=(send,))
+
+    def allgather_sizes_async(self, n: int, device: torch.device) -> "AsyncResult":
+        if self._size == 1:
+            return AsyncResult(None, torch.tensor([n], dtype=torch.int64))
+        t = torch.tensor([n], dtype=torch.int64, device=device)
+        out = torch.empty(self._size, dtype=torch.int64, device=device)
+        work = dist.all_gather_into_tensor(out, t, group=self.group,
+                                           async_op=True)
+        return AsyncResult(work, out, keep=(t,), finish=lambda o: o.cpu())
+
+    def allgatherv_async(self, t: torch.Tensor,
+                         sizes: Sequence[int]) -> "AsyncResult":
+        """Pad-to-max allgather with KNOWN sizes (get them from
+        allgather_sizes[_async] first); .wait() returns the stripped concat."""
+        flat = t.reshape(-1).contiguous()
+        if self._size == 1:
+            return AsyncResult(None, flat.clone())
+        sizes = [int(s) for s in sizes]
+        mx = max(sizes) if sizes else 0
+        if mx == 0:
+            return AsyncResult(
+                None, torch.empty(0, dtype=flat.dtype, device=flat.device))
+        send = flat
+        if flat.numel() < mx:
+            send = torch.zeros(mx, dtype=flat.dtype, device=flat.device)
+            send[: flat.numel()] = flat
+        out = torch.empty(self._size * mx, dtype=flat.dtype, device=flat.device)
+        work = dist.all_gather_into_tensor(out, send, group=self.group,
+                                           async_op=True)
+        def strip(o):
+            if all(s == mx for s in sizes):
+                return o
+            return torch.cat([o[i * mx : i * mx + sizes[i]]
+                              for i in range(self._size)])
+        return AsyncResult(work, out, keep=(send,), finish=strip)
 
     # -- p2p (pipeline parallelism) --------------------------------------
     def isend(self, t: torch.Tensor, dst: int, tag: int = 0):

@@ -98,3 +98,54 @@ def _ddp_worker(rank):
 
 def test_ddp_alternative_trainer():
     run_dist(_ddp_worker, 2)
+
+
+def _async_matches_sync(rank):
+    """Async Comm variants == sync results; interleaved enqueue of several
+    in-flight collectives drains correctly (docs/overlap_design.md step 1)."""
+    import torch
+    import torch.distributed as dist
+    from oktopk_amd.comm import Comm
+
+    comm = Comm(dist.group.WORLD)
+    P = comm.size
+    dev = comm.device
+
+    send_sizes = [(rank + d) % 5 + 1 for d in range(P)]
+    sync_sizes = comm.alltoall_sizes(send_sizes, dev)
+    a1 = comm.alltoall_sizes_async(send_sizes, dev)
+
+    n = 3 + rank
+    sync_g = comm.allgather_sizes(n, dev)
+    a2 = comm.allgather_sizes_async(n, dev)
+
+    payload = torch.arange(sum(send_sizes), dtype=torch.float32) + 100 * rank
+    a3 = comm.alltoallv_async(payload, send_sizes, sync_sizes)
+
+    v = torch.arange(n, dtype=torch.float32) + 10 * rank
+    a4 = comm.allgatherv_async(v, [int(x) for x in sync_g])
+
+    # waits deferred past all enqueues, resolved out of order
+    got_v, _ = comm.allgatherv(v, sizes=[int(x) for x in sync_g])
+    assert torch.equal(a4.wait(), got_v)
+    sync_p = comm.alltoallv(payload, send_sizes, sync_sizes)
+    assert torch.equal(a3.wait(), sync_p)
+    assert [int(x) for x in a2.wait()] == [int(x) for x in sync_g]
+    assert a1.wait() == sync_sizes
+    assert a1.wait() == sync_sizes  # idempotent
+
+
+def test_async_collectives_world3():
+    run_dist(_async_matches_sync, 3)
+
+
+def test_async_collectives_world1():
+    import torch
+    from oktopk_amd.comm import Comm
+
+    c = Comm(None)
+    assert c.alltoall_sizes_async([4], torch.device("cpu")).wait() == [4]
+    t = torch.arange(4.0)
+    assert torch.equal(c.alltoallv_async(t, [4], [4]).wait(), t)
+    assert torch.equal(
+        c.allgatherv_async(t, [4]).wait(), t)
