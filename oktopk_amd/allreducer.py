@@ -147,8 +147,12 @@ class AllReducer:
         t = tensor.reshape(-1)
         if self.timing_sync and t.is_cuda:
             torch.cuda.synchronize()
-        st = self.state(name, t)
         comp = self.cfg.compressor
+        if comp == "oktopk" and self.cfg.oktopk.pipeline_chunks > 1:
+            # chunked stage-interleaved engine keeps per-chunk states and
+            # skips the full-tensor state below (docs/overlap_design.md)
+            return self._oktopk_chunked(name, tensor, grad_src)
+        st = self.state(name, t)
         ok = self.cfg.oktopk
         fused_ef = comp in ("oktopk", "topkAopt", "topkSA", "gaussiankSA")
         if grad_src is not None and not (
@@ -335,31 +339,7 @@ class AllReducer:
             self._dump_grad_stats(name, it, t, tau)
 
         # --- 2. balanced region repartition (reuses the selection) -------
-        if st.boundaries is None:
-            st.boundaries = self._uniform_boundaries(n)
-        if it % ok.region_repartition_interval == 0 and P > 1:
-            m = idx.numel()
-            # The allreduce below is COLLECTIVE: every rank must reach it on
-            # repartition iterations regardless of its local selection size
-            # (a rank-local `if m >= P: allreduce` would deadlock RCCL when
-            # one rank's selection degenerates). Degenerate ranks contribute
-            # uniform-split quantiles instead.
-            if m >= P:
-                step = m // P
-                pos = torch.arange(1, P, dtype=torch.int64, device=idx.device) * step
-                q_local = idx.long()[pos]
-            else:
-                q_local = torch.arange(1, P, dtype=torch.int64, device=idx.device) * (n // P)
-            q = comm.to_comm(q_local)
-            comm.allreduce_(q)
-            q = (q // P).cpu()
-            b = torch.empty(P + 1, dtype=torch.int64)
-            b[0] = 0
-            b[1:P] = q
-            b[P] = n
-            # guard against degenerate (non-monotone) boundaries
-            if bool((b[1:] >= b[:-1]).all()):
-                st.boundaries = b
+        self._repartition(st, idx, n, it)
         bounds = st.boundaries
         lo, hi = int(bounds[rank]), int(bounds[rank + 1])
 
@@ -464,17 +444,234 @@ class AllReducer:
         result = t  # reuse the gradient storage, reference VGG/allreducer.py:838
         ops.fill_sparse_scaled_(result, g_sel_idx, g_sel_val, 1.0 / P)
 
-        # residual credit: zero residual at locally-sent indices that made
-        # the global top-k (reference intersect1d + update_residuals,
-        # VGG/allreducer.py:844-845,1051-1052).
+        self._residual_credit(st, idx, g_sel_idx, n, t.device)
+        self._time(name, "merge", time.perf_counter() - s4)
+        return result
+
+    def _residual_credit(self, st: TensorState, idx: torch.Tensor,
+                         g_sel_idx: torch.Tensor, n: int, device) -> None:
+        """Zero residual at locally-sent indices that made the global top-k
+        (reference intersect1d + update_residuals,
+        VGG/allreducer.py:844-845,1051-1052)."""
         if idx.numel() and g_sel_idx.numel():
             if st.mask is None or st.mask.numel() != n:
-                st.mask = torch.zeros(n, dtype=torch.bool, device=t.device)
+                st.mask = torch.zeros(n, dtype=torch.bool, device=device)
             st.mask[g_sel_idx.long()] = True
             ops.zero_at_masked_(st.residual, idx, st.mask)
             st.mask[g_sel_idx.long()] = False  # cheap sparse reset
+
+    # -- chunked Ok-Topk (docs/overlap_design.md) ------------------------
+    def _oktopk_chunked(self, name: str, tensor: torch.Tensor,
+                        grad_src: Optional[torch.Tensor]) -> torch.Tensor:
+        """Stage-interleaved Ok-Topk over pipeline_chunks slices.
+
+        Semantics: identical to running C independent engines on the C
+        slices (the VGG reference merges layers into groups and runs the
+        whole pipeline per group, VGG/allreducer.py:272-366) — verified
+        bit-equal in tests.  Collectives are issued chunk-by-chunk in
+        deterministic order with DEFERRED waits (comm.AsyncResult), so on
+        RCCL chunk i's exchange overlaps chunk i+1's host+GPU stages.
+        Opt-in via OkTopkConfig.pipeline_chunks; balanced_allgather and
+        profiling_norm are not supported in this mode.
+        """
+        cfg, ok, comm = self.cfg, self.cfg.oktopk, self.comm
+        if cfg.profiling_norm:
+            raise ValueError("profiling_norm requires pipeline_chunks == 1")
+        P, rank = comm.size, comm.rank
+        t = tensor.reshape(-1)
+        n = t.numel()
+        C = max(1, min(ok.pipeline_chunks, n // 8 or 1))
+        g = grad_src.reshape(-1) if grad_src is not None else None
+
+        meta = self.states.get(name + "/meta")
+        if meta is None:
+            meta = TensorState(residual=t.new_zeros(0))
+            self.states[name + "/meta"] = meta
+        it = meta.counter
+        meta.counter += 1
+        if it < ok.dense_warmup_iters:
+            if g is not None:
+                t.copy_(g.to(t.dtype))
+            self._dense(name, t)
+            return tensor
+
+        base = max(8, (n // C) & ~7)  # slice starts 8-aligned (float4/bf16x8)
+        chunks = []
+        for i in range(C):
+            lo = i * base
+            hi = n if i == C - 1 else min(n, (i + 1) * base)
+            if lo >= hi:
+                break
+            sl = t[lo:hi]
+            st = self.state(f"{name}/c{i}", sl)
+            st.grad_src = g[lo:hi] if g is not None else None
+            chunks.append({"sl": sl, "st": st})
+
+        # --- stage 1: local selection (bulk GPU work, all chunks) --------
+        s0 = time.perf_counter()
+        for c in chunks:
+            st, sl = c["st"], c["sl"]
+            k = c["k"] = self._k(sl.numel())
+            if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
+                self._ef_restore(sl, st)
+                st.tau_local = ops.kth_abs_value(sl, k)
+                c["idx"], c["val"] = ops.compact_gt(sl, st.tau_local)
+            else:
+                taus = [st.tau_local * ok.bump_scale ** j
+                        for j in range(ok.bump_max_loops + 1)]
+                grad_c = st.grad_src
+                st.grad_src = None
+                c["idx"], c["val"], chosen, _ = ops.compact_adaptive_ef(
+                    sl, st.residual, grad_c, taus, 4 * k // 3)
+                st.tau_local = taus[chosen]
+            sel = c["idx"].numel()
+            if sel < ok.local_lo_num * k // ok.local_lo_den:
+                st.tau_local /= ok.scale_local
+            elif sel > ok.local_hi_num * k // ok.local_hi_den:
+                st.tau_local *= ok.scale_local
+        self._time(name, "compress", time.perf_counter() - s0)
+
+        if P == 1:
+            s4 = time.perf_counter()
+            for c in chunks:
+                self._chunk_round2_local(c, it)
+            self._time(name, "merge", time.perf_counter() - s4)
+            return tensor
+
+        # --- stage 2: repartition + round-1 exchange (deferred waits) ----
+        s1 = time.perf_counter()
+        for c in chunks:
+            st, idx = c["st"], c["idx"]
+            nl = c["sl"].numel()
+            self._repartition(st, idx, nl, it)
+            bounds = st.boundaries
+            c["lo_r"], c["hi_r"] = int(bounds[rank]), int(bounds[rank + 1])
+            split_pts = torch.searchsorted(
+                idx.long(), bounds[1:P].to(idx.device)).cpu()
+            cuts = [0] + [int(x) for x in split_pts] + [idx.numel()]
+            c["elem_counts"] = [cuts[j + 1] - cuts[j] for j in range(P)]
+            segs = [self._pack(idx[cuts[j]:cuts[j + 1]],
+                               c["val"][cuts[j]:cuts[j + 1]]) for j in range(P)]
+            c["send"] = comm.to_comm(torch.cat(segs))
+            c["w_sizes"] = comm.alltoall_sizes_async(c["elem_counts"], comm.device)
+        for c in chunks:
+            recv_elems = c["recv_elems"] = c["w_sizes"].wait()
+            c["w_payload"] = comm.alltoallv_async(
+                c["send"],
+                [self._pack_ints(x) for x in c["elem_counts"]],
+                [self._pack_ints(x) for x in recv_elems])
+        self._time(name, "alltoall", time.perf_counter() - s1)
+
+        # --- stage 3: reduce + round-2 select + allgather (deferred) -----
+        s2 = time.perf_counter()
+        for c in chunks:
+            st = c["st"]
+            r_idx, r_val = self._unpack(c["w_payload"].wait(), c["recv_elems"])
+            reduced = torch.zeros(c["hi_r"] - c["lo_r"], dtype=t.dtype,
+                                  device=t.device)
+            if r_idx.numel():
+                ops.scatter_add_(reduced, r_idx.to(t.device) - c["lo_r"],
+                                 r_val.to(t.device))
+            exact = c["exact"] = (
+                it % ok.global_threshold_recompute_interval == 0
+                or st.tau_global <= 0.0)
+            gidx, gval = ops.compact_gt(reduced, 0.0 if exact else st.tau_global)
+            c["gidx"] = gidx + c["lo_r"]
+            c["gval"] = gval
+            c["w_gsizes"] = comm.allgather_sizes_async(gidx.numel(), comm.device)
+        for c in chunks:
+            sizes = c["gsizes"] = [int(x) for x in c["w_gsizes"].wait()]
+            pack = comm.to_comm(self._pack(c["gidx"], c["gval"]))
+            c["w_gather"] = comm.allgatherv_async(
+                pack, [self._pack_ints(x) for x in sizes])
+        self._time(name, "allgather", time.perf_counter() - s2)
+
+        # --- stage 4: merge ----------------------------------------------
+        s4 = time.perf_counter()
+        for c in chunks:
+            st, sl = c["st"], c["sl"]
+            all_idx, all_val = self._unpack(c["w_gather"].wait(), c["gsizes"])
+            all_idx = all_idx.to(t.device)
+            all_val = all_val.to(t.device)
+            if c["exact"]:
+                kk = min(c["k"], all_val.numel())
+                if kk > 0:
+                    top = torch.topk(all_val.abs(), kk, sorted=True)
+                    st.tau_global = float(top.values[-1].item())
+                    g_sel_idx = all_idx[top.indices]
+                    g_sel_val = all_val[top.indices]
+                else:
+                    g_sel_idx, g_sel_val = all_idx, all_val
+            else:
+                g_sel_idx, g_sel_val = all_idx, all_val
+                gsz = g_sel_idx.numel()
+                if gsz < ok.global_lo_num * c["k"] // ok.global_lo_den:
+                    st.tau_global /= ok.scale_global_increase
+                elif gsz > ok.global_hi_num * c["k"] // ok.global_hi_den:
+                    st.tau_global *= ok.scale_global_decrease
+            ops.fill_sparse_scaled_(sl, g_sel_idx, g_sel_val, 1.0 / P)
+            self._residual_credit(st, c["idx"], g_sel_idx, sl.numel(), t.device)
         self._time(name, "merge", time.perf_counter() - s4)
-        return result
+        return tensor
+
+    def _chunk_round2_local(self, c: dict, it: int) -> None:
+        """World-1 round 2 for one chunk (no comm — mirror of the fast path
+        in _oktopk)."""
+        ok = self.cfg.oktopk
+        st, sl = c["st"], c["sl"]
+        idx, val, k = c["idx"], c["val"], c["k"]
+        exact = it % ok.global_threshold_recompute_interval == 0 or st.tau_global <= 0.0
+        if exact:
+            g_sel_idx, g_sel_val = idx, val
+            kk = min(k, val.numel())
+            if kk > 0:
+                top = torch.topk(val.abs(), kk, sorted=True)
+                st.tau_global = float(top.values[-1].item())
+                g_sel_idx = idx[top.indices]
+                g_sel_val = val[top.indices]
+        else:
+            sel = (val.abs() > st.tau_global).nonzero(as_tuple=False).squeeze(1)
+            g_sel_idx, g_sel_val = idx[sel], val[sel]
+            gsz = g_sel_idx.numel()
+            if gsz < ok.global_lo_num * k // ok.global_lo_den:
+                st.tau_global /= ok.scale_global_increase
+            elif gsz > ok.global_hi_num * k // ok.global_hi_den:
+                st.tau_global *= ok.scale_global_decrease
+        ops.fill_sparse_scaled_(sl, g_sel_idx, g_sel_val, 1.0)
+        self._residual_credit(st, idx, g_sel_idx, sl.numel(), sl.device)
+
+    def _repartition(self, st: TensorState, idx: torch.Tensor, n: int,
+                     it: int) -> None:
+        """Balanced region repartition from the local selection's quantiles
+        (reference VGG/allreducer.py:620-651).  The allreduce is COLLECTIVE:
+        every rank must reach it on repartition iterations regardless of its
+        local selection size (a rank-local `if m >= P: allreduce` would
+        deadlock RCCL when one rank's selection degenerates) — degenerate
+        ranks contribute uniform-split quantiles instead."""
+        comm = self.comm
+        P = comm.size
+        ok = self.cfg.oktopk
+        if st.boundaries is None:
+            st.boundaries = self._uniform_boundaries(n)
+        if it % ok.region_repartition_interval != 0 or P <= 1:
+            return
+        m = idx.numel()
+        if m >= P:
+            step = m // P
+            pos = torch.arange(1, P, dtype=torch.int64, device=idx.device) * step
+            q_local = idx.long()[pos]
+        else:
+            q_local = torch.arange(1, P, dtype=torch.int64, device=idx.device) * (n // P)
+        q = comm.to_comm(q_local)
+        comm.allreduce_(q)
+        q = (q // P).cpu()
+        b = torch.empty(P + 1, dtype=torch.int64)
+        b[0] = 0
+        b[1:P] = q
+        b[P] = n
+        # guard against degenerate (non-monotone) boundaries
+        if bool((b[1:] >= b[:-1]).all()):
+            st.boundaries = b
 
     def _balanced_round2(
         self, gidx: torch.Tensor, gval: torch.Tensor

@@ -49,6 +49,13 @@ class OkTopkConfig:
     global_lo_den: int = 3
     global_hi_num: int = 4
     global_hi_den: int = 3
+    # Chunked engine pipeline (docs/overlap_design.md): split the flat
+    # gradient into this many slices, each with independent thresholds /
+    # regions / residual (the VGG reference's per-group processing), and
+    # interleave the stages so chunk i's collectives overlap chunk i+1's
+    # compress on RCCL.  1 = the plain single-pass engine (default; flip
+    # only after the 8-GPU win is measured).
+    pipeline_chunks: int = 1
     # Round-2 load-balanced redistribution (reference
     # BERT/bert/allreducer.py:615-715): re-balance the per-owner survivors
     # into equal ceil(S/P) blocks with one alltoallv, then exchange with a

@@ -348,3 +348,77 @@ def test_world8_gtopk():
 def test_world8_oktopk():
     # oktopk IS EF-complete at any world size — full mass invariant
     run_dist(_mass_conservation, 8, args=("oktopk",))
+
+
+def _chunked_equivalence(rank):
+    """pipeline_chunks=C == C independent engines on the C slices
+    (docs/overlap_design.md) — bit-equal outputs and residuals."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    C = 3
+    n = N - 40  # non-divisible: last chunk takes the tail
+    # dense_warmup_iters=0: the warmup dense allreduce is full-tensor in the
+    # chunked engine but per-slice in the reference engines — RCCL/gloo ring
+    # segmentation differs by message size, giving 1-ulp differences that
+    # would break bit-equality (warmup covered by _chunked_warmup below)
+    okc = dict(dense_warmup_iters=0, region_repartition_interval=4)
+    cfg_c = EngineConfig(compressor="oktopk", density=DENSITY,
+                         oktopk=OkTopkConfig(pipeline_chunks=C, **okc))
+    eng_c = AllReducer(Comm(dist.group.WORLD), cfg_c)
+
+    cfg_1 = EngineConfig(compressor="oktopk", density=DENSITY,
+                         oktopk=OkTopkConfig(**okc))
+    engs = [AllReducer(Comm(dist.group.WORLD), cfg_1) for _ in range(C)]
+
+    base = max(8, (n // C) & ~7)
+    bounds = [i * base for i in range(C)] + [n]
+
+    for it in range(ITERS):
+        t = _grad(rank, it, n)
+        ref = t.clone()
+        out_c = eng_c.run("w", t.clone())
+        parts = []
+        for i in range(C):
+            lo, hi = bounds[i], bounds[i + 1]
+            parts.append(engs[i].run("w", ref[lo:hi].clone()))
+        out_ref = torch.cat(parts)
+        assert torch.equal(out_c, out_ref), (it, (out_c - out_ref).abs().max())
+    for i in range(C):
+        lo, hi = bounds[i], bounds[i + 1]
+        assert torch.equal(eng_c.states[f"w/c{i}"].residual,
+                           engs[i].states["w"].residual)
+
+
+def _chunked_warmup(rank):
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    cfg = EngineConfig(compressor="oktopk", density=DENSITY,
+                       oktopk=OkTopkConfig(pipeline_chunks=2,
+                                           dense_warmup_iters=2))
+    eng = AllReducer(Comm(dist.group.WORLD), cfg)
+    world = dist.get_world_size()
+    for it in range(3):
+        out = eng.run("w", _grad(rank, it))
+        if it < 2:  # dense warmup: exact dense mean
+            assert torch.allclose(out, _dense_mean(it, world), atol=1e-5)
+        assert torch.isfinite(out).all()
+
+
+def test_chunked_warmup_world2():
+    run_dist(_chunked_warmup, 2)
+
+
+def test_chunked_equivalence_world1():
+    run_dist(_chunked_equivalence, 1)
+
+
+def test_chunked_equivalence_world2():
+    run_dist(_chunked_equivalence, 2)
+
+
+def test_chunked_equivalence_world4():
+    run_dist(_chunked_equivalence, 4)
