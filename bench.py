@@ -32,6 +32,8 @@ def parse_args():
                    help="dtype of sparse values on the wire (indices stay int32)")
     p.add_argument("--balanced-allgather", action="store_true",
                    help="round-2 load-balanced redistribution (equal AllGather blocks)")
+    p.add_argument("--pipeline-chunks", type=int, default=1,
+                   help="chunked engine pipeline (overlap comm with compress)")
     p.add_argument("--dense-baseline-steps", type=int, default=-1,
                    help="steps for the in-run dense baseline (-1: min(steps,10); 0: skip)")
     return p.parse_args()
@@ -63,7 +65,8 @@ def build_trainer(args, comm, compressor):
     )
     cfg = EngineConfig.preset(preset, compressor=compressor, density=args.density,
                               wire_dtype=args.wire_dtype, dense_warmup_iters=0,
-                              balanced_allgather=args.balanced_allgather)
+                              balanced_allgather=args.balanced_allgather,
+                              pipeline_chunks=args.pipeline_chunks)
     # LSTM recipes run fp32 (MIOpen fused RNN has no bf16 path — 4x
     # slower under autocast, measured); conv/transformer recipes bf16
     dtype = "bf16" if torch.cuda.is_available() else "fp32"

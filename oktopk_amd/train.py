@@ -42,6 +42,8 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--sigma-scale", type=float, default=2.5)
     p.add_argument("--balanced-allgather", action="store_true",
                    help="oktopk round-2 load-balanced redistribution")
+    p.add_argument("--pipeline-chunks", type=int, default=1,
+                   help="chunked engine pipeline (docs/overlap_design.md)")
     p.add_argument("--dense-warmup", type=int, default=None,
                    help="dense allreduce iterations before sparsifying")
     # optimizer
@@ -75,7 +77,8 @@ def main(argv=None) -> int:
               else ("lstm" if args.dnn.startswith("lstm") else "vgg"))
     overrides = dict(compressor=args.compressor, density=args.density,
                      profiling=args.profiling, profiling_norm=args.profiling_norm,
-                     balanced_allgather=args.balanced_allgather)
+                     balanced_allgather=args.balanced_allgather,
+                     pipeline_chunks=args.pipeline_chunks)
     if args.dense_warmup is not None:
         overrides["dense_warmup_iters"] = args.dense_warmup
     cfg = EngineConfig.preset(preset, **overrides)

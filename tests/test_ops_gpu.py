@@ -476,3 +476,28 @@ def test_compact_adaptive_ef_matches_reference(n, with_grad):
     assert torch.allclose(val.cpu(), val2, atol=1e-6)
     assert torch.allclose(t_g.cpu(), t_cpu, atol=1e-6)
     assert torch.allclose(r_g.cpu(), r_cpu, atol=1e-6)
+
+
+def test_engine_chunked_gpu_matches_cpu_world1():
+    """Chunked engine (pipeline_chunks=3) on GPU vs CPU torch reference."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    def build():
+        return AllReducer(
+            Comm(None),
+            EngineConfig(compressor="oktopk", density=0.01,
+                         oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                             pipeline_chunks=3)),
+        )
+
+    eng_gpu, eng_cpu = build(), build()
+    for it in range(4):
+        g = torch.Generator().manual_seed(it)
+        t = torch.randn(100_001, generator=g)  # odd size: ragged last chunk
+        out_gpu = eng_gpu.run("w", t.cuda()).cpu()
+        out_cpu = eng_cpu.run("w", t.clone())
+        assert torch.allclose(out_gpu, out_cpu, atol=1e-5), it
+    for i in range(3):
+        assert torch.allclose(eng_gpu.states[f"w/c{i}"].residual.cpu(),
+                              eng_cpu.states[f"w/c{i}"].residual, atol=1e-5)
