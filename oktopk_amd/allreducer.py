@@ -799,14 +799,19 @@ class AllReducer:
         it = st.counter
 
         s0 = time.perf_counter()
-        self._ef_restore(t, st)
         if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
+            self._ef_restore(t, st)
             st.tau_local = ops.kth_abs_value(t, k)
             idx, val = ops.compact_gt(t, st.tau_local)
         else:
+            # steady state: EF restore + candidate counting in one pass
+            # (same fusion as oktopk — taus derive from the previous tau)
             taus = [st.tau_local * ok.bump_scale ** i
                     for i in range(ok.bump_max_loops + 1)]
-            idx, val, chosen, _cnt = ops.compact_adaptive(t, taus, 4 * k // 3)
+            grad = st.grad_src.reshape(-1) if st.grad_src is not None else None
+            st.grad_src = None
+            idx, val, chosen, _cnt = ops.compact_adaptive_ef(
+                t, st.residual, grad, taus, 4 * k // 3)
             st.tau_local = taus[chosen]
         sel = idx.numel()
         if sel < ok.local_lo_num * k // ok.local_lo_den:
