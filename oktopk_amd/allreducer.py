@@ -901,8 +901,8 @@ class AllReducer:
         it = st.counter
 
         s0 = time.perf_counter()
-        self._ef_restore(t, st)
         if gaussian:
+            self._ef_restore(t, st)
             tau = _gaussian_threshold(t, self.cfg.density)
             for _ in range(3):
                 cnt = ops.count_gt(t, tau)
@@ -916,12 +916,17 @@ class AllReducer:
             idx, val = ops.compact_gt(t, tau)
         else:
             if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
+                self._ef_restore(t, st)
                 st.tau_local = ops.kth_abs_value(t, k)
                 idx, val = ops.compact_gt(t, st.tau_local)
             else:
+                # fused EF restore + candidate counting (same as oktopk)
                 taus = [st.tau_local * ok.bump_scale ** i
                         for i in range(ok.bump_max_loops + 1)]
-                idx, val, chosen, _cnt = ops.compact_adaptive(t, taus, 4 * k // 3)
+                grad = st.grad_src.reshape(-1) if st.grad_src is not None else None
+                st.grad_src = None
+                idx, val, chosen, _cnt = ops.compact_adaptive_ef(
+                    t, st.residual, grad, taus, 4 * k // 3)
                 st.tau_local = taus[chosen]
             tau = st.tau_local
 
