@@ -501,3 +501,28 @@ def test_engine_chunked_gpu_matches_cpu_world1():
     for i in range(3):
         assert torch.allclose(eng_gpu.states[f"w/c{i}"].residual.cpu(),
                               eng_cpu.states[f"w/c{i}"].residual, atol=1e-5)
+
+
+@pytest.mark.parametrize("comp", ["topkAopt", "topkSA", "gaussiank", "gtopk"])
+def test_engine_baselines_gpu_match_cpu_world1(comp):
+    """Baseline compressors, GPU (HIP kernels incl. fused EF+count where
+    adopted) vs CPU torch reference."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    def build():
+        return AllReducer(
+            Comm(None),
+            EngineConfig(compressor=comp, density=0.01,
+                         oktopk=OkTopkConfig(dense_warmup_iters=0)),
+        )
+
+    eng_gpu, eng_cpu = build(), build()
+    for it in range(4):
+        g = torch.Generator().manual_seed(100 + it)
+        t = torch.randn(200_000, generator=g)
+        out_gpu = eng_gpu.run("w", t.cuda()).cpu()
+        out_cpu = eng_cpu.run("w", t.clone())
+        assert torch.allclose(out_gpu, out_cpu, atol=1e-4), (comp, it)
+    assert torch.allclose(eng_gpu.states["w"].residual.cpu(),
+                          eng_cpu.states["w"].residual, atol=1e-4)
