@@ -204,3 +204,31 @@ def test_lr_schedule_step_decay():
     assert tr.opt.param_groups[0]["lr"] == pytest.approx(0.001)
     tr.set_epoch(5)  # schedules are absolute, not cumulative
     assert tr.opt.param_groups[0]["lr"] == pytest.approx(0.1)
+
+
+def test_chunked_state_roundtrip():
+    """Chunked engine states (w/c{i} + w/meta) survive a state_dict round
+    trip through the optimizer-style rebuild path."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    from oktopk_amd.allreducer import TensorState
+
+    cfg = EngineConfig(compressor="oktopk", density=0.02,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                           pipeline_chunks=2))
+    eng = AllReducer(Comm(None), cfg)
+    g = torch.Generator().manual_seed(5)
+    for it in range(3):
+        eng.run("w", torch.randn(4096, generator=g))
+    saved = {n: s.state_dict() for n, s in eng.states.items()}
+    assert set(saved) == {"w/c0", "w/c1", "w/meta"}
+    assert saved["w/meta"]["counter"] == 3
+
+    eng2 = AllReducer(Comm(None), cfg)
+    for n, sd in saved.items():
+        eng2.states[n] = TensorState(residual=sd["residual"].clone())
+        eng2.states[n].load_state_dict(sd)
+    t = torch.randn(4096, generator=torch.Generator().manual_seed(99))
+    o1 = eng.run("w", t.clone())
+    o2 = eng2.run("w", t.clone())
+    assert torch.equal(o1, o2)
