@@ -57,3 +57,35 @@ def test_compact_matches_reference_fuzz(n, density, seed):
     idx, val = R.compact_gt(t, tau)
     assert (t.abs()[idx.long()] > tau).all()
     assert int((t.abs() > tau).sum()) == idx.numel()
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    n=st.integers(min_value=1, max_value=4096),
+    density=st.floats(min_value=0.001, max_value=1.0),
+    chunks=st.integers(min_value=2, max_value=5),
+    seed=st.integers(min_value=0, max_value=10_000),
+)
+def test_chunked_engine_invariants_world1(n, density, chunks, seed):
+    """Chunked engine keeps the EF mass invariant for any size/density/C
+    (incl. n smaller than the chunk alignment)."""
+    cfg = EngineConfig(compressor="oktopk", density=density,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                           pipeline_chunks=chunks))
+    eng = AllReducer(Comm(None), cfg)
+    g = torch.Generator().manual_seed(seed)
+    total_in = torch.zeros(n)
+    total_out = torch.zeros(n)
+    for _ in range(3):
+        t = torch.randn(n, generator=g)
+        total_in += t
+        out = eng.run("w", t.clone())
+        assert torch.isfinite(out).all()
+        total_out += out
+    parts = [stt.residual for key, stt in
+             sorted(((k, s) for k, s in eng.states.items() if "/c" in k),
+                    key=lambda kv: int(kv[0].rsplit("c", 1)[1]))]
+    res = torch.cat(parts)
+    assert res.numel() == n  # chunks tile the tensor exactly
+    err = (total_out + res - total_in).abs().max().item()
+    assert err <= 1e-4, (n, density, chunks, err)
