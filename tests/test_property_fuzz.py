@@ -11,7 +11,7 @@ from oktopk_amd import AllReducer, Comm, EngineConfig
 from oktopk_amd.config import OkTopkConfig
 
 CONSERVING = ["oktopk", "topkA", "topkAopt", "topkSA", "gtopk", "gaussiank",
-              "gaussiankSA", "dense"]
+              "gaussiankconcat", "gaussiankSA", "dense"]
 
 
 @settings(max_examples=40, deadline=None)
