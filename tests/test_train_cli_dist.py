@@ -31,3 +31,13 @@ def test_cli_train_and_resume(tmp_path):
 
     out2 = _run_cli(tmp, ["--pretrain", ck, "--max-epochs", "1"])
     assert out2.returncode == 0, out2.stderr[-2000:]
+
+
+def test_cli_chunked_balanced_flags(tmp_path):
+    """--pipeline-chunks and --balanced-allgather through the full CLI
+    (2 gloo ranks): both engine modes train end-to-end."""
+    tmp = str(tmp_path)
+    out = _run_cli(tmp, ["--pipeline-chunks", "2"])
+    assert out.returncode == 0, out.stderr[-2000:]
+    out2 = _run_cli(tmp, ["--balanced-allgather"])
+    assert out2.returncode == 0, out2.stderr[-2000:]
