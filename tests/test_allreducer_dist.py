@@ -422,3 +422,40 @@ def test_chunked_equivalence_world2():
 
 def test_chunked_equivalence_world4():
     run_dist(_chunked_equivalence, 4)
+
+
+def _hook_optimizer_chunked(rank):
+    """Hook-driven DistributedOptimizer with pipeline_chunks=2: per-bucket
+    chunked states don't collide, DP replicas stay in sync."""
+    import torch
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig, OkTopkConfig
+    from oktopk_amd.optimizer import DistributedOptimizer
+    import torch.distributed as dist
+
+    torch.manual_seed(7)  # same init everywhere
+    model = torch.nn.Sequential(
+        torch.nn.Linear(64, 128), torch.nn.ReLU(), torch.nn.Linear(128, 10))
+    inner = torch.optim.SGD(model.parameters(), lr=0.05)
+    cfg = EngineConfig(compressor="oktopk", density=0.1,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                           pipeline_chunks=2))
+    opt = DistributedOptimizer(inner, model.named_parameters(),
+                               comm=Comm(dist.group.WORLD), cfg=cfg)
+    g = torch.Generator().manual_seed(100 + rank)  # different data per rank
+    for _ in range(3):
+        opt.zero_grad()
+        x = torch.randn(8, 64, generator=g)
+        model(x).sum().backward()
+        opt.step()
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    assert torch.isfinite(flat).all()
+    ref = flat.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.allclose(flat, ref, atol=1e-6), (flat - ref).abs().max()
+    # chunked states exist per bucket
+    assert any(k.endswith("/c0") for k in opt.reducer.states)
+
+
+def test_hook_optimizer_chunked_world2():
+    run_dist(_hook_optimizer_chunked, 2)
