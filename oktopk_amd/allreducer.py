@@ -1130,12 +1130,16 @@ def _benchmark_main():  # pragma: no cover
     ap.add_argument("--numel", type=int, default=25_000_000)
     ap.add_argument("--density", type=float, default=0.001)
     ap.add_argument("--iters", type=int, default=10)
+    ap.add_argument("--pipeline-chunks", type=int, default=1)
+    ap.add_argument("--balanced-allgather", action="store_true")
     args = ap.parse_args()
 
     comm = init_from_env()
     dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
     cfg = EngineConfig(compressor=args.compressor, density=args.density)
     cfg.oktopk.dense_warmup_iters = 0
+    cfg.oktopk.pipeline_chunks = args.pipeline_chunks
+    cfg.oktopk.balanced_allgather = args.balanced_allgather
     eng = AllReducer(comm, cfg)
     g = torch.Generator().manual_seed(comm.rank)
     t0 = torch.randn(args.numel, generator=g).to(dev)
