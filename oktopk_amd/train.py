@@ -140,6 +140,11 @@ def main(argv=None) -> int:
             for name, phases in red.timers.items():
                 logger.info("timing[%s]: %s", name,
                             {k: round(v, 4) for k, v in phases.items()})
+        if args.profiling and comm.rank == 0:
+            mem = trainer.memory_stats()
+            logger.info("gpu-mem: %s", mem)
+            writer.add_dict({f"mem/{k}": v for k, v in mem.items()},
+                            trainer.iteration)
         if args.checkpoint_dir:
             path = os.path.join(args.checkpoint_dir, f"checkpoint.epoch.{epoch}.pth")
             save_checkpoint(path, trainer.model, trainer.opt, trainer.iteration,

@@ -290,6 +290,20 @@ class Trainer:
             if hook_driven:
                 self.opt.local = False
 
+    def memory_stats(self) -> dict:
+        """GPU memory snapshot in MiB (reference DLTrainer's GPU-mem
+        profiling, VGG/dl_trainer.py — logged per epoch under --profiling).
+        Zeros on CPU."""
+        if self.device.type != "cuda":
+            return {"allocated_mib": 0.0, "max_allocated_mib": 0.0,
+                    "reserved_mib": 0.0}
+        mib = 1024 * 1024
+        return {
+            "allocated_mib": round(torch.cuda.memory_allocated() / mib, 1),
+            "max_allocated_mib": round(torch.cuda.max_memory_allocated() / mib, 1),
+            "reserved_mib": round(torch.cuda.memory_reserved() / mib, 1),
+        }
+
     def set_epoch(self, epoch: int) -> None:
         """Advance the engine's dynamic density schedule (reference
         train_epoch plumbing, VGG/allreducer.py:207-208) and apply the LR

@@ -97,3 +97,13 @@ def test_checkpoint_path_scheme(tmp_path):
 
     p = checkpoint_path(str(tmp_path), "bert", epoch=3, stage=1)
     assert p.endswith("bert/checkpoint.1.pth.tar.epoch.3")
+
+
+def test_trainer_memory_stats_cpu():
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.trainer import Trainer
+
+    tr = Trainer(model_name="resnet20", batch_size=2, comm=Comm(None), dtype="fp32")
+    m = tr.memory_stats()
+    assert set(m) == {"allocated_mib", "max_allocated_mib", "reserved_mib"}
+    assert all(v == 0.0 for v in m.values())  # CPU
