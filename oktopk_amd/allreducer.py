@@ -103,6 +103,7 @@ class AllReducer:
         self.states: Dict[str, TensorState] = {}
         self.timers: Dict[str, Dict[str, float]] = {}
         self.eps_log: List[Tuple[int, float]] = []
+        self.randk_log: List[Tuple[int, float]] = []
         self.train_epoch = 0  # drives the dynamic density schedule
         # Phase timers are host perf_counter spans around async GPU launches;
         # without a device sync at entry the first host-blocking point inside
@@ -179,7 +180,7 @@ class AllReducer:
             eps_topk = torch.zeros_like(eps_ref)
             eps_topk[top.indices] = eps_ref[top.indices]
             eps_den = max(ops.l2norm(eps_ref), 1e-30)
-            eps_ref = (eps_topk, eps_den)
+            eps_ref = (eps_topk, eps_den, eps_ref)  # (target, norm, full dense)
 
         if comp in ("none", "dense") or st.counter < ok.dense_warmup_iters:
             out = self._dense(name, t)
@@ -201,9 +202,18 @@ class AllReducer:
             raise AssertionError(comp)
 
         if eps_ref is not None:
-            eps_topk, eps_den = eps_ref
+            eps_topk, eps_den, dense_full = eps_ref
             num = ops.l2norm(out - eps_topk)
             self.eps_log.append((st.counter, num / eps_den))
+            # rand-k baseline for the research dumps (reference's randk norm
+            # arrays, VGG/main_trainer.py:107-138): keeping k RANDOM entries
+            # of the dense mean — the floor any informed selection must beat
+            kk = self._k(t.numel())
+            perm = torch.randperm(t.numel(), device=dense_full.device)[:kk]
+            randk = torch.zeros_like(dense_full)
+            randk[perm] = dense_full[perm]
+            self.randk_log.append(
+                (st.counter, ops.l2norm(randk - eps_topk) / eps_den))
 
         # per-phase timing table (reference prints every 50 iterations,
         # VGG/allreducer.py:379-439)

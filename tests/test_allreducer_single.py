@@ -232,3 +232,21 @@ def test_chunked_state_roundtrip():
     o1 = eng.run("w", t.clone())
     o2 = eng2.run("w", t.clone())
     assert torch.equal(o1, o2)
+
+
+def test_randk_research_log():
+    """profiling_norm also logs the rand-k floor (reference randk norm
+    arrays): informed selection (EPS) must beat random selection."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    torch.manual_seed(0)
+    cfg = EngineConfig(compressor="oktopk", density=0.02, profiling_norm=True,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0))
+    eng = AllReducer(Comm(None), cfg)
+    g = torch.Generator().manual_seed(3)
+    for it in range(4):
+        eng.run("w", torch.randn(8192, generator=g))
+    assert len(eng.randk_log) == 4 and len(eng.eps_log) == 4
+    for (_, eps), (_, rk) in zip(eng.eps_log, eng.randk_log):
+        assert eps < rk  # informed beats random
