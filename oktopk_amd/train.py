@@ -134,6 +134,8 @@ def main(argv=None) -> int:
         if args.profiling_norm and red is not None and comm.rank == 0:
             for step_i, eps in red.eps_log:
                 writer.add_scalar("oracle/eps", eps, step_i)
+            for step_i, rk in red.randk_log:
+                writer.add_scalar("oracle/randk", rk, step_i)
             logger.info("EPS oracle: %d samples, last %.4f",
                         len(red.eps_log), red.eps_log[-1][1] if red.eps_log else -1)
         if args.profiling and red is not None and comm.rank == 0:
