@@ -104,6 +104,7 @@ class AllReducer:
         self.timers: Dict[str, Dict[str, float]] = {}
         self.eps_log: List[Tuple[int, float]] = []
         self.randk_log: List[Tuple[int, float]] = []
+        self.upbound_log: List[Tuple[int, float]] = []
         self.train_epoch = 0  # drives the dynamic density schedule
         # Phase timers are host perf_counter spans around async GPU launches;
         # without a device sync at entry the first host-blocking point inside
@@ -214,6 +215,10 @@ class AllReducer:
             randk[perm] = dense_full[perm]
             self.randk_log.append(
                 (st.counter, ops.l2norm(randk - eps_topk) / eps_den))
+            # upbound: the truncation floor — error of the best possible
+            # k-sparse approximation of the dense mean
+            self.upbound_log.append(
+                (st.counter, ops.l2norm(dense_full - eps_topk) / eps_den))
 
         # per-phase timing table (reference prints every 50 iterations,
         # VGG/allreducer.py:379-439)

@@ -136,6 +136,8 @@ def main(argv=None) -> int:
                 writer.add_scalar("oracle/eps", eps, step_i)
             for step_i, rk in red.randk_log:
                 writer.add_scalar("oracle/randk", rk, step_i)
+            for step_i, ub in red.upbound_log:
+                writer.add_scalar("oracle/upbound", ub, step_i)
             logger.info("EPS oracle: %d samples, last %.4f",
                         len(red.eps_log), red.eps_log[-1][1] if red.eps_log else -1)
         if args.profiling and red is not None and comm.rank == 0:

@@ -248,5 +248,8 @@ def test_randk_research_log():
     for it in range(4):
         eng.run("w", torch.randn(8192, generator=g))
     assert len(eng.randk_log) == 4 and len(eng.eps_log) == 4
-    for (_, eps), (_, rk) in zip(eng.eps_log, eng.randk_log):
-        assert eps < rk  # informed beats random
+    assert len(eng.upbound_log) == 4
+    for (_, eps), (_, rk), (_, ub) in zip(eng.eps_log, eng.randk_log,
+                                          eng.upbound_log):
+        assert eps < rk      # informed beats random
+        assert 0.0 <= ub < 1.0  # truncation floor is a proper fraction
