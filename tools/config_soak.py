@@ -6,9 +6,13 @@ fewer dimensions more deeply.)
     python tools/config_soak.py [--trials 400] [--seed 0]
 """
 import argparse
+import os
 import random
+import sys
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from oktopk_amd import AllReducer, Comm, EngineConfig
 from oktopk_amd.config import OkTopkConfig
