@@ -92,11 +92,18 @@ class _DistributedOptimizer:
         comm: Optional[Comm] = None,
         cfg: Optional[EngineConfig] = None,
         max_grad_norm: float = 0.0,
+        momentum_correction: float = 0.0,
     ):
         self.optimizer = optimizer
         # post-reduce gradient clipping (reference clips for the LSTM recipe,
         # VGG/main_trainer.py:96-99); 0 disables
         self.max_grad_norm = max_grad_norm
+        # pre-reduce momentum (reference momentum_correction,
+        # VGG/distributed_optimizer.py:56,81-88: momentum accumulates on the
+        # LOCAL gradient BEFORE sparsification, and the inner step must then
+        # run with momentum=0).  0 disables (the reference's default too).
+        self.momentum_correction = momentum_correction
+        self._mc_bufs: Dict[str, torch.Tensor] = {}
         self.comm = comm or Comm(None)
         self.cfg = cfg or EngineConfig()
         self.reducer = AllReducer(self.comm, self.cfg)
@@ -148,10 +155,20 @@ class _DistributedOptimizer:
             b = self._param_bucket[p]
             b.pending -= 1
             if b.pending == 0:
+                self._apply_mc(b)
                 self.reducer.run(b.name, b.flat)
                 b.ready = True
 
         return hook
+
+    def _apply_mc(self, b) -> None:
+        if not self.momentum_correction:
+            return
+        buf = self._mc_bufs.get(b.name)
+        if buf is None:
+            buf = self._mc_bufs[b.name] = torch.zeros_like(b.flat)
+        buf.mul_(self.momentum_correction).add_(b.flat)
+        b.flat.copy_(buf)
 
     # -- optimizer surface ----------------------------------------------
     @property
@@ -175,6 +192,7 @@ class _DistributedOptimizer:
         VGG/distributed_optimizer.py:96-105."""
         for b in self.buckets:
             if not b.ready and not self.local:
+                self._apply_mc(b)
                 self.reducer.run(b.name, b.flat)
                 b.ready = True
 
@@ -226,6 +244,7 @@ def DistributedOptimizer(
     is_sparse: Optional[bool] = None,
     density: Optional[float] = None,
     norm_clip: Optional[float] = None,
+    momentum_correction: float = 0.0,
     **_ignored,
 ) -> _DistributedOptimizer:
     """Factory with the reference's calling convention
@@ -239,7 +258,8 @@ def DistributedOptimizer(
     if density is not None:
         cfg.density = density
     return _DistributedOptimizer(optimizer, named_parameters, comm, cfg,
-                                 max_grad_norm=norm_clip or 0.0)
+                                 max_grad_norm=norm_clip or 0.0,
+                                 momentum_correction=momentum_correction)
 
 
 class FlatBertAdam:

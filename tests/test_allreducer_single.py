@@ -253,3 +253,35 @@ def test_randk_research_log():
                                           eng.upbound_log):
         assert eps < rk      # informed beats random
         assert 0.0 <= ub < 1.0  # truncation floor is a proper fraction
+
+
+def test_momentum_correction_matches_sgd_momentum():
+    """momentum_correction (pre-reduce momentum, reference
+    distributed_optimizer.py:56,81-88) with an inner momentum-0 SGD equals
+    plain SGD momentum (dampening 0) on the dense path."""
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.optimizer import DistributedOptimizer
+
+    torch.manual_seed(1)
+    m1 = torch.nn.Linear(16, 8)
+    m2 = torch.nn.Linear(16, 8)
+    m2.load_state_dict(m1.state_dict())
+
+    opt1 = DistributedOptimizer(
+        torch.optim.SGD(m1.parameters(), lr=0.1, momentum=0.0),
+        m1.named_parameters(), comm=Comm(None),
+        cfg=EngineConfig(compressor="dense"), momentum_correction=0.9)
+    opt2 = torch.optim.SGD(m2.parameters(), lr=0.1, momentum=0.9, dampening=0.0)
+
+    g = torch.Generator().manual_seed(2)
+    for _ in range(4):
+        x = torch.randn(4, 16, generator=g)
+        opt1.zero_grad()
+        m1(x).sum().backward()
+        opt1.step()
+        opt2.zero_grad()
+        m2(x).sum().backward()
+        opt2.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
