@@ -161,6 +161,17 @@ class _DistributedOptimizer:
 
         return hook
 
+    # reference API surface (VGG/distributed_optimizer.py:185-201) ------
+    def stop(self) -> None:
+        """Reference stops its background reducer thread; there is none
+        here (hooks fire inline over async RCCL) — kept for API parity."""
+
+    def add_train_epoch(self) -> None:
+        self.reducer.train_epoch += 1
+
+    def get_current_density(self) -> float:
+        return self.reducer.get_current_density()
+
     def _apply_mc(self, b) -> None:
         if not self.momentum_correction:
             return
