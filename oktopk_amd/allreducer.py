@@ -1116,6 +1116,20 @@ class AllReducer:
             )
 
 
+def get_approximate_sigma_scale(density: float) -> float:
+    """Density -> sigma-scale lookup (reference
+    VGG/allreducer.py:460-471).  The engine's gaussian modes compute the
+    threshold exactly via the normal ppf instead (below); this helper is
+    kept for callers that used the reference's approximation."""
+    if density > 0.7:
+        return 0.5
+    if density > 0.05:
+        return 1.5
+    if density > 0.01:
+        return 2.0
+    return 3.0
+
+
 def _gaussian_threshold(t: torch.Tensor, density: float) -> float:
     """Gaussian-fit threshold (reference utils.gen_threshold_from_normal_distribution,
     VGG/utils.py:136-138): right tail ppf of N(mean, std)."""
