@@ -51,10 +51,18 @@ def warmup_linear(x: float, warmup: float = 0.002) -> float:
     return max((x - 1.0) / (warmup - 1.0), 0.0)
 
 
+def warmup_poly(x: float, warmup: float = 0.002, degree: float = 0.5) -> float:
+    # reference optimization.py:55-58
+    if x < warmup:
+        return x / warmup
+    return (1.0 - x) ** degree
+
+
 SCHEDULES = {
     "warmup_cosine": warmup_cosine,
     "warmup_constant": warmup_constant,
     "warmup_linear": warmup_linear,
+    "warmup_poly": warmup_poly,
     "none": lambda x, warmup=0: 1.0,
 }
 

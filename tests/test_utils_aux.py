@@ -107,3 +107,14 @@ def test_trainer_memory_stats_cpu():
     m = tr.memory_stats()
     assert set(m) == {"allocated_mib", "max_allocated_mib", "reserved_mib"}
     assert all(v == 0.0 for v in m.values())  # CPU
+
+
+def test_warmup_schedules_shapes():
+    from oktopk_amd.optimizer import SCHEDULES
+
+    for name, f in SCHEDULES.items():
+        assert 0.0 <= f(0.001) <= 1.0
+        assert 0.0 <= f(0.5) <= 1.0 + 1e-9, name
+    # warmup ramp is linear for all warmup_* schedules
+    assert abs(SCHEDULES["warmup_poly"](0.001) - 0.5) < 1e-9
+    assert abs(SCHEDULES["warmup_linear"](0.002) - 1.0) < 1e-9
