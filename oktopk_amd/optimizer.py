@@ -381,6 +381,13 @@ class FlatBertAdam:
             return self.lr * self.schedule(self.step_count / self.t_total, self.warmup)
         return self.lr
 
+    def get_lr(self) -> list:
+        """Reference BertAdam.get_lr surface (optimization.py:120-133):
+        one scheduled lr per param group — here one flat group."""
+        if self.step_count == 0:
+            return [0]
+        return [self.current_lr()]
+
     def step(self):
         # 1. sparse allreduce of the whole flat gradient
         if self.cfg.compressor in ("dense", "none") and self.model_dtype != torch.float32:
