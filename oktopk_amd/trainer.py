@@ -325,6 +325,20 @@ class Trainer:
             self._base_lr = base = groups[0]["lr"]
         if self.model_name.startswith("bert"):
             return  # BertAdam has its own warmup schedule
+        if self.model_name == "lstman4":
+            # reference _adjust_learning_rate_lstman4: lr /= 1.01 per epoch
+            for g in groups:
+                g["lr"] = base / (1.01 ** epoch)
+            return
+        if self.model_name == "lstm":
+            # reference _adjust_learning_rate_lstmptb (:514-529).  As written
+            # there, first=63 > second=60 makes the 0.1x branch unreachable —
+            # effective schedule: base below 63, x0.01 at 63, x0.001 at 80;
+            # reproduced faithfully.
+            factor = 1.0 if epoch < 63 else (0.01 if epoch < 80 else 0.001)
+            for g in groups:
+                g["lr"] = base * factor
+            return
         milestones = (81, 122) if self.batches.family == "cifar" else (10, 20)
         factor = 0.1 ** sum(1 for m in milestones if epoch >= m)
         for g in groups:

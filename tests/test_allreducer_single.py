@@ -285,3 +285,26 @@ def test_momentum_correction_matches_sgd_momentum():
         opt2.step()
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
+
+
+def test_lr_schedules_per_recipe():
+    """Per-recipe LR schedules (reference _adjust_learning_rate_*,
+    VGG/dl_trainer.py:507-570)."""
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.trainer import Trainer
+
+    tr = Trainer(model_name="lstman4", batch_size=2, comm=Comm(None),
+                 dtype="fp32", lr=1.0)
+    tr.adjust_learning_rate(0)
+    assert abs(tr.opt.param_groups[0]["lr"] - 1.0) < 1e-9
+    tr.adjust_learning_rate(2)
+    assert abs(tr.opt.param_groups[0]["lr"] - 1.0 / 1.01 ** 2) < 1e-9
+
+    tr2 = Trainer(model_name="lstm", batch_size=2, comm=Comm(None),
+                  dtype="fp32", lr=1.0)
+    tr2.adjust_learning_rate(10)
+    assert abs(tr2.opt.param_groups[0]["lr"] - 1.0) < 1e-9
+    tr2.adjust_learning_rate(70)
+    assert abs(tr2.opt.param_groups[0]["lr"] - 0.01) < 1e-9
+    tr2.adjust_learning_rate(85)
+    assert abs(tr2.opt.param_groups[0]["lr"] - 0.001) < 1e-9
