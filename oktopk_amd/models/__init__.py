@@ -8,6 +8,10 @@ from .vgg import VGG, vgg16
 from .lstm import DeepSpeech, deepspeech_an4
 from .bert import BertConfig, BertForPreTraining, bert_base, bert_large
 from .resnet import (
+    preresnet20,
+    preresnet110,
+    resnet_mod20,
+    resnet_mod110,
     resnet20, resnet32, resnet44, resnet56, resnet110,
     resnet18, resnet34, resnet50, resnet101, resnet152,
 )
@@ -26,6 +30,10 @@ _REGISTRY = {
     "resnet44": resnet44,
     "resnet56": resnet56,
     "resnet110": resnet110,
+    "preresnet20": preresnet20,
+    "preresnet110": preresnet110,
+    "resnet_mod20": resnet_mod20,
+    "resnet_mod110": resnet_mod110,
     "alexnet": AlexNet,
     "caffe_cifar": CaffeCifar,
     "densenet": DenseNetCifar,

@@ -157,3 +157,120 @@ def resnet101(**kw):
 
 def resnet152(**kw):
     return ImagenetResNet(Bottleneck, [3, 8, 36, 3], **kw)
+
+
+class _PadChannelDownsample(nn.Module):
+    """Parameter-free shortcut: stride-2 subsample + zero-pad channels
+    (option-A shortcut of the ResNet paper; reference res_utils.DownsampleA)."""
+
+    def __init__(self, stride=2):
+        super().__init__()
+        self.pool = nn.AvgPool2d(1, stride=stride)
+
+    def forward(self, x):
+        x = self.pool(x)
+        return torch.cat([x, x.mul(0.0)], dim=1)
+
+
+class PreActBlock(nn.Module):
+    """Pre-activation basic block (BN-ReLU-conv ordering, He et al. 2016;
+    reference preresnet.py behavior: the FIRST block of a stage pre-activates
+    the shared input, later blocks keep a plain residual)."""
+
+    def __init__(self, cin, cout, stride=1, downsample=None, both_preact=False):
+        super().__init__()
+        self.bn1 = nn.BatchNorm2d(cin)
+        self.conv1 = _conv3x3(cin, cout, stride)
+        self.bn2 = nn.BatchNorm2d(cout)
+        self.conv2 = _conv3x3(cout, cout)
+        self.downsample = downsample
+        self.both_preact = both_preact
+
+    def forward(self, x):
+        pre = F.relu(self.bn1(x), inplace=False)
+        identity = pre if self.both_preact else x
+        if self.downsample is not None:
+            identity = self.downsample(identity)
+        out = self.conv1(pre)
+        out = self.conv2(F.relu(self.bn2(out), inplace=True))
+        return out + identity
+
+
+class CifarPreResNet(nn.Module):
+    """Pre-activation 6n+2 resnet for 32x32 (reference preresnet.py)."""
+
+    def __init__(self, depth: int = 20, num_classes: int = 10):
+        super().__init__()
+        assert (depth - 2) % 6 == 0, "depth must be 6n+2"
+        n = (depth - 2) // 6
+        self.conv1 = _conv3x3(3, 16)
+        self.inplanes = 16
+        self.layer1 = self._make_layer(16, n, 1)
+        self.layer2 = self._make_layer(32, n, 2)
+        self.layer3 = self._make_layer(64, n, 2)
+        self.bn = nn.BatchNorm2d(64)
+        self.fc = nn.Linear(64, num_classes)
+
+    def _make_layer(self, planes, blocks, stride):
+        down = _PadChannelDownsample(stride) if stride != 1 else None
+        layers = [PreActBlock(self.inplanes, planes, stride, down,
+                              both_preact=True)]
+        self.inplanes = planes
+        for _ in range(1, blocks):
+            layers.append(PreActBlock(planes, planes))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = self.conv1(x)
+        x = self.layer3(self.layer2(self.layer1(x)))
+        x = F.relu(self.bn(x), inplace=True)
+        x = F.adaptive_avg_pool2d(x, 1).flatten(1)
+        return self.fc(x)
+
+
+class CifarResNetMod(nn.Module):
+    """fb.resnet.torch-style cifar resnet (reference resnet_mod.py): same
+    6n+2 structure as CifarResNet but with the parameter-free option-A
+    zero-pad shortcut instead of a 1x1-conv projection."""
+
+    def __init__(self, depth: int = 20, num_classes: int = 10):
+        super().__init__()
+        assert (depth - 2) % 6 == 0, "depth must be 6n+2"
+        n = (depth - 2) // 6
+        self.conv1 = _conv3x3(3, 16)
+        self.bn1 = nn.BatchNorm2d(16)
+        self.inplanes = 16
+        self.layer1 = self._make_layer(16, n, 1)
+        self.layer2 = self._make_layer(32, n, 2)
+        self.layer3 = self._make_layer(64, n, 2)
+        self.fc = nn.Linear(64, num_classes)
+
+    def _make_layer(self, planes, blocks, stride):
+        down = _PadChannelDownsample(stride) if stride != 1 else None
+        layers = [BasicBlock(self.inplanes, planes, stride, down)]
+        self.inplanes = planes
+        for _ in range(1, blocks):
+            layers.append(BasicBlock(planes, planes))
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        x = F.relu(self.bn1(self.conv1(x)), inplace=True)
+        x = self.layer3(self.layer2(self.layer1(x)))
+        x = F.adaptive_avg_pool2d(x, 1).flatten(1)
+        return self.fc(x)
+
+
+def preresnet20(**kw):
+    return CifarPreResNet(20, **kw)
+
+
+def preresnet110(**kw):
+    return CifarPreResNet(110, **kw)
+
+
+def resnet_mod20(**kw):
+    return CifarResNetMod(20, **kw)
+
+
+def resnet_mod110(**kw):
+    return CifarResNetMod(110, **kw)

@@ -93,3 +93,21 @@ def test_sdpa_path_matches_manual_cpu(monkeypatch):
     ctx = torch.nn.functional.scaled_dot_product_attention(q[0], q[1], q[2])
     y_sdpa = m.out(ctx.transpose(1, 2).reshape(2, 16, 64))
     assert torch.allclose(y_manual, y_sdpa, atol=1e-5), (y_manual - y_sdpa).abs().max()
+
+
+def test_resnet_variants_forward_backward():
+    """preresnet (pre-activation) + resnet_mod (option-A zero-pad shortcut)
+    variants (reference preresnet.py / resnet_mod.py)."""
+    from oktopk_amd import models
+
+    for name in ("preresnet20", "resnet_mod20"):
+        m = models.create_net(name)
+        y = m(torch.randn(2, 3, 32, 32))
+        assert y.shape == (2, 10)
+        y.sum().backward()
+        assert all(p.grad is not None for p in m.parameters())
+    # option-A shortcut adds no parameters at stage boundaries:
+    # resnet_mod20 must have FEWER params than projection-shortcut resnet20
+    a = sum(p.numel() for p in models.create_net("resnet_mod20").parameters())
+    b = sum(p.numel() for p in models.create_net("resnet20").parameters())
+    assert a < b
