@@ -9,8 +9,14 @@ from .lstm import DeepSpeech, deepspeech_an4
 from .bert import BertConfig, BertForPreTraining, bert_base, bert_large
 from .resnet import (
     preresnet20,
+    preresnet32,
+    preresnet44,
+    preresnet56,
     preresnet110,
     resnet_mod20,
+    resnet_mod32,
+    resnet_mod44,
+    resnet_mod56,
     resnet_mod110,
     resnet20, resnet32, resnet44, resnet56, resnet110,
     resnet18, resnet34, resnet50, resnet101, resnet152,
@@ -31,8 +37,14 @@ _REGISTRY = {
     "resnet56": resnet56,
     "resnet110": resnet110,
     "preresnet20": preresnet20,
+    "preresnet32": preresnet32,
+    "preresnet44": preresnet44,
+    "preresnet56": preresnet56,
     "preresnet110": preresnet110,
     "resnet_mod20": resnet_mod20,
+    "resnet_mod32": resnet_mod32,
+    "resnet_mod44": resnet_mod44,
+    "resnet_mod56": resnet_mod56,
     "resnet_mod110": resnet_mod110,
     "alexnet": AlexNet,
     "caffe_cifar": CaffeCifar,

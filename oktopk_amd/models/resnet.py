@@ -274,3 +274,27 @@ def resnet_mod20(**kw):
 
 def resnet_mod110(**kw):
     return CifarResNetMod(110, **kw)
+
+
+def preresnet32(**kw):
+    return CifarPreResNet(32, **kw)
+
+
+def preresnet44(**kw):
+    return CifarPreResNet(44, **kw)
+
+
+def preresnet56(**kw):
+    return CifarPreResNet(56, **kw)
+
+
+def resnet_mod32(**kw):
+    return CifarResNetMod(32, **kw)
+
+
+def resnet_mod44(**kw):
+    return CifarResNetMod(44, **kw)
+
+
+def resnet_mod56(**kw):
+    return CifarResNetMod(56, **kw)
