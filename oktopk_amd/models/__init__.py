@@ -22,7 +22,7 @@ from .resnet import (
     resnet18, resnet34, resnet50, resnet101, resnet152,
 )
 from .small_cnns import (
-    AlexNet, CaffeCifar, DenseNetCifar, ResNeXtCifar, PTBLSTM,
+    AlexNet, CaffeCifar, DenseNetCifar, MnistNet, ResNeXtCifar, PTBLSTM,
 )
 
 _REGISTRY = {
@@ -47,6 +47,7 @@ _REGISTRY = {
     "resnet_mod56": resnet_mod56,
     "resnet_mod110": resnet_mod110,
     "alexnet": AlexNet,
+    "mnistnet": MnistNet,
     "caffe_cifar": CaffeCifar,
     "densenet": DenseNetCifar,
     "resnext": ResNeXtCifar,

@@ -162,3 +162,25 @@ class PTBLSTM(nn.Module):
         w = next(self.parameters())
         shape = (self.nlayers, bsz, self.hidden_size)
         return (w.new_zeros(shape), w.new_zeros(shape))
+
+
+class MnistNet(nn.Module):
+    """The 2-conv MNIST net the reference trains for its smallest recipe
+    (reference VGG/dl_trainer.py:59-76): conv5x5(1->10) -> pool -> conv5x5
+    (10->20)+dropout2d -> pool -> fc320->50 -> dropout -> fc50->10."""
+
+    def __init__(self, num_classes: int = 10):
+        super().__init__()
+        self.conv1 = nn.Conv2d(1, 10, kernel_size=5)
+        self.conv2 = nn.Conv2d(10, 20, kernel_size=5)
+        self.conv2_drop = nn.Dropout2d()
+        self.fc1 = nn.Linear(320, 50)
+        self.fc2 = nn.Linear(50, num_classes)
+
+    def forward(self, x):
+        x = torch.relu(torch.max_pool2d(self.conv1(x), 2))
+        x = torch.relu(torch.max_pool2d(self.conv2_drop(self.conv2(x)), 2))
+        x = x.flatten(1)
+        x = torch.relu(self.fc1(x))
+        x = nn.functional.dropout(x, training=self.training)
+        return self.fc2(x)

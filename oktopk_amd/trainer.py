@@ -20,7 +20,10 @@ from .optimizer import DistributedOptimizer, FlatBertAdam
 
 
 _CIFAR = {"vgg11", "vgg13", "vgg16", "vgg19", "resnet20", "resnet32", "resnet44",
-          "resnet56", "resnet110", "alexnet", "caffe_cifar", "densenet", "resnext"}
+          "resnet56", "resnet110", "preresnet20", "preresnet32", "preresnet44",
+          "preresnet56", "preresnet110", "resnet_mod20", "resnet_mod32",
+          "resnet_mod44", "resnet_mod56", "resnet_mod110", "alexnet",
+          "caffe_cifar", "densenet", "resnext"}
 _IMAGENET = {"resnet18", "resnet34", "resnet50", "resnet101", "resnet152"}
 
 
@@ -29,6 +32,8 @@ def model_family(name: str) -> str:
         return "cifar"
     if name in _IMAGENET:
         return "imagenet"
+    if name == "mnistnet":
+        return "mnist"
     if name == "lstm":
         return "ptb"
     if name == "lstman4":
@@ -57,6 +62,9 @@ class SyntheticBatches:
         elif self.family == "imagenet":
             self.x = torch.randn(batch_size, 3, 224, 224, generator=g)
             self.y = torch.randint(0, 1000, (batch_size,), generator=g)
+        elif self.family == "mnist":
+            self.x = torch.randn(batch_size, 1, 28, 28, generator=g)
+            self.y = torch.randint(0, 10, (batch_size,), generator=g)
         elif self.family == "ptb":
             # PTB word-LM: (seq, batch) int tokens + shifted targets
             self.tokens = torch.randint(0, 10000, (35, batch_size), generator=g)
@@ -90,7 +98,7 @@ class SyntheticBatches:
         captured hipGraph that reads them stays valid)."""
         self._roll += 1
         g = torch.Generator().manual_seed(99991 * self._roll + 7)
-        if self.family in ("cifar", "imagenet"):
+        if self.family in ("cifar", "imagenet", "mnist"):
             self.x.copy_(torch.randn(self.x.shape, generator=g).to(self.x.device))
             self.y.copy_(torch.randint(0, int(self.y.max().clamp(min=9)) + 1,
                                        self.y.shape, generator=g).to(self.y.device))
@@ -207,7 +215,7 @@ class Trainer:
     # ------------------------------------------------------------------
     def _forward_loss(self) -> torch.Tensor:
         b = self.batches
-        if b.family in ("cifar", "imagenet"):
+        if b.family in ("cifar", "imagenet", "mnist"):
             out = self.model(b.x)
             return torch.nn.functional.cross_entropy(out, b.y)
         if b.family == "ptb":
@@ -354,7 +362,7 @@ class Trainer:
         self.model.eval()
         try:
             b = self.batches
-            if b.family in ("cifar", "imagenet"):
+            if b.family in ("cifar", "imagenet", "mnist"):
                 out = self.model(b.x)
                 top1 = accuracy_topk(out.float(), b.y, (1,))[0]
                 return {"top1": top1}

@@ -111,3 +111,14 @@ def test_resnet_variants_forward_backward():
     a = sum(p.numel() for p in models.create_net("resnet_mod20").parameters())
     b = sum(p.numel() for p in models.create_net("resnet20").parameters())
     assert a < b
+
+
+def test_mnistnet_trains():
+    """mnistnet recipe (reference dl_trainer.py:59-76,87) trains end-to-end
+    on synthetic MNIST-shape batches."""
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.trainer import Trainer
+
+    tr = Trainer(model_name="mnistnet", batch_size=4, comm=Comm(None), dtype="fp32")
+    losses = [tr.step() for _ in range(3)]
+    assert all(torch.isfinite(torch.tensor(losses)))
