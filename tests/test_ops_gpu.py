@@ -526,3 +526,17 @@ def test_engine_baselines_gpu_match_cpu_world1(comp):
         assert torch.allclose(out_gpu, out_cpu, atol=1e-4), (comp, it)
     assert torch.allclose(eng_gpu.states["w"].residual.cpu(),
                           eng_cpu.states["w"].residual, atol=1e-4)
+
+
+def test_model_variants_forward_backward_gpu():
+    """Late-round model additions (preresnet/resnet_mod/mnistnet) run
+    fwd+bwd on the GPU."""
+    from oktopk_amd import models
+
+    cases = [("preresnet20", (2, 3, 32, 32)), ("resnet_mod20", (2, 3, 32, 32)),
+             ("mnistnet", (2, 1, 28, 28))]
+    for name, shape in cases:
+        m = models.create_net(name).cuda()
+        y = m(torch.randn(*shape, device="cuda"))
+        y.sum().backward()
+        assert torch.isfinite(y).all(), name
