@@ -26,7 +26,7 @@ thread-per-tensor-per-peer because gloo CPU sends block
 from __future__ import annotations
 
 from collections import deque
-from typing import List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Sequence, Tuple
 
 import torch
 import torch.distributed as dist
@@ -535,3 +535,27 @@ def make_hybrid_groups(num_stages: int, dp_degree: int):
     prev_rank = rank - dp_degree if stage_id > 0 else None
     next_rank = rank + dp_degree if stage_id < num_stages - 1 else None
     return stage_id, replica_id, Comm(dp_group), prev_rank, next_rank
+
+
+def load_stage_map(path: str) -> Dict[int, List[int]]:
+    """Load a reference-format stage_to_rank_map JSON (the conf files under
+    BERT/bert/tests/depth=N/: {"stage_to_rank_map": {"0": [ranks...], ...}};
+    main_bert.py:881-889 int-keys it the same way).  The exercised reference
+    configs map ALL ranks to stage 0 (pure DP); arbitrary maps partition the
+    world into per-stage DP groups for make_hybrid_groups-style setups."""
+    import json
+
+    with open(path) as f:
+        m = json.load(f)["stage_to_rank_map"]
+    out = {int(k): [int(r) for r in v] for k, v in m.items()}
+    all_ranks = [r for v in out.values() for r in v]
+    if len(all_ranks) != len(set(all_ranks)):
+        raise ValueError("stage_to_rank_map assigns a rank to two stages")
+    return out
+
+
+def stage_of_rank(stage_map: Dict[int, List[int]], rank: int) -> int:
+    for s, ranks in stage_map.items():
+        if rank in ranks:
+            return s
+    raise ValueError(f"rank {rank} not in stage map")

@@ -279,3 +279,22 @@ def _pipeline4_worker(rank):
 
 def test_pipeline_4stage_1f1b_matches_unsplit():
     run_dist(_pipeline4_worker, 4)
+
+
+def test_stage_map_loader(tmp_path):
+    """Reference-format stage_to_rank_map JSON (BERT/bert/tests/depth=4
+    conf files, main_bert.py:881-889)."""
+    import json
+    from oktopk_amd.pipeline import load_stage_map, stage_of_rank
+
+    p = tmp_path / "conf.json"
+    p.write_text(json.dumps(
+        {"stage_to_rank_map": {"0": [0, 1], "1": [2, 3]}}))
+    m = load_stage_map(str(p))
+    assert m == {0: [0, 1], 1: [2, 3]}
+    assert stage_of_rank(m, 2) == 1
+    p2 = tmp_path / "bad.json"
+    p2.write_text(json.dumps({"stage_to_rank_map": {"0": [0], "1": [0]}}))
+    import pytest
+    with pytest.raises(ValueError):
+        load_stage_map(str(p2))
