@@ -459,3 +459,27 @@ def _hook_optimizer_chunked(rank):
 
 def test_hook_optimizer_chunked_world2():
     run_dist(_hook_optimizer_chunked, 2)
+
+
+def _chunked_zero_grads(rank):
+    """Zero gradients through the chunked async pipeline: empty selections
+    on every rank/chunk must not wedge the collectives (0-size
+    all_to_all/all_gather payloads)."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    cfg = EngineConfig(compressor="oktopk", density=0.01,
+                       oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                           pipeline_chunks=3))
+    eng = AllReducer(Comm(dist.group.WORLD), cfg)
+    for it in range(3):
+        out = eng.run("w", torch.zeros(4096))
+        assert torch.equal(out, torch.zeros(4096))
+    # then a real gradient still flows
+    out = eng.run("w", _grad(rank, 9, 4096))
+    assert torch.isfinite(out).all() and out.abs().sum() > 0
+
+
+def test_chunked_zero_grads_world2():
+    run_dist(_chunked_zero_grads, 2)
