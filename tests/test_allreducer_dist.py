@@ -483,3 +483,27 @@ def _chunked_zero_grads(rank):
 
 def test_chunked_zero_grads_world2():
     run_dist(_chunked_zero_grads, 2)
+
+
+def _balanced_tiny_s(rank):
+    """Balanced redistribution when global survivor count S <= P (some
+    balanced blocks empty): still bit-equal to pad-to-max."""
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    import torch.distributed as dist
+
+    outs = {}
+    for bal in (False, True):
+        cfg = EngineConfig(compressor="oktopk", density=0.001,
+                           oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                               balanced_allgather=bal))
+        eng = AllReducer(Comm(dist.group.WORLD), cfg)
+        g0 = torch.zeros(900)
+        g0[rank * 3] = rank + 1.0  # one entry per rank
+        outs[bal] = [eng.run("w", g0.clone()).clone() for _ in range(4)]
+    for a, b in zip(outs[False], outs[True]):
+        assert torch.equal(a, b)
+
+
+def test_balanced_tiny_s_world4():
+    run_dist(_balanced_tiny_s, 4)
