@@ -149,3 +149,48 @@ def test_async_collectives_world1():
     assert torch.equal(c.alltoallv_async(t, [4], [4]).wait(), t)
     assert torch.equal(
         c.allgatherv_async(t, [4]).wait(), t)
+
+
+def _comm_fuzz(rank):
+    """Randomized sizes through alltoallv/allgatherv, sync and async, many
+    rounds — the exact variable-size patterns the engine emits."""
+    import random
+    import torch
+    import torch.distributed as dist
+    from oktopk_amd.comm import Comm
+
+    comm = Comm(dist.group.WORLD)
+    P = comm.size
+    rng = random.Random(42)  # SAME stream on every rank: sizes agree
+    for round_i in range(25):
+        sizes = [[rng.randint(0, 9) for _ in range(P)] for _ in range(P)]
+        send_sizes = sizes[rank]
+        recv_sizes = [sizes[src][rank] for src in range(P)]
+        payload = torch.arange(sum(send_sizes), dtype=torch.float32) + 1000 * rank
+        if round_i % 2 == 0:
+            got = comm.alltoallv(payload, send_sizes, recv_sizes)
+        else:
+            got = comm.alltoallv_async(payload, send_sizes, recv_sizes).wait()
+        # verify contents from each source
+        off = 0
+        for src in range(P):
+            n = recv_sizes[src]
+            seg = got[off:off + n]
+            base = sum(sizes[src][:rank])
+            expect = torch.arange(base, base + n, dtype=torch.float32) + 1000 * src
+            assert torch.equal(seg, expect), (round_i, src)
+            off += n
+
+        n_mine = rng.randint(0, 7) + rank  # rank-dependent but derived from shared rng
+        v = torch.full((n_mine,), float(rank))
+        all_sizes = [int(x) for x in comm.allgather_sizes(n_mine, comm.device)]
+        if round_i % 2 == 0:
+            got2, _ = comm.allgatherv(v, sizes=all_sizes)
+        else:
+            got2 = comm.allgatherv_async(v, all_sizes).wait()
+        expect2 = torch.cat([torch.full((all_sizes[r],), float(r)) for r in range(P)])
+        assert torch.equal(got2, expect2), round_i
+
+
+def test_comm_fuzz_world3():
+    run_dist(_comm_fuzz, 3)
