@@ -26,3 +26,14 @@ def test_task_map():
     assert set(s) == {"acc", "f1"}
     with pytest.raises(ValueError):
         compute_glue_scores("nope", [], [])
+
+
+def test_metric_edge_cases():
+    """Degenerate inputs: constant predictions (MCC denominator 0), perfect
+    F1, anti-correlated pearson."""
+    from oktopk_amd.glue import matthews_corrcoef, f1, pearson
+
+    assert matthews_corrcoef([1, 1, 1, 1], [0, 1, 0, 1]) == 0.0  # undefined -> 0
+    assert f1([1, 0, 1], [1, 0, 1]) == 1.0
+    assert f1([0, 0, 0], [1, 1, 1]) == 0.0
+    assert abs(pearson([1, 2, 3], [3, 2, 1]) + 1.0) < 1e-9
