@@ -48,3 +48,17 @@ def test_bench_world2_torchrun():
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 4
+
+
+def test_bench_dense_compressor_json():
+    """--compressor dense: no dense arm, speedup null, allreduce phase only."""
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--model", "vgg16", "--batch-size", "2",
+         "--steps", "2", "--warmup", "1", "--compressor", "dense"],
+        cwd=REPO, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["config"]["compressor"] == "dense"
+    assert d["config"]["speedup_vs_dense"] is None
+    assert d["config"]["dense_ms_per_step"] is None
