@@ -118,3 +118,17 @@ def test_warmup_schedules_shapes():
     # warmup ramp is linear for all warmup_* schedules
     assert abs(SCHEDULES["warmup_poly"](0.001) - 0.5) < 1e-9
     assert abs(SCHEDULES["warmup_linear"](0.002) - 1.0) < 1e-9
+
+
+def test_flops_counter_vgg16_magnitude():
+    """vgg16 on 32x32: ~0.3 GMac forward (standard figure for CIFAR VGG-16);
+    assert the counter lands in the right decade."""
+    import torch
+    from oktopk_amd import models
+    from oktopk_amd.utils import get_model_complexity_info
+
+    m = models.create_net("vgg16")
+    flops, params = get_model_complexity_info(
+        m, input_constructor=lambda _: torch.randn(1, 3, 32, 32), input_res=None)
+    assert 1e8 < flops < 1e9, flops
+    assert 1e7 < params < 4e7, params  # ~15M conv+fc params
