@@ -41,3 +41,22 @@ def test_cli_chunked_balanced_flags(tmp_path):
     assert out.returncode == 0, out.stderr[-2000:]
     out2 = _run_cli(tmp, ["--balanced-allgather"])
     assert out2.returncode == 0, out2.stderr[-2000:]
+
+
+def test_baseline_config1_vgg_cpu_world2(tmp_path):
+    """BASELINE.json config #1 verbatim: VGG-16 Ok-Topk density=1%, CPU,
+    world_size=2 (plumbing check, no GPU)."""
+    tmp = str(tmp_path)
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29553", "-m", "oktopk_amd.train",
+           "--dnn", "vgg16", "--batch-size", "4", "--density", "0.01",
+           "--compressor", "oktopk", "--iters-per-epoch", "2",
+           "--max-epochs", "1", "--dtype", "fp32",
+           "--logdir", os.path.join(tmp, "logs")]
+    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                         timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    metrics = os.path.join(tmp, "logs", "metrics.jsonl")
+    lines = [json.loads(l) for l in open(metrics)]
+    assert any(d["tag"] == "train/loss" for d in lines)
