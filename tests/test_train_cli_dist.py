@@ -13,6 +13,7 @@ def _run_cli(tmp, extra):
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
            "--master-port", "29549", "-m", "oktopk_amd.train",
            "--dnn", "resnet20", "--batch-size", "2", "--density", "0.05",
+           "--dense-warmup", "0",  # exercise the SPARSE path, not warmup
            "--iters-per-epoch", "3", "--max-epochs", "1", "--dtype", "fp32",
            "--logdir", os.path.join(tmp, "logs"),
            "--checkpoint-dir", os.path.join(tmp, "ck")] + extra
