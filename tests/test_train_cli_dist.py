@@ -52,7 +52,8 @@ def test_baseline_config1_vgg_cpu_world2(tmp_path):
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
            "--master-port", "29553", "-m", "oktopk_amd.train",
            "--dnn", "vgg16", "--batch-size", "4", "--density", "0.01",
-           "--compressor", "oktopk", "--iters-per-epoch", "2",
+           "--compressor", "oktopk", "--dense-warmup", "0",
+           "--iters-per-epoch", "2",
            "--max-epochs", "1", "--dtype", "fp32",
            "--logdir", os.path.join(tmp, "logs")]
     out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
