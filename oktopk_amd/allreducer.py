@@ -1136,8 +1136,10 @@ def _gaussian_threshold(t: torch.Tensor, density: float) -> float:
     from scipy import stats
 
     mean = float(t.mean().item())
-    std = float(t.std().item())
-    if std == 0.0:
+    std = float(t.std().item()) if t.numel() > 1 else 0.0
+    if not math.isfinite(std) or std == 0.0:
+        # degenerate (n<2 or constant tensor): threshold at |mean| selects
+        # nothing strictly above it — EF defers everything, stays finite
         return abs(mean)
     right = stats.norm.ppf(1 - density / 2, loc=mean, scale=std)
     return float(abs(right))
