@@ -16,10 +16,12 @@ from oktopk_amd.config import EngineConfig
 from oktopk_amd.trainer import Trainer
 
 
-def run(compressor, steps, profiling_norm=False, dtype="bf16"):
+def run(compressor, steps, profiling_norm=False, dtype="bf16", density=0.001,
+        chunks=1):
     torch.manual_seed(0)
-    cfg = EngineConfig.preset("bert", compressor=compressor, density=0.001,
-                              dense_warmup_iters=0, profiling_norm=profiling_norm)
+    cfg = EngineConfig.preset("bert", compressor=compressor, density=density,
+                              dense_warmup_iters=0, profiling_norm=profiling_norm,
+                              pipeline_chunks=chunks)
     tr = Trainer("bert_base", batch_size=8, seq_len=128, cfg=cfg, dtype=dtype)
     losses = []
     for i in range(steps):
@@ -35,11 +37,15 @@ def run(compressor, steps, profiling_norm=False, dtype="bf16"):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--steps", type=int, default=150)
+    ap.add_argument("--compressor", default="oktopk")
+    ap.add_argument("--density", type=float, default=0.001)
+    ap.add_argument("--pipeline-chunks", type=int, default=1)
     args = ap.parse_args()
     t0 = time.time()
     dense_losses, _ = run("dense", args.steps)
     dense32_losses, _ = run("dense", args.steps, dtype="fp32")
-    ok_losses, _ = run("oktopk", args.steps)
+    ok_losses, _ = run(args.compressor, args.steps, density=args.density,
+                       chunks=args.pipeline_chunks)
     _, eps = run("oktopk", 10, profiling_norm=True)
     print(f"steps={args.steps} wall={time.time()-t0:.0f}s")
     for i in range(0, args.steps, max(1, args.steps // 15)):
