@@ -31,26 +31,37 @@ def free_port() -> int:
 
 
 def run_dist(fn, world_size: int, args=(), backend: str = "gloo", timeout: float = 180.0):
-    """Spawn `world_size` processes, init gloo on 127.0.0.1, run fn(rank, *args)."""
+    """Spawn `world_size` processes, init gloo on 127.0.0.1, run fn(rank, *args).
+
+    free_port() is racy (the port is released before the children bind, so a
+    concurrent test worker can steal it); a rendezvous-flavored failure gets
+    ONE retry on a fresh port — a genuine test failure still fails twice."""
     import torch.multiprocessing as mp
 
-    port = free_port()
-
-    ctx = mp.get_context("spawn")
-    procs = []
-    for rank in range(world_size):
-        p = ctx.Process(
-            target=_dist_entry, args=(fn, rank, world_size, port, backend, args)
-        )
-        p.start()
-        procs.append(p)
-    for p in procs:
-        p.join(timeout)
-    for rank, p in enumerate(procs):
-        if p.is_alive():
-            p.terminate()
-            raise TimeoutError(f"rank {rank} timed out")
-        assert p.exitcode == 0, f"rank {rank} exited with {p.exitcode}"
+    last_err = None
+    for attempt in range(2):
+        port = free_port()
+        ctx = mp.get_context("spawn")
+        procs = []
+        for rank in range(world_size):
+            p = ctx.Process(
+                target=_dist_entry, args=(fn, rank, world_size, port, backend, args)
+            )
+            p.start()
+            procs.append(p)
+        for p in procs:
+            p.join(timeout)
+        failed = None
+        for rank, p in enumerate(procs):
+            if p.is_alive():
+                p.terminate()
+                failed = TimeoutError(f"rank {rank} timed out")
+            elif p.exitcode != 0 and failed is None:
+                failed = AssertionError(f"rank {rank} exited with {p.exitcode}")
+        if failed is None:
+            return
+        last_err = failed
+    raise last_err
 
 
 def _dist_entry(fn, rank, world, port, backend, args):
