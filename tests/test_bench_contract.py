@@ -5,6 +5,9 @@ import os
 import subprocess
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+from conftest import free_port
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
@@ -36,7 +39,7 @@ def test_bench_world2_torchrun():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "2961", "bench.py", "--gpus", "2", "--model",
+         "--master-port", str(free_port()), "bench.py", "--gpus", "2", "--model",
          "vgg16", "--batch-size", "2", "--steps", "2", "--warmup", "1",
          "--density", "0.05", "--dense-baseline-steps", "1"],
         cwd=REPO, capture_output=True, text=True, timeout=600,

@@ -5,13 +5,16 @@ import os
 import subprocess
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+from conftest import free_port
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def _run_cli(tmp, extra):
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29549", "-m", "oktopk_amd.train",
+           "--master-port", str(free_port()), "-m", "oktopk_amd.train",
            "--dnn", "resnet20", "--batch-size", "2", "--density", "0.05",
            "--dense-warmup", "0",  # exercise the SPARSE path, not warmup
            "--iters-per-epoch", "3", "--max-epochs", "1", "--dtype", "fp32",
@@ -50,7 +53,7 @@ def test_baseline_config1_vgg_cpu_world2(tmp_path):
     tmp = str(tmp_path)
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29553", "-m", "oktopk_amd.train",
+           "--master-port", str(free_port()), "-m", "oktopk_amd.train",
            "--dnn", "vgg16", "--batch-size", "4", "--density", "0.01",
            "--compressor", "oktopk", "--dense-warmup", "0",
            "--iters-per-epoch", "2",
