@@ -492,6 +492,10 @@ class AllReducer:
         cfg, ok, comm = self.cfg, self.cfg.oktopk, self.comm
         if cfg.profiling_norm:
             raise ValueError("profiling_norm requires pipeline_chunks == 1")
+        if ok.balanced_allgather:
+            raise ValueError("balanced_allgather requires pipeline_chunks == 1 "
+                             "(the chunked engine's per-chunk allgathers are "
+                             "already size-balanced by the chunk split)")
         P, rank = comm.size, comm.rank
         t = tensor.reshape(-1)
         n = t.numel()

@@ -101,6 +101,75 @@ def load_tsv_column(path: str, column, has_header: bool = True) -> List[str]:
     return out
 
 
+# -- per-task example loading (reference's DataProcessor subclasses,
+# compute_glue_scores.py:199-524) — one table-driven loader instead of nine
+# near-identical classes.  Columns follow the public GLUE TSV layouts:
+# (text_a_col, text_b_col|None, label_col, labels|None for regression,
+# has_header, test text_a/text_b cols — test TSVs lead with an id column).
+_TASK_SPECS = {
+    "mrpc":  {"a": 3, "b": 4,    "label": 0,  "labels": ["0", "1"],
+              "header": True,  "test_a": 3, "test_b": 4},
+    "mnli":  {"a": 8, "b": 9,    "label": -1,
+              "labels": ["contradiction", "entailment", "neutral"],
+              "header": True,  "test_a": 8, "test_b": 9},
+    "cola":  {"a": 3, "b": None, "label": 1,  "labels": ["0", "1"],
+              "header": False, "test_a": 1, "test_b": None},
+    "sst-2": {"a": 0, "b": None, "label": 1,  "labels": ["0", "1"],
+              "header": True,  "test_a": 1, "test_b": None},
+    "sts-b": {"a": 7, "b": 8,    "label": -1, "labels": None,
+              "header": True,  "test_a": 7, "test_b": 8},
+    "qqp":   {"a": 3, "b": 4,    "label": 5,  "labels": ["0", "1"],
+              "header": True,  "test_a": 1, "test_b": 2},
+    "qnli":  {"a": 1, "b": 2,    "label": -1,
+              "labels": ["entailment", "not_entailment"],
+              "header": True,  "test_a": 1, "test_b": 2},
+    "rte":   {"a": 1, "b": 2,    "label": -1,
+              "labels": ["entailment", "not_entailment"],
+              "header": True,  "test_a": 1, "test_b": 2},
+    "wnli":  {"a": 1, "b": 2,    "label": -1, "labels": ["0", "1"],
+              "header": True,  "test_a": 1, "test_b": 2},
+}
+
+
+def task_labels(task: str):
+    """Label vocabulary for a classification task (None for STS-B
+    regression) — reference get_labels()."""
+    return _TASK_SPECS[task.lower()]["labels"]
+
+
+def load_examples(task: str, tsv_path: str, set_type: str = "dev"
+                  ) -> List[Dict[str, object]]:
+    """Read one GLUE split TSV into [{text_a, text_b, label}] dicts
+    (reference _create_examples of each DataProcessor).  `label` is the
+    label-vocabulary index (classification), a float (sts-b), or None for
+    test splits (unlabeled)."""
+    spec = _TASK_SPECS[task.lower()]
+    test = set_type == "test"
+    a_col = spec["test_a"] if test else spec["a"]
+    b_col = spec["test_b"] if test else spec["b"]
+    out: List[Dict[str, object]] = []
+    with open(tsv_path, newline="", encoding="utf-8") as f:
+        reader = csv.reader(f, delimiter="\t", quotechar=None)
+        if spec["header"] or test:  # test splits always carry a header row
+            next(reader)
+        for line in reader:
+            if len(line) <= max(a_col, b_col or 0):
+                continue
+            label = None
+            if not test:
+                raw = line[spec["label"]]
+                if spec["labels"] is None:
+                    label = float(raw)
+                else:
+                    label = spec["labels"].index(raw)
+            out.append({
+                "text_a": line[a_col],
+                "text_b": line[b_col] if b_col is not None else None,
+                "label": label,
+            })
+    return out
+
+
 def score_files(task: str, pred_file: str, label_file: str,
                 pred_col=-1, label_col=-1) -> Dict[str, float]:
     preds = load_tsv_column(pred_file, pred_col)

@@ -125,3 +125,22 @@ def _dense_fallback_odd_n(rank):
 
 def test_topkSA_dense_fallback_odd_n_world2():
     run_dist(_dense_fallback_odd_n, 2)
+
+
+def test_chunked_rejects_balanced_allgather():
+    """pipeline_chunks>1 + balanced_allgather is an unsupported combination
+    and must fail loudly, mirroring the profiling_norm guard."""
+    import pytest
+
+    from oktopk_amd.allreducer import AllReducer
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig, OkTopkConfig
+
+    eng = AllReducer(
+        Comm(None),
+        EngineConfig(compressor="oktopk",
+                     oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                         pipeline_chunks=2,
+                                         balanced_allgather=True)))
+    with pytest.raises(ValueError, match="balanced_allgather"):
+        eng.run("w", torch.randn(64))

@@ -37,3 +37,52 @@ def test_metric_edge_cases():
     assert f1([1, 0, 1], [1, 0, 1]) == 1.0
     assert f1([0, 0, 0], [1, 1, 1]) == 0.0
     assert abs(pearson([1, 2, 3], [3, 2, 1]) + 1.0) < 1e-9
+
+
+def test_load_examples_per_task(tmp_path):
+    """Table-driven GLUE TSV loaders (reference DataProcessor subclasses,
+    compute_glue_scores.py:199-524): column layout + label vocabulary."""
+    from oktopk_amd.glue import load_examples, task_labels
+
+    # MRPC dev: label, id1, id2, s1, s2 (header row)
+    mrpc = tmp_path / "mrpc_dev.tsv"
+    mrpc.write_text("Quality\t#1 ID\t#2 ID\t#1 String\t#2 String\n"
+                    "1\ta\tb\tfirst sentence\tsecond sentence\n"
+                    "0\tc\td\tthird\tfourth\n")
+    ex = load_examples("mrpc", str(mrpc))
+    assert len(ex) == 2
+    assert ex[0] == {"text_a": "first sentence", "text_b": "second sentence",
+                     "label": 1}
+    assert ex[1]["label"] == 0
+
+    # CoLA dev: no header, cols = source, label, star, sentence
+    cola = tmp_path / "cola_dev.tsv"
+    cola.write_text("src\t1\t*\tthe cat sat\nsrc\t0\t*\tcat the sat the\n")
+    ex = load_examples("cola", str(cola))
+    assert [e["label"] for e in ex] == [1, 0]
+    assert ex[0]["text_a"] == "the cat sat"
+    assert ex[0]["text_b"] is None
+
+    # STS-B: regression label in the last column
+    sts = tmp_path / "sts_dev.tsv"
+    sts.write_text("index\tgenre\tfile\tyear\told1\told2\tsrc\ts1\ts2\tscore\n"
+                   "0\tg\tf\ty\t-\t-\t-\ta sent\tb sent\t3.8\n")
+    ex = load_examples("sts-b", str(sts))
+    assert ex[0]["label"] == 3.8 and ex[0]["text_b"] == "b sent"
+    assert task_labels("sts-b") is None
+
+    # MNLI dev: textual labels -> vocabulary index
+    mnli = tmp_path / "mnli_dev.tsv"
+    header = "\t".join(f"c{i}" for i in range(10)) + "\tgold_label\n"
+    mnli.write_text(header +
+                    "\t".join("x" * 1 for _ in range(8)) +
+                    "\tpremise\thypothesis\tentailment\n")
+    ex = load_examples("mnli", str(mnli))
+    assert ex[0]["label"] == task_labels("mnli").index("entailment")
+
+    # test split: unlabeled, id-leading columns
+    qnli_t = tmp_path / "qnli_test.tsv"
+    qnli_t.write_text("index\tquestion\tsentence\n"
+                      "0\twho?\tthe answer.\n")
+    ex = load_examples("qnli", str(qnli_t), set_type="test")
+    assert ex[0] == {"text_a": "who?", "text_b": "the answer.", "label": None}

@@ -1,0 +1,138 @@
+#!/usr/bin/env python3
+"""CPU convergence evidence: fixed-batch memorization, sparse vs dense.
+
+Each rank trains on ONE fixed synthetic batch (rank-seeded), so the loss must
+drop toward zero and the error-feedback compressors must track the dense
+trajectory (EF defers gradient mass, never loses it — reference
+PROFILING_NORM methodology, VGG/main_trainer.py:107-138).  Runs every
+compressor mode at world 1 and (gloo) world 2, fp32, no GPU needed.
+
+Writes profiles/convergence_cpu.json and prints the trajectory table.
+"""
+import argparse
+import json
+import os
+import socket
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+COMPRESSORS = ["dense", "oktopk", "topkA", "topkA2", "topkAopt", "topkSA",
+               "gtopk", "gaussiank", "gaussiankconcat", "gaussiankSA"]
+
+
+def run_rank(model, compressor, density, steps, record_every, lr=None):
+    """Train on this rank's fixed batch; return sampled losses (must be
+    called with torch.distributed already initialised, or at world 1)."""
+    import torch.distributed as dist
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.trainer import Trainer
+
+    torch.manual_seed(0)
+    comm = Comm(dist.group.WORLD) if dist.is_initialized() else None
+    cfg = EngineConfig.preset("vgg", compressor=compressor, density=density,
+                              dense_warmup_iters=0)
+    tr = Trainer(model, batch_size=32, comm=comm, cfg=cfg, dtype="fp32", lr=lr)
+    out = []
+    for i in range(steps):
+        loss = tr.step()
+        if i % record_every == 0 or i == steps - 1:
+            out.append(round(loss, 4))
+    return out
+
+
+def _child(rank, world, port, model, compressor, density, steps, record_every, lr, q):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        losses = run_rank(model, compressor, density, steps, record_every, lr)
+        if rank == 0:
+            q.put(losses)
+    finally:
+        dist.destroy_process_group()
+
+
+def run(model, compressor, density, steps, world, record_every, lr=None):
+    if world == 1:
+        return run_rank(model, compressor, density, steps, record_every, lr)
+    import torch.multiprocessing as mp
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_child,
+                         args=(r, world, port, model, compressor, density,
+                               steps, record_every, lr, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    losses = q.get(timeout=600)
+    for p in procs:
+        p.join(120)
+    return losses
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="mnistnet")
+    ap.add_argument("--steps", type=int, default=300)
+    ap.add_argument("--density", type=float, default=0.01)
+    ap.add_argument("--lr", type=float, default=None)
+    ap.add_argument("--worlds", default="1,2")
+    ap.add_argument("--compressors", default=",".join(COMPRESSORS))
+    ap.add_argument("--out", default=os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "profiles", "convergence_cpu.json"))
+    args = ap.parse_args()
+
+    record_every = max(1, args.steps // 10)
+    results = {}
+    for world in [int(w) for w in args.worlds.split(",")]:
+        for comp in args.compressors.split(","):
+            losses = run(args.model, comp, args.density, args.steps, world,
+                         record_every, args.lr)
+            key = f"world{world}/{comp}"
+            results[key] = losses
+            print(f"{key:28s} " + " ".join(f"{x:7.3f}" for x in losses))
+
+    # tracking verdict: every sparse run's final loss within the larger of
+    # 0.15 absolute or 2x of dense's final at the same world size
+    verdict = {}
+    for world in [int(w) for w in args.worlds.split(",")]:
+        dense_final = results[f"world{world}/dense"][-1]
+        for comp in args.compressors.split(","):
+            if comp == "dense":
+                continue
+            final = results[f"world{world}/{comp}"][-1]
+            ok = final <= max(dense_final * 2.0, dense_final + 0.15)
+            verdict[f"world{world}/{comp}"] = {"final": final,
+                                               "dense_final": dense_final,
+                                               "tracks_dense": ok}
+    out = {
+        "setup": {"model": args.model, "batch_size": 32, "dtype": "fp32",
+                  "device": "cpu", "density": args.density,
+                  "steps": args.steps, "record_every": record_every,
+                  "lr": args.lr,
+                  "task": "fixed-batch memorization (one rank-seeded batch "
+                          "per rank, loss must drop toward 0)"},
+        "loss_trajectories": results,
+        "verdict": verdict,
+    }
+    with open(args.out, "w") as f:
+        json.dump(out, f, indent=1)
+    bad = [k for k, v in verdict.items() if not v["tracks_dense"]]
+    print(f"\nwrote {args.out}")
+    print("tracking verdict:", "ALL TRACK DENSE" if not bad else f"DIVERGED: {bad}")
+
+
+if __name__ == "__main__":
+    main()
