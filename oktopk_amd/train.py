@@ -39,7 +39,11 @@ def build_argparser() -> argparse.ArgumentParser:
     # sparsification (reference --compressor/--density/--sigma-scale)
     p.add_argument("--compressor", type=str, default="oktopk")
     p.add_argument("--density", type=float, default=0.02)
-    p.add_argument("--sigma-scale", type=float, default=2.5)
+    p.add_argument("--sigma-scale", type=float, default=2.5,
+                   help="accepted for reference-script compatibility; the "
+                        "gaussian modes compute the exact normal-ppf "
+                        "threshold (reference utils.py:136) so the manual "
+                        "sigma-scale approximation is not needed")
     p.add_argument("--balanced-allgather", action="store_true",
                    help="oktopk round-2 load-balanced redistribution")
     p.add_argument("--pipeline-chunks", type=int, default=1,
