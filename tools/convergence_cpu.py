@@ -23,7 +23,7 @@ COMPRESSORS = ["dense", "oktopk", "topkA", "topkA2", "topkAopt", "topkSA",
                "gtopk", "gaussiank", "gaussiankconcat", "gaussiankSA"]
 
 
-def run_rank(model, compressor, density, steps, record_every, lr=None):
+def run_rank(model, compressor, density, steps, record_every, lr=None, bs=32):
     """Train on this rank's fixed batch; return sampled losses (must be
     called with torch.distributed already initialised, or at world 1)."""
     import torch.distributed as dist
@@ -33,9 +33,11 @@ def run_rank(model, compressor, density, steps, record_every, lr=None):
 
     torch.manual_seed(0)
     comm = Comm(dist.group.WORLD) if dist.is_initialized() else None
-    cfg = EngineConfig.preset("vgg", compressor=compressor, density=density,
+    preset = ("lstm" if model.startswith("lstm") else
+              ("bert" if model.startswith("bert") else "vgg"))
+    cfg = EngineConfig.preset(preset, compressor=compressor, density=density,
                               dense_warmup_iters=0)
-    tr = Trainer(model, batch_size=32, comm=comm, cfg=cfg, dtype="fp32", lr=lr)
+    tr = Trainer(model, batch_size=bs, comm=comm, cfg=cfg, dtype="fp32", lr=lr)
     out = []
     for i in range(steps):
         loss = tr.step()
@@ -44,23 +46,23 @@ def run_rank(model, compressor, density, steps, record_every, lr=None):
     return out
 
 
-def _child(rank, world, port, model, compressor, density, steps, record_every, lr, q):
+def _child(rank, world, port, model, compressor, density, steps, record_every, lr, bs, q):
     import torch.distributed as dist
 
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
-        losses = run_rank(model, compressor, density, steps, record_every, lr)
+        losses = run_rank(model, compressor, density, steps, record_every, lr, bs)
         if rank == 0:
             q.put(losses)
     finally:
         dist.destroy_process_group()
 
 
-def run(model, compressor, density, steps, world, record_every, lr=None):
+def run(model, compressor, density, steps, world, record_every, lr=None, bs=32):
     if world == 1:
-        return run_rank(model, compressor, density, steps, record_every, lr)
+        return run_rank(model, compressor, density, steps, record_every, lr, bs)
     import torch.multiprocessing as mp
 
     s = socket.socket()
@@ -71,11 +73,11 @@ def run(model, compressor, density, steps, world, record_every, lr=None):
     q = ctx.Queue()
     procs = [ctx.Process(target=_child,
                          args=(r, world, port, model, compressor, density,
-                               steps, record_every, lr, q))
+                               steps, record_every, lr, bs, q))
              for r in range(world)]
     for p in procs:
         p.start()
-    losses = q.get(timeout=600)
+    losses = q.get(timeout=1800)
     for p in procs:
         p.join(120)
     return losses
@@ -87,6 +89,7 @@ def main():
     ap.add_argument("--steps", type=int, default=300)
     ap.add_argument("--density", type=float, default=0.01)
     ap.add_argument("--lr", type=float, default=None)
+    ap.add_argument("--batch-size", type=int, default=32)
     ap.add_argument("--worlds", default="1,2")
     ap.add_argument("--compressors", default=",".join(COMPRESSORS))
     ap.add_argument("--out", default=os.path.join(
@@ -104,7 +107,7 @@ def main():
         run_steps = prev.get("run_steps", {})
     except (OSError, ValueError):
         results, run_steps = {}, {}
-    setup = {"model": args.model, "batch_size": 32, "dtype": "fp32",
+    setup = {"model": args.model, "batch_size": args.batch_size, "dtype": "fp32",
              "device": "cpu", "density": args.density,
              "lr": args.lr,
              "task": "fixed-batch memorization (one rank-seeded batch "
@@ -112,7 +115,7 @@ def main():
     for world in [int(w) for w in args.worlds.split(",")]:
         for comp in args.compressors.split(","):
             losses = run(args.model, comp, args.density, args.steps, world,
-                         record_every, args.lr)
+                         record_every, args.lr, args.batch_size)
             key = f"world{world}/{comp}"
             results[key] = losses
             run_steps[key] = args.steps
