@@ -23,7 +23,8 @@ COMPRESSORS = ["dense", "oktopk", "topkA", "topkA2", "topkAopt", "topkSA",
                "gtopk", "gaussiank", "gaussiankconcat", "gaussiankSA"]
 
 
-def run_rank(model, compressor, density, steps, record_every, lr=None, bs=32):
+def run_rank(model, compressor, density, steps, record_every, lr=None, bs=32,
+             seq_len=128, model_kwargs=None):
     """Train on this rank's fixed batch; return sampled losses (must be
     called with torch.distributed already initialised, or at world 1)."""
     import torch.distributed as dist
@@ -37,7 +38,8 @@ def run_rank(model, compressor, density, steps, record_every, lr=None, bs=32):
               ("bert" if model.startswith("bert") else "vgg"))
     cfg = EngineConfig.preset(preset, compressor=compressor, density=density,
                               dense_warmup_iters=0)
-    tr = Trainer(model, batch_size=bs, comm=comm, cfg=cfg, dtype="fp32", lr=lr)
+    tr = Trainer(model, batch_size=bs, seq_len=seq_len, comm=comm, cfg=cfg,
+                 dtype="fp32", lr=lr, model_kwargs=model_kwargs)
     out = []
     for i in range(steps):
         loss = tr.step()
@@ -46,23 +48,27 @@ def run_rank(model, compressor, density, steps, record_every, lr=None, bs=32):
     return out
 
 
-def _child(rank, world, port, model, compressor, density, steps, record_every, lr, bs, q):
+def _child(rank, world, port, model, compressor, density, steps, record_every,
+           lr, bs, seq_len, model_kwargs, q):
     import torch.distributed as dist
 
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
-        losses = run_rank(model, compressor, density, steps, record_every, lr, bs)
+        losses = run_rank(model, compressor, density, steps, record_every, lr, bs,
+                          seq_len, model_kwargs)
         if rank == 0:
             q.put(losses)
     finally:
         dist.destroy_process_group()
 
 
-def run(model, compressor, density, steps, world, record_every, lr=None, bs=32):
+def run(model, compressor, density, steps, world, record_every, lr=None, bs=32,
+        seq_len=128, model_kwargs=None):
     if world == 1:
-        return run_rank(model, compressor, density, steps, record_every, lr, bs)
+        return run_rank(model, compressor, density, steps, record_every, lr, bs,
+                        seq_len, model_kwargs)
     import torch.multiprocessing as mp
 
     s = socket.socket()
@@ -73,7 +79,8 @@ def run(model, compressor, density, steps, world, record_every, lr=None, bs=32):
     q = ctx.Queue()
     procs = [ctx.Process(target=_child,
                          args=(r, world, port, model, compressor, density,
-                               steps, record_every, lr, bs, q))
+                               steps, record_every, lr, bs,
+                               seq_len, model_kwargs, q))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -90,6 +97,10 @@ def main():
     ap.add_argument("--density", type=float, default=0.01)
     ap.add_argument("--lr", type=float, default=None)
     ap.add_argument("--batch-size", type=int, default=32)
+    ap.add_argument("--seq-len", type=int, default=128)
+    ap.add_argument("--model-kwargs", default="",
+                    help='JSON dict of model-config overrides, e.g. a tiny '
+                         'BERT: {"num_hidden_layers":2,"hidden_size":128}')
     ap.add_argument("--worlds", default="1,2")
     ap.add_argument("--compressors", default=",".join(COMPRESSORS))
     ap.add_argument("--out", default=os.path.join(
@@ -97,6 +108,7 @@ def main():
         "profiles", "convergence_cpu.json"))
     args = ap.parse_args()
 
+    mkw = json.loads(args.model_kwargs) if args.model_kwargs else None
     record_every = max(1, args.steps // 10)
     # merge into any existing evidence file and write INCREMENTALLY after
     # every run, so a killed/timed-out sweep keeps its completed runs
@@ -107,7 +119,8 @@ def main():
         run_steps = prev.get("run_steps", {})
     except (OSError, ValueError):
         results, run_steps = {}, {}
-    setup = {"model": args.model, "batch_size": args.batch_size, "dtype": "fp32",
+    setup = {"model": args.model, "batch_size": args.batch_size,
+             "seq_len": args.seq_len, "model_kwargs": mkw, "dtype": "fp32",
              "device": "cpu", "density": args.density,
              "lr": args.lr,
              "task": "fixed-batch memorization (one rank-seeded batch "
@@ -115,7 +128,8 @@ def main():
     for world in [int(w) for w in args.worlds.split(",")]:
         for comp in args.compressors.split(","):
             losses = run(args.model, comp, args.density, args.steps, world,
-                         record_every, args.lr, args.batch_size)
+                         record_every, args.lr, args.batch_size,
+                         args.seq_len, mkw)
             key = f"world{world}/{comp}"
             results[key] = losses
             run_steps[key] = args.steps
@@ -140,8 +154,11 @@ def main():
             if not key.startswith(w + "/") or key == dense_key:
                 continue
             final = losses[-1]
-            ok = final < 1.0 and final <= max(dense_final * 4.0,
-                                              dense_final + 0.3)
+            # scale-free: clear descent from the untrained start AND within
+            # a generous band of dense's floor (EF lags 2-3x in steps)
+            descended = final <= 0.4 * losses[0]
+            ok = descended and final <= max(dense_final * 4.0,
+                                            dense_final + 0.3)
             verdict[key] = {"final": final, "dense_final": dense_final,
                             "tracks_dense": ok}
     with open(args.out, "w") as f:
