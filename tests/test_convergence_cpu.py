@@ -34,3 +34,27 @@ def test_sparse_descends_like_dense(compressor):
     # must be well below the untrained plateau (2.30) and still descending
     assert min(losses[-30:]) < 1.0, f"{compressor} stuck: tail={losses[-5:]}"
     assert min(losses[-30:]) < min(losses[:30]) / 2
+
+
+def test_flat_adam_engine_descends_like_dense():
+    """The FLAT engine path (FlatBertAdam: bulk EF restore + fused Adam —
+    the bench flagship's optimizer) on a tiny BERT: oktopk must track the
+    dense trajectory (full matrix: profiles/convergence_cpu_bert_tiny.json)."""
+    kw = dict(num_hidden_layers=2, hidden_size=128, num_attention_heads=4,
+              intermediate_size=512)
+
+    def losses(compressor, steps=120):
+        torch.manual_seed(0)
+        cfg = EngineConfig.preset("bert", compressor=compressor,
+                                  density=0.01, dense_warmup_iters=0)
+        tr = Trainer("bert_base", batch_size=8, seq_len=64, cfg=cfg,
+                     dtype="fp32", model_kwargs=kw)
+        from oktopk_amd.optimizer import FlatBertAdam
+        assert isinstance(tr.opt, FlatBertAdam)
+        return [tr.step() for _ in range(steps)]
+
+    dense = losses("dense")
+    ok = losses("oktopk")
+    assert dense[-1] < 0.4 * dense[0]
+    assert ok[-1] < 0.4 * ok[0]
+    assert ok[-1] <= max(4 * dense[-1], dense[-1] + 0.3), (ok[-5:], dense[-5:])
