@@ -151,9 +151,15 @@ def load_examples(task: str, tsv_path: str, set_type: str = "dev"
     with open(tsv_path, newline="", encoding="utf-8") as f:
         reader = csv.reader(f, delimiter="\t", quotechar=None)
         if spec["header"] or test:  # test splits always carry a header row
-            next(reader)
+            next(reader, None)
         for line in reader:
-            if len(line) <= max(a_col, b_col or 0):
+            # guard every column we will index, including the label column —
+            # real QQP train.tsv contains short malformed rows (the reference
+            # QqpProcessor wraps row access in try/except IndexError)
+            need = max(a_col, b_col or 0)
+            if not test:
+                need = max(need, spec["label"])
+            if len(line) <= need:
                 continue
             label = None
             if not test:

@@ -86,3 +86,23 @@ def test_load_examples_per_task(tmp_path):
                       "0\twho?\tthe answer.\n")
     ex = load_examples("qnli", str(qnli_t), set_type="test")
     assert ex[0] == {"text_a": "who?", "text_b": "the answer.", "label": None}
+
+
+def test_load_examples_malformed_rows(tmp_path):
+    """Short rows are skipped, not crashed on — real QQP train.tsv contains
+    rows shorter than the label column (reference QqpProcessor wraps row
+    access in try/except IndexError); empty files must not StopIteration."""
+    from oktopk_amd.glue import load_examples
+
+    # QQP layout: id, qid1, qid2, question1, question2, is_duplicate(=col 5).
+    # Row 2 has the text columns but NOT the label column.
+    qqp = tmp_path / "qqp_train.tsv"
+    qqp.write_text("id\tqid1\tqid2\tquestion1\tquestion2\tis_duplicate\n"
+                   "0\ta\tb\tq one\tq two\t1\n"
+                   "1\tc\td\tshort row q1\tshort row q2\n")
+    ex = load_examples("qqp", str(qqp), set_type="train")
+    assert len(ex) == 1 and ex[0]["label"] == 1
+
+    empty = tmp_path / "empty.tsv"
+    empty.write_text("")
+    assert load_examples("qqp", str(empty), set_type="train") == []

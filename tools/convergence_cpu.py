@@ -84,9 +84,35 @@ def run(model, compressor, density, steps, world, record_every, lr=None, bs=32,
              for r in range(world)]
     for p in procs:
         p.start()
-    losses = q.get(timeout=1800)
+    # wait on the result queue while watching child exitcodes — a crashed
+    # rank otherwise leaves the survivors hung in collectives for the full
+    # timeout and zombies behind
+    import queue as _queue
+    import time as _time
+
+    losses = None
+    deadline = _time.monotonic() + 1800
+    while losses is None:
+        try:
+            losses = q.get(timeout=5)
+        except _queue.Empty:
+            dead = [p for p in procs if p.exitcode not in (None, 0)]
+            if dead or _time.monotonic() > deadline:
+                for p in procs:
+                    if p.is_alive():
+                        p.terminate()
+                for p in procs:
+                    p.join(30)
+                codes = [p.exitcode for p in procs]
+                raise RuntimeError(
+                    f"convergence child failed (exitcodes {codes})")
     for p in procs:
         p.join(120)
+        if p.is_alive():
+            p.terminate()
+            p.join(30)
+        if p.exitcode not in (0, None):
+            raise RuntimeError(f"convergence child exited {p.exitcode}")
     return losses
 
 
@@ -117,8 +143,9 @@ def main():
             prev = json.load(f)
         results = prev.get("loss_trajectories", {})
         run_steps = prev.get("run_steps", {})
+        run_setups = prev.get("run_setups", {})
     except (OSError, ValueError):
-        results, run_steps = {}, {}
+        results, run_steps, run_setups = {}, {}, {}
     setup = {"model": args.model, "batch_size": args.batch_size,
              "seq_len": args.seq_len, "model_kwargs": mkw, "dtype": "fp32",
              "device": "cpu", "density": args.density,
@@ -133,9 +160,11 @@ def main():
             key = f"world{world}/{comp}"
             results[key] = losses
             run_steps[key] = args.steps
+            run_setups[key] = setup  # per-run settings: merged files may mix
             print(f"{key:28s} " + " ".join(f"{x:7.3f}" for x in losses), flush=True)
             with open(args.out, "w") as f:
                 json.dump({"setup": setup, "run_steps": run_steps,
+                           "run_setups": run_setups,
                            "loss_trajectories": results}, f, indent=1)
 
     # tracking verdict over EVERYTHING in the evidence file: sparse final
@@ -163,6 +192,7 @@ def main():
                             "tracks_dense": ok}
     with open(args.out, "w") as f:
         json.dump({"setup": setup, "run_steps": run_steps,
+                   "run_setups": run_setups,
                    "loss_trajectories": results, "verdict": verdict},
                   f, indent=1)
     bad = [k for k, v in verdict.items() if not v["tracks_dense"]]
