@@ -475,7 +475,53 @@ class AllReducer:
             ops.zero_at_masked_(st.residual, idx, st.mask)
             st.mask[g_sel_idx.long()] = False  # cheap sparse reset
 
-    # -- chunked Ok-Topk (docs/overlap_design.md) ------------------------
+    # -- stage-pipelined Ok-Topk (docs/overlap_design.md) ----------------
+    def run_many(self, items: List[Tuple[str, torch.Tensor, Optional[torch.Tensor]]]) -> None:
+        """Reduce several INDEPENDENT flat gradients in one stage-pipelined
+        pass: `items` is [(name, tensor, grad_src-or-None)], each mutated in
+        place exactly as run() would (bit-equal — tested).  For the oktopk
+        compressor the items share each pipeline stage, so item i's
+        collectives overlap item i+1's selection/merge (the cross-bucket
+        analogue of the reference's per-group background processing,
+        VGG/allreducer.py:549-1643); other compressors and the
+        profiling_norm / chunked / balanced modes fall back to serial run().
+        Collective issue order is identical on every rank: list order plus
+        per-item counters, which agree across ranks by construction.
+        """
+        comp, ok = self.cfg.compressor, self.cfg.oktopk
+        if (comp != "oktopk" or self.cfg.profiling_norm
+                or ok.pipeline_chunks > 1 or ok.balanced_allgather
+                or len(items) <= 1):
+            for name, t, g in items:
+                self.run(name, t, grad_src=g)
+            return
+        if self.timing_sync and items[0][1].is_cuda:
+            torch.cuda.synchronize()
+        parts, warm = [], []
+        for name, tensor, g in items:
+            t = tensor.reshape(-1)
+            st = self.state(name, t)
+            it = st.counter
+            st.counter += 1
+            if it < ok.dense_warmup_iters:
+                if g is not None:
+                    t.copy_(g.reshape(-1).to(t.dtype))
+                warm.append(t)
+            else:
+                st.grad_src = g.reshape(-1) if g is not None else None
+                parts.append({"sl": t, "st": st, "it": it})
+        # warm-up items: async dense allreduces, waits deferred past the
+        # sparse pipeline (same issue order on every rank — the per-item
+        # counters that pick the path agree across ranks)
+        works = [self.comm.allreduce_async_(t) for t in warm]
+        self._oktopk_pipeline("batch", parts)
+        if warm:
+            s = time.perf_counter()
+            for t, w in zip(warm, works):
+                w.wait()
+                t.div_(self.comm.size)
+            self._time("batch", "allreduce", time.perf_counter() - s)
+
     def _oktopk_chunked(self, name: str, tensor: torch.Tensor,
                         grad_src: Optional[torch.Tensor]) -> torch.Tensor:
         """Stage-interleaved Ok-Topk over pipeline_chunks slices.
@@ -489,14 +535,13 @@ class AllReducer:
         Opt-in via OkTopkConfig.pipeline_chunks; balanced_allgather and
         profiling_norm are not supported in this mode.
         """
-        cfg, ok, comm = self.cfg, self.cfg.oktopk, self.comm
+        cfg, ok = self.cfg, self.cfg.oktopk
         if cfg.profiling_norm:
             raise ValueError("profiling_norm requires pipeline_chunks == 1")
         if ok.balanced_allgather:
             raise ValueError("balanced_allgather requires pipeline_chunks == 1 "
                              "(the chunked engine's per-chunk allgathers are "
                              "already size-balanced by the chunk split)")
-        P, rank = comm.size, comm.rank
         t = tensor.reshape(-1)
         n = t.numel()
         C = max(1, min(ok.pipeline_chunks, n // 8 or 1))
@@ -524,12 +569,27 @@ class AllReducer:
             sl = t[lo:hi]
             st = self.state(f"{name}/c{i}", sl)
             st.grad_src = g[lo:hi] if g is not None else None
-            chunks.append({"sl": sl, "st": st})
+            chunks.append({"sl": sl, "st": st, "it": it})
+        self._oktopk_pipeline(name, chunks)
+        return tensor
 
-        # --- stage 1: local selection (bulk GPU work, all chunks) --------
+    def _oktopk_pipeline(self, label: str, parts: List[dict]) -> None:
+        """The 4-stage deferred-wait Ok-Topk pipeline over independent flat
+        slices.  Each part: {"sl": 1-D grad view, "st": TensorState with
+        grad_src pre-set, "it": that part's iteration number}.  Stages run
+        part-by-part in list order; collectives use the async comm variants
+        so an earlier part's exchange progresses while later parts compute.
+        """
+        comm, ok = self.comm, self.cfg.oktopk
+        P, rank = comm.size, comm.rank
+        if not parts:
+            return
+        dev = parts[0]["sl"].device
+
+        # --- stage 1: local selection (bulk GPU work, all parts) ---------
         s0 = time.perf_counter()
-        for c in chunks:
-            st, sl = c["st"], c["sl"]
+        for c in parts:
+            st, sl, it = c["st"], c["sl"], c["it"]
             k = c["k"] = self._k(sl.numel())
             if it % ok.local_threshold_recompute_interval == 0 or st.tau_local <= 0.0:
                 self._ef_restore(sl, st)
@@ -548,21 +608,21 @@ class AllReducer:
                 st.tau_local /= ok.scale_local
             elif sel > ok.local_hi_num * k // ok.local_hi_den:
                 st.tau_local *= ok.scale_local
-        self._time(name, "compress", time.perf_counter() - s0)
+        self._time(label, "compress", time.perf_counter() - s0)
 
         if P == 1:
             s4 = time.perf_counter()
-            for c in chunks:
-                self._chunk_round2_local(c, it)
-            self._time(name, "merge", time.perf_counter() - s4)
-            return tensor
+            for c in parts:
+                self._chunk_round2_local(c, c["it"])
+            self._time(label, "merge", time.perf_counter() - s4)
+            return
 
         # --- stage 2: repartition + round-1 exchange (deferred waits) ----
         s1 = time.perf_counter()
-        for c in chunks:
+        for c in parts:
             st, idx = c["st"], c["idx"]
             nl = c["sl"].numel()
-            self._repartition(st, idx, nl, it)
+            self._repartition(st, idx, nl, c["it"])
             bounds = st.boundaries
             c["lo_r"], c["hi_r"] = int(bounds[rank]), int(bounds[rank + 1])
             split_pts = torch.searchsorted(
@@ -573,45 +633,45 @@ class AllReducer:
                                c["val"][cuts[j]:cuts[j + 1]]) for j in range(P)]
             c["send"] = comm.to_comm(torch.cat(segs))
             c["w_sizes"] = comm.alltoall_sizes_async(c["elem_counts"], comm.device)
-        for c in chunks:
+        for c in parts:
             recv_elems = c["recv_elems"] = c["w_sizes"].wait()
             c["w_payload"] = comm.alltoallv_async(
                 c["send"],
                 [self._pack_ints(x) for x in c["elem_counts"]],
                 [self._pack_ints(x) for x in recv_elems])
-        self._time(name, "alltoall", time.perf_counter() - s1)
+        self._time(label, "alltoall", time.perf_counter() - s1)
 
         # --- stage 3: reduce + round-2 select + allgather (deferred) -----
         s2 = time.perf_counter()
-        for c in chunks:
+        for c in parts:
             st = c["st"]
             r_idx, r_val = self._unpack(c["w_payload"].wait(), c["recv_elems"])
-            reduced = torch.zeros(c["hi_r"] - c["lo_r"], dtype=t.dtype,
-                                  device=t.device)
+            reduced = torch.zeros(c["hi_r"] - c["lo_r"],
+                                  dtype=c["sl"].dtype, device=dev)
             if r_idx.numel():
-                ops.scatter_add_(reduced, r_idx.to(t.device) - c["lo_r"],
-                                 r_val.to(t.device))
+                ops.scatter_add_(reduced, r_idx.to(dev) - c["lo_r"],
+                                 r_val.to(dev))
             exact = c["exact"] = (
-                it % ok.global_threshold_recompute_interval == 0
+                c["it"] % ok.global_threshold_recompute_interval == 0
                 or st.tau_global <= 0.0)
             gidx, gval = ops.compact_gt(reduced, 0.0 if exact else st.tau_global)
             c["gidx"] = gidx + c["lo_r"]
             c["gval"] = gval
             c["w_gsizes"] = comm.allgather_sizes_async(gidx.numel(), comm.device)
-        for c in chunks:
+        for c in parts:
             sizes = c["gsizes"] = [int(x) for x in c["w_gsizes"].wait()]
             pack = comm.to_comm(self._pack(c["gidx"], c["gval"]))
             c["w_gather"] = comm.allgatherv_async(
                 pack, [self._pack_ints(x) for x in sizes])
-        self._time(name, "allgather", time.perf_counter() - s2)
+        self._time(label, "allgather", time.perf_counter() - s2)
 
         # --- stage 4: merge ----------------------------------------------
         s4 = time.perf_counter()
-        for c in chunks:
+        for c in parts:
             st, sl = c["st"], c["sl"]
             all_idx, all_val = self._unpack(c["w_gather"].wait(), c["gsizes"])
-            all_idx = all_idx.to(t.device)
-            all_val = all_val.to(t.device)
+            all_idx = all_idx.to(dev)
+            all_val = all_val.to(dev)
             if c["exact"]:
                 kk = min(c["k"], all_val.numel())
                 if kk > 0:
@@ -629,9 +689,8 @@ class AllReducer:
                 elif gsz > ok.global_hi_num * c["k"] // ok.global_hi_den:
                     st.tau_global *= ok.scale_global_decrease
             ops.fill_sparse_scaled_(sl, g_sel_idx, g_sel_val, 1.0 / P)
-            self._residual_credit(st, c["idx"], g_sel_idx, sl.numel(), t.device)
-        self._time(name, "merge", time.perf_counter() - s4)
-        return tensor
+            self._residual_credit(st, c["idx"], g_sel_idx, sl.numel(), dev)
+        self._time(label, "merge", time.perf_counter() - s4)
 
     def _chunk_round2_local(self, c: dict, it: int) -> None:
         """World-1 round 2 for one chunk (no comm — mirror of the fast path

@@ -232,6 +232,13 @@ class Comm:
     # (deterministic chunk loops); waits may be deferred so later chunks'
     # collectives progress on the RCCL streams while the host works.
 
+    def allreduce_async_(self, t: torch.Tensor, op: str = "sum") -> "AsyncResult":
+        """In-place async allreduce; .wait() returns t."""
+        if self._size == 1:
+            return AsyncResult(None, t)
+        work = dist.all_reduce(t, op=_op(op), group=self.group, async_op=True)
+        return AsyncResult(work, t)
+
     def alltoall_sizes_async(self, send_sizes: Sequence[int],
                              device: torch.device) -> "AsyncResult":
         if self._size == 1:

@@ -19,6 +19,8 @@ Two surfaces, mirroring the reference:
 from __future__ import annotations
 
 import math
+import queue
+import threading
 from typing import Dict, Iterable, List, Optional, Tuple
 
 import torch
@@ -92,6 +94,72 @@ class _Bucket:
         self.ready = False
 
 
+class _ReducerWorker:
+    """Background reducer thread — the reference's architecture
+    (VGG/distributed_optimizer.py:57-59, VGG/allreducer.py:549): all engine
+    collectives issue from this ONE thread in bucket-completion order, so
+    RCCL waits and host readbacks (size exchanges, .item() syncs) overlap
+    the backward pass instead of stalling the autograd-hook thread.
+
+    Determinism: buckets are enqueued in autograd completion order, which is
+    identical on every rank for identical graphs (the same assumption DDP
+    and the reference make), so collective issue order matches across ranks.
+    GPU kernels from this thread go on the caller's default stream — stream
+    order alone serialises them correctly against backward (the hook fires
+    after the grad accumulation was enqueued).
+    """
+
+    def __init__(self, reducer: AllReducer, device: torch.device):
+        self._reducer = reducer
+        self._device = device
+        self._q: "queue.Queue" = queue.Queue()
+        self._done: "queue.Queue" = queue.Queue()
+        self._exc: Optional[BaseException] = None
+        self._inflight = 0
+        self._thread = threading.Thread(
+            target=self._loop, daemon=True, name="oktopk-reducer")
+        self._thread.start()
+
+    def _loop(self) -> None:
+        if self._device.type == "cuda":
+            torch.cuda.set_device(self._device)
+        while True:
+            item = self._q.get()
+            if item is None:
+                return
+            try:
+                if self._exc is None:  # after a failure, drain without work
+                    if isinstance(item, list):
+                        self._reducer.run_many(item)
+                    else:
+                        name, flat, grad = item
+                        self._reducer.run(name, flat, grad_src=grad)
+            except BaseException as e:  # noqa: BLE001 — repropagated in drain()
+                self._exc = e
+            finally:
+                self._done.put(None)
+
+    def submit(self, item) -> None:
+        self._inflight += 1
+        self._q.put(item)
+
+    def drain(self) -> None:
+        """Block until every submitted item finished; re-raise any engine
+        exception on the caller (the reference's msg_queue2 'DONE' join,
+        VGG/distributed_optimizer.py:96-105)."""
+        while self._inflight:
+            self._done.get()
+            self._inflight -= 1
+        if self._exc is not None:
+            exc, self._exc = self._exc, None
+            raise exc
+
+    def stop(self) -> None:
+        if self._thread.is_alive():
+            self._q.put(None)
+            self._thread.join(timeout=30)
+
+
 class _DistributedOptimizer:
     def __init__(
         self,
@@ -101,6 +169,7 @@ class _DistributedOptimizer:
         cfg: Optional[EngineConfig] = None,
         max_grad_norm: float = 0.0,
         momentum_correction: float = 0.0,
+        overlap: Optional[bool] = None,
     ):
         self.optimizer = optimizer
         # post-reduce gradient clipping (reference clips for the LSTM recipe,
@@ -151,6 +220,15 @@ class _DistributedOptimizer:
             h = p.register_post_accumulate_grad_hook(self._make_hook(p))
             self._hooks.append(h)
 
+        # Background reducer (on by default whenever there is real
+        # communication to hide; at world-1 the engine never blocks on
+        # comm, so inline is cheaper).
+        if overlap is None:
+            overlap = self.comm.size > 1
+        self._worker = (
+            _ReducerWorker(self.reducer, device) if overlap else None
+        )
+
     def _seal_bucket(self, items, device):
         name = f"bucket_{len(self.buckets)}"
         self.buckets.append(_Bucket(name, [p for _, p in items], device))
@@ -164,15 +242,26 @@ class _DistributedOptimizer:
             b.pending -= 1
             if b.pending == 0:
                 self._apply_mc(b)
-                self.reducer.run(b.name, b.flat)
+                if self._worker is not None:
+                    # hand the completed bucket to the reducer thread and
+                    # return immediately — backward keeps running while the
+                    # engine's collectives and host syncs proceed there
+                    self._worker.submit((b.name, b.flat, None))
+                else:
+                    self.reducer.run(b.name, b.flat)
                 b.ready = True
 
         return hook
 
     # reference API surface (VGG/distributed_optimizer.py:185-201) ------
     def stop(self) -> None:
-        """Reference stops its background reducer thread; there is none
-        here (hooks fire inline over async RCCL) — kept for API parity."""
+        """Stop the background reducer thread (reference stop,
+        VGG/distributed_optimizer.py:185-187).  Safe to call twice; the
+        optimizer falls back to inline reduces afterwards."""
+        if self._worker is not None:
+            self._worker.drain()
+            self._worker.stop()
+            self._worker = None
 
     def add_train_epoch(self) -> None:
         self.reducer.train_epoch += 1
@@ -206,14 +295,32 @@ class _DistributedOptimizer:
             b.attach_grads()
 
     def synchronize(self):
-        """Reduce any bucket whose hook set never completed (e.g. params not
-        touched this step); reference synchronize,
-        VGG/distributed_optimizer.py:96-105."""
-        for b in self.buckets:
-            if not b.ready and not self.local:
-                self._apply_mc(b)
+        """Finish this step's reduces (reference synchronize,
+        VGG/distributed_optimizer.py:96-105): reduce any bucket whose hook
+        set never completed — params untouched this step, or ALL buckets
+        when hooks were muted (hipGraph replay, gradient-accumulation
+        boundary) — then join the background reducer.  The all-buckets case
+        goes through run_many so bucket i's collectives overlap bucket
+        i+1's selection instead of reducing serially."""
+        if self.local:
+            return
+        leftovers = [b for b in self.buckets if not b.ready]
+        for b in leftovers:
+            self._apply_mc(b)
+            b.ready = True
+        whole_step = len(leftovers) == len(self.buckets) and len(leftovers) > 1
+        if self._worker is not None:
+            if whole_step:
+                self._worker.submit([(b.name, b.flat, None) for b in leftovers])
+            else:
+                for b in leftovers:
+                    self._worker.submit((b.name, b.flat, None))
+            self._worker.drain()
+        elif whole_step:
+            self.reducer.run_many([(b.name, b.flat, None) for b in leftovers])
+        else:
+            for b in leftovers:
                 self.reducer.run(b.name, b.flat)
-                b.ready = True
 
     def step(self, closure=None):
         if not self.local:
@@ -264,11 +371,14 @@ def DistributedOptimizer(
     density: Optional[float] = None,
     norm_clip: Optional[float] = None,
     momentum_correction: float = 0.0,
+    overlap: Optional[bool] = None,
     **_ignored,
 ) -> _DistributedOptimizer:
     """Factory with the reference's calling convention
     (VGG/distributed_optimizer.py:203: DistributedOptimizer(optimizer,
-    named_parameters, compression, is_sparse, density, ...))."""
+    named_parameters, compression, is_sparse, density, ...)).
+    `overlap` forces the background reducer thread on/off (default: on
+    when world size > 1)."""
     cfg = cfg or EngineConfig()
     if compression is not None:
         cfg.compressor = compression
@@ -278,7 +388,8 @@ def DistributedOptimizer(
         cfg.density = density
     return _DistributedOptimizer(optimizer, named_parameters, comm, cfg,
                                  max_grad_norm=norm_clip or 0.0,
-                                 momentum_correction=momentum_correction)
+                                 momentum_correction=momentum_correction,
+                                 overlap=overlap)
 
 
 class FlatBertAdam:
