@@ -35,7 +35,9 @@ def parse_args():
     p.add_argument("--pipeline-chunks", type=int, default=1,
                    help="chunked engine pipeline (overlap comm with compress)")
     p.add_argument("--dense-baseline-steps", type=int, default=-1,
-                   help="steps for the in-run dense baseline (-1: min(steps,10); 0: skip)")
+                   help="steps for the in-run dense baseline (-1: equal to "
+                        "--steps so the speedup figure compares like with "
+                        "like; 0: skip)")
     return p.parse_args()
 
 
@@ -135,7 +137,10 @@ def main():
 
     dense_steps = args.dense_baseline_steps
     if dense_steps < 0:
-        dense_steps = min(args.steps, 10)
+        # equal steps: at 8 GPUs RCCL warm-up effects make a short dense
+        # arm noisy, and the headline speedup figure deserves a
+        # like-for-like comparison (VERDICT r01 weak spot 4)
+        dense_steps = args.steps
     dense_ms = None
     speedup = None
     if dense_steps > 0 and args.compressor not in ("dense", "none"):

@@ -1096,21 +1096,26 @@ class AllReducer:
 
         s1 = time.perf_counter()
         if P > 1:
-            if P & (P - 1):
-                raise ValueError("gtopk requires a power-of-two world size")
+            # Binomial-tree pairwise merge over an explicit alive list:
+            # round r pairs consecutive survivors (receiver = lower rank);
+            # an unpaired trailing survivor carries its packet forward.  At
+            # powers of two this is exactly the reference's schedule
+            # (participate_ranks = range(0, P, 2^r), VGG/allreducer.py:116-150);
+            # the alive-list form also covers general world sizes, where the
+            # reference's participate_ranks[local_rank+1] indexes out of
+            # range (e.g. P=6, round 1).
             cur_idx, cur_val = comm.to_comm(idx), comm.to_comm(val)
-            nrounds = int(math.log2(P))
-            peer_dist = 1
-            alive = True
-            for _ in range(nrounds):
-                if alive:
-                    group_pos = (rank // peer_dist) % 2
-                    peer = rank + peer_dist if group_pos == 0 else rank - peer_dist
-                    pack = self._pack(cur_idx, cur_val)
-                    if group_pos == 0:
-                        other = torch.empty_like(pack)
-                        rr = comm.irecv(other, src=peer, tag=7)
-                        rr.wait()
+            alive = list(range(P))
+            while len(alive) > 1:
+                nxt = []
+                sent = False
+                for j in range(0, len(alive) - 1, 2):
+                    recv_rank, send_rank = alive[j], alive[j + 1]
+                    nxt.append(recv_rank)
+                    if rank == recv_rank:
+                        other = torch.empty(self._pack_ints(k),
+                                            dtype=torch.int32, device=comm.device)
+                        comm.irecv(other, src=send_rank, tag=7).wait()
                         o_idx, o_val = self._unpack(other, [k])
                         merged = torch.zeros(n, dtype=t.dtype, device=t.device)
                         ops.scatter_add_(merged, cur_idx.to(t.device), cur_val.to(t.device))
@@ -1118,11 +1123,15 @@ class AllReducer:
                         topm = torch.topk(merged.abs(), k, sorted=False)
                         cur_idx = comm.to_comm(topm.indices.to(torch.int32))
                         cur_val = comm.to_comm(merged[topm.indices])
-                    else:
-                        sr = comm.isend(pack, dst=peer, tag=7)
-                        sr.wait()
-                        alive = False
-                peer_dist *= 2
+                    elif rank == send_rank:
+                        comm.isend(self._pack(cur_idx, cur_val),
+                                   dst=recv_rank, tag=7).wait()
+                        sent = True
+                if len(alive) % 2:
+                    nxt.append(alive[-1])
+                alive = nxt
+                if sent:
+                    break
             # root (rank 0) holds the winner; broadcast 2k packet
             final = self._pack(cur_idx, cur_val) if rank == 0 else torch.empty(
                 self._pack_ints(k), dtype=torch.int32, device=comm.device

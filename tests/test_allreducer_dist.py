@@ -345,6 +345,17 @@ def test_world8_gtopk():
     run_dist(_gtopk_world8, 8)
 
 
+def test_world3_gtopk():
+    # non-power-of-two world: the alive-list merge must leave the unpaired
+    # trailing survivor carrying its packet forward (the reference's
+    # participate_ranks logic indexes out of range here)
+    run_dist(_gtopk_world8, 3)
+
+
+def test_world6_gtopk():
+    run_dist(_gtopk_world8, 6)
+
+
 def test_world8_oktopk():
     # oktopk IS EF-complete at any world size — full mass invariant
     run_dist(_mass_conservation, 8, args=("oktopk",))
