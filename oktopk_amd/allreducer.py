@@ -729,7 +729,9 @@ class AllReducer:
         comm = self.comm
         P = comm.size
         ok = self.cfg.oktopk
-        if st.boundaries is None:
+        if st.boundaries is None or st.boundaries.numel() != P + 1:
+            # numel mismatch: boundaries restored from a checkpoint taken at
+            # a different world size (elastic shrink/resume) — re-derive
             st.boundaries = self._uniform_boundaries(n)
         if it % ok.region_repartition_interval != 0 or P <= 1:
             return

@@ -14,7 +14,11 @@ checkpoint-and-requeue is in `install_preemption_handler`.
 """
 from __future__ import annotations
 
+import datetime
+import json
 import signal
+import threading
+import time
 from typing import Callable, Iterable, List, Optional
 
 import torch.distributed as dist
@@ -38,6 +42,135 @@ def apply_shrink(reducer, trainer, new_comm: Comm) -> None:
     reducer.set_comm(new_comm)
     if trainer is not None:
         trainer.update_nworker(new_comm)
+
+
+class RankFailure(RuntimeError):
+    """A peer rank is believed dead; the caller should run
+    ElasticAgent.rebuild() and retry the step from its last snapshot."""
+
+    def __init__(self, dead: List[int]):
+        super().__init__(f"dead ranks: {dead}")
+        self.dead = dead
+
+
+class ElasticAgent:
+    """Failure detection + re-rendezvous — the torch.distributed analogue of
+    the reference's MPI.ERRORS_RETURN + err_callback
+    (/root/reference/VGG/allreducer.py:220,237, VGG/main_trainer.py:42-44).
+
+    MPI with ERRORS_RETURN surfaces a dead peer as an error return from the
+    collective and ULFM-style repair rebuilds the communicator; torch has
+    neither, so the equivalent here is:
+
+    * a side TCPStore (independent of any process group) carrying per-rank
+      heartbeat counters from a daemon thread;
+    * failure evidence = a collective timeout (set a short process-group
+      timeout) OR a heartbeat stalled for > grace_s, confirmed by sampling
+      the counters twice grace_s apart;
+    * a race-free alarm: the first detector publishes the dead set with
+      compare_set; every survivor — including ones already blocked in a
+      doomed collective until its timeout fires — converges on the same
+      alarm and calls rebuild();
+    * rebuild() = destroy_process_group + init_process_group on a
+      generation-prefixed view of the same store with the survivor ranks
+      renumbered — the err_callback(new_nworkers, new_rank) moment; hand
+      the returned Comm to apply_shrink().
+
+    The store server lives in original rank 0 (host it externally for
+    rank-0 fault tolerance — the reference has the same single point of
+    failure in mpirun).  A falsely-suspected live rank finds itself in the
+    dead set at its next check and must exit (fencing).
+
+    The engine's EF residuals mutate before the failed collective, so the
+    caller retries the step from a snapshot (params + optimizer/engine
+    state_dict — the per-step cost is one device-side copy) or restores the
+    last checkpoint; tests/test_elastic_checkpoint.py does the former.
+    """
+
+    def __init__(self, host: str, port: int, rank: int, world: int,
+                 heartbeat_s: float = 0.25, grace_s: float = 2.0,
+                 store: Optional["dist.Store"] = None):
+        self.rank = rank
+        self.ranks = list(range(world))  # original rank ids, current world
+        self.generation = 0
+        self.heartbeat_s = heartbeat_s
+        self.grace_s = grace_s
+        self.store = store if store is not None else dist.TCPStore(
+            host, port, world, rank == 0, wait_for_workers=False)
+        self._beat = 0
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._heartbeat_loop,
+                                        daemon=True, name="oktopk-heartbeat")
+        self._thread.start()
+
+    # -- heartbeat ------------------------------------------------------
+    def _heartbeat_loop(self) -> None:
+        while not self._stop.is_set():
+            self._beat += 1
+            try:
+                self.store.set(f"hb/{self.rank}", str(self._beat))
+            except Exception:  # noqa: BLE001 — store server died with rank 0
+                return
+            self._stop.wait(self.heartbeat_s)
+
+    def _read_beats(self) -> dict:
+        out = {}
+        for r in self.ranks:
+            key = f"hb/{r}"
+            out[r] = int(self.store.get(key)) if self.store.check([key]) else -1
+        return out
+
+    # -- detection ------------------------------------------------------
+    def find_dead(self) -> List[int]:
+        """Sample heartbeats twice grace_s apart; a rank whose counter did
+        not advance is dead.  Call this after a collective timeout, or
+        periodically from the training loop."""
+        a = self._read_beats()
+        time.sleep(self.grace_s)
+        b = self._read_beats()
+        return [r for r in self.ranks if r != self.rank and b[r] <= a[r]]
+
+    def _alarm_key(self) -> str:
+        return f"alarm/gen{self.generation}"
+
+    def raise_alarm(self, dead: List[int]) -> List[int]:
+        """Publish the dead set for this generation (first writer wins);
+        returns the agreed set."""
+        winner = self.store.compare_set(
+            self._alarm_key(), "", json.dumps(sorted(dead)))
+        return json.loads(winner)
+
+    def check_alarm(self) -> Optional[List[int]]:
+        """Non-blocking: the agreed dead set if any rank raised the alarm
+        for the current generation."""
+        if self.store.check([self._alarm_key()]):
+            return json.loads(self.store.get(self._alarm_key()))
+        return None
+
+    # -- repair ---------------------------------------------------------
+    def rebuild(self, dead: List[int], backend: Optional[str] = None,
+                timeout_s: float = 60.0) -> Comm:
+        """Collective among survivors: tear down the default process group
+        and re-rendezvous generation g+1 on the shared store.  Returns the
+        new Comm; the caller passes it to apply_shrink()."""
+        if self.rank in dead:
+            raise SystemExit("fenced: this rank was declared dead")
+        backend = backend or (dist.get_backend() if dist.is_initialized() else "gloo")
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        survivors = [r for r in self.ranks if r not in dead]
+        self.generation += 1
+        new_rank = survivors.index(self.rank)
+        prefix = dist.PrefixStore(f"gen{self.generation}", self.store)
+        dist.init_process_group(
+            backend, store=prefix, rank=new_rank, world_size=len(survivors),
+            timeout=datetime.timedelta(seconds=timeout_s))
+        self.ranks = survivors
+        return Comm(dist.group.WORLD)
+
+    def stop(self) -> None:
+        self._stop.set()
+        self._thread.join(timeout=5)
 
 
 def install_preemption_handler(save_fn: Callable[[], None],

@@ -80,3 +80,135 @@ def test_checkpoint_roundtrip_sgd(tmp_path):
     ):
         assert n1 == n2
         assert torch.allclose(b1.residual, b2.residual)
+
+
+# ---------------------------------------------------------------------------
+# failure-path elastic: a rank DIES mid-training and survivors continue
+# (VERDICT r01 what's-missing 5; reference bar: MPI.ERRORS_RETURN +
+# err_callback -> update_nworker, VGG/allreducer.py:220,237,
+# VGG/dl_trainer.py:472-493)
+# ---------------------------------------------------------------------------
+
+def _spawn_elastic(fn, world, die_rank, die_step, timeout=240.0):
+    """run_dist-style spawner with a SHORT process-group timeout (failure
+    evidence = collective timeout) and a side TCPStore port for the
+    ElasticAgent."""
+    import torch.multiprocessing as mp
+    from conftest import free_port
+
+    port_pg, port_store = free_port(), free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_elastic_entry,
+                         args=(fn, r, world, port_pg, port_store,
+                               die_rank, die_step))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout)
+    for r, p in enumerate(procs):
+        if p.is_alive():
+            p.terminate()
+            raise TimeoutError(f"rank {r} timed out")
+        assert p.exitcode == 0, f"rank {r} exited {p.exitcode}"
+
+
+def _elastic_entry(fn, rank, world, port_pg, port_store, die_rank, die_step):
+    import datetime
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port_pg)
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            timeout=datetime.timedelta(seconds=8))
+    fn(rank, world, port_store, die_rank, die_step)
+
+
+def _snapshot(tr):
+    return {
+        "params": [p.detach().clone() for p in tr.model.parameters()],
+        "opt": {
+            "optimizer": {
+                k: ([{kk: (vv.clone() if torch.is_tensor(vv) else vv)
+                      for kk, vv in d.items()} for d in v]
+                    if k == "state_list" else v)
+                for k, v in [("sd", tr.opt.optimizer.state_dict())]
+            },
+        },
+        "reducer": {n: {k: (v.clone() if torch.is_tensor(v) else v)
+                        for k, v in s.state_dict().items()}
+                    for n, s in tr.opt.reducer.states.items()},
+        "iteration": tr.iteration,
+    }
+
+
+def _restore(tr, snap):
+    with torch.no_grad():
+        for p, saved in zip(tr.model.parameters(), snap["params"]):
+            p.copy_(saved)
+    tr.opt.optimizer.load_state_dict(snap["opt"]["optimizer"]["sd"])
+    for n, d in snap["reducer"].items():
+        tr.opt.reducer.states[n].load_state_dict(d)
+    tr.iteration = snap["iteration"]
+
+
+def _failure_worker(rank, world, port_store, die_rank, die_step):
+    import torch.distributed as dist
+
+    from oktopk_amd import Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    from oktopk_amd.elastic import ElasticAgent, apply_shrink
+    from oktopk_amd.trainer import Trainer
+
+    agent = ElasticAgent("127.0.0.1", port_store, rank, world,
+                         heartbeat_s=0.2, grace_s=1.5)
+    comm = Comm(dist.group.WORLD)
+    cfg = EngineConfig(compressor="oktopk", density=0.05,
+                       oktopk=OkTopkConfig(dense_warmup_iters=1,
+                                           region_repartition_interval=4))
+    tr = Trainer("mnistnet", batch_size=16, comm=comm, cfg=cfg, dtype="fp32")
+
+    steps = 8
+    losses = []
+    shrunk = False
+    for step in range(steps):
+        if rank == die_rank and step == die_step:
+            os._exit(0)  # silent death: heartbeat stops, no goodbye
+        dead = agent.check_alarm()
+        loss = None
+        if dead is None:
+            snap = _snapshot(tr)
+            try:
+                loss = tr.step()
+            except RuntimeError:
+                # collective timed out: confirm who is dead, publish alarm
+                dead = agent.raise_alarm(agent.find_dead())
+        if dead is not None and not shrunk:
+            assert dead == [die_rank], dead
+            new_comm = agent.rebuild(dead)
+            # restore FIRST: the snapshot carries old-world region
+            # boundaries; apply_shrink's set_comm resets them for the new P
+            _restore(tr, snap)
+            apply_shrink(tr.opt.reducer, tr, new_comm)
+            loss = tr.step()  # retry the failed step in the shrunken world
+            shrunk = True
+        losses.append(loss)
+
+    assert shrunk, "failure was never detected"
+    assert all(l is not None and torch.isfinite(torch.tensor(l)) for l in losses)
+    # training made progress through the failure
+    assert losses[-1] < losses[0], losses
+    # survivors hold identical parameters (same reduced gradients applied)
+    flat = torch.cat([p.detach().reshape(-1) for p in tr.model.parameters()])
+    ref = flat.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.equal(flat, ref)
+    agent.stop()
+
+
+def test_elastic_rank_failure_world4():
+    """Kill rank 3 of a world-4 run at step 3; ranks 0-2 detect the death
+    (collective timeout + heartbeat stall), agree via the alarm key,
+    re-rendezvous as world 3, restore the pre-step snapshot, retry the
+    step, and train on to a lower loss with rank-identical parameters."""
+    _spawn_elastic(_failure_worker, 4, die_rank=3, die_step=3)
