@@ -142,7 +142,17 @@ def partition_bert(bert_model, num_stages: int) -> List[nn.Module]:
 
 class StageComm:
     """Shape-prefixed p2p tensor send/recv between pipeline neighbours
-    (reference BERT/communication.py:469-516)."""
+    (reference BERT/communication.py:469-516).
+
+    Tag semantics: NCCL/RCCL IGNORES p2p tags — matching there is purely by
+    posting order per (src, dst) pair.  That is safe here because the
+    pipeline schedules are deterministic and every message is a
+    meta-then-payload pair posted in the same order on both sides; the tag
+    is still passed for gloo, and — so the ordering assumption is CHECKED
+    rather than assumed — the meta packet carries the logical tag, which
+    the receiver verifies.  A schedule divergence therefore fails loudly
+    on every backend instead of silently delivering the wrong tensor.
+    """
 
     def __init__(self, group=None, device: Optional[torch.device] = None):
         self.group = group
@@ -153,7 +163,8 @@ class StageComm:
         # non-blocking: a blocking send deadlocks 1F1B (both neighbours in
         # send at once); buffers are kept alive on the pending list.
         meta = torch.tensor(
-            [t.dim()] + list(t.shape) + [0] * (8 - t.dim()), dtype=torch.int64
+            [tag, t.dim()] + list(t.shape) + [0] * (8 - t.dim()),
+            dtype=torch.int64,
         ).to(self.device)
         payload = t.contiguous().to(self.device)
         self._pending.append((dist.isend(meta, dst=dst, tag=tag, group=self.group), meta))
@@ -168,11 +179,17 @@ class StageComm:
         self._pending.clear()
 
     def recv(self, src: int, dtype: torch.dtype, device, tag: int = 0) -> torch.Tensor:
-        meta = torch.zeros(9, dtype=torch.int64, device=self.device)
+        meta = torch.zeros(10, dtype=torch.int64, device=self.device)
         dist.recv(meta, src=src, tag=tag, group=self.group)
         meta = meta.cpu()
-        dim = int(meta[0])
-        shape = [int(x) for x in meta[1 : 1 + dim]]
+        got = int(meta[0])
+        if got != tag:
+            raise RuntimeError(
+                f"pipeline p2p message mismatch: expected tag {tag}, got "
+                f"{got} from rank {src} (schedule divergence — on NCCL "
+                f"tags are ignored and matching is by posting order)")
+        dim = int(meta[1])
+        shape = [int(x) for x in meta[2 : 2 + dim]]
         buf = torch.empty(shape, dtype=dtype, device=self.device)
         dist.recv(buf, src=src, tag=tag + 1, group=self.group)
         return buf.to(device)

@@ -144,3 +144,52 @@ def test_chunked_rejects_balanced_allgather():
                                          balanced_allgather=True)))
     with pytest.raises(ValueError, match="balanced_allgather"):
         eng.run("w", torch.randn(64))
+
+
+def _adversarial_repartition(rank):
+    """World-8 repartition with adversarially skewed index distributions
+    (VERDICT r01 weak 3): all ranks concentrate mass in ONE narrow band, a
+    degenerate rank selects nothing (uniform quantile contribution), one
+    rank dominates magnitudes.  Boundaries must stay monotone and the EF
+    mass invariant must hold."""
+    import torch.distributed as dist
+
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    world = dist.get_world_size()
+    n = 16384
+    cfg = EngineConfig(
+        compressor="oktopk", density=0.02,
+        oktopk=OkTopkConfig(dense_warmup_iters=0,
+                            region_repartition_interval=1))
+    eng = AllReducer(Comm(dist.group.WORLD), cfg)
+    in_sum = torch.zeros(n)
+    out_sum = torch.zeros(n)
+    for it in range(6):
+        g = torch.Generator().manual_seed(31 * rank + it)
+        t = torch.zeros(n)
+        if rank == 0:
+            pass  # degenerate: empty selection every iteration
+        elif rank == 1:
+            t = torch.randn(n, generator=g) * 1e4  # magnitude-dominant
+        else:
+            # everyone else: all mass inside the same narrow band
+            band = slice(n // 2, n // 2 + n // 64)
+            t[band] = torch.randn(n // 64, generator=g) * 10
+        in_sum += t
+        out = eng.run("w", t.clone())
+        out_sum += out
+        assert torch.isfinite(out).all()
+        b = eng.states["w"].boundaries
+        assert b is not None and bool((b[1:] >= b[:-1]).all()), b
+        assert int(b[0]) == 0 and int(b[-1]) == n
+    dist.all_reduce(in_sum)
+    res = eng.states["w"].residual.clone()
+    dist.all_reduce(res)
+    err = (world * out_sum + res - in_sum).abs().max().item()
+    assert err < 2e-2, f"EF mass leak under skew: {err}"
+
+
+def test_adversarial_repartition_world8():
+    run_dist(_adversarial_repartition, 8)
