@@ -230,3 +230,43 @@ def test_hook_muted_drain_uses_run_many(monkeypatch):
     opt.synchronize()
     assert calls == [len(opt.buckets)]
     opt.step()
+
+
+# ---------------------------------------------------------------------------
+# GPU: the worker thread must be CUDA-safe (stream ordering, set_device)
+# ---------------------------------------------------------------------------
+import pytest
+
+
+@pytest.mark.gpu
+def test_overlap_worker_cuda_bitequal():
+    """World-1, overlap FORCED on: results must bit-match the inline path
+    with the engine running entirely on the reducer thread over CUDA."""
+    from oktopk_amd import Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    from oktopk_amd.optimizer import DistributedOptimizer
+
+    def train(overlap):
+        torch.manual_seed(11)
+        model = _mlp(seed=4).cuda()
+        cfg = EngineConfig(compressor="oktopk", density=0.05,
+                           bucket_bytes=8 << 10,
+                           oktopk=OkTopkConfig(dense_warmup_iters=1))
+        opt = DistributedOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.05),
+            model.named_parameters(), comm=Comm(None), cfg=cfg,
+            overlap=overlap)
+        for it in range(5):
+            g = torch.Generator().manual_seed(it)
+            x = torch.randn(16, 64, generator=g).cuda()
+            y = torch.randint(0, 8, (16,), generator=g).cuda()
+            opt.zero_grad()
+            torch.nn.functional.cross_entropy(model(x), y).backward()
+            opt.step()
+        opt.stop()
+        torch.cuda.synchronize()
+        return torch.cat([p.detach().reshape(-1) for p in model.parameters()]).cpu()
+
+    a = train(False)
+    b = train(True)
+    assert torch.equal(a, b), (a - b).abs().max()
