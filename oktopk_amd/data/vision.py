@@ -49,6 +49,43 @@ def cifar_like_dataset(
     return TensorDataset(x, y)
 
 
+def digits_dataset(split: str = "train", image_size: int = 32,
+                   channels: int = 3, test_fraction: float = 0.2,
+                   seed: int = 0) -> Dataset:
+    """REAL image data available offline: the UCI handwritten-digits set
+    bundled with scikit-learn (1797 8x8 grayscale images, 10 classes).
+
+    Serves the role of the reference's real CIFAR-10 accuracy runs
+    (VGG/dl_trainer.py:286,709-784) in an offline image: images are
+    bilinearly upsampled to `image_size` and channel-repeated so the
+    CIFAR-shaped model zoo (resnet20/vgg16/caffe_cifar) trains unchanged.
+    Deterministic seeded train/test split; per-channel standardization
+    from TRAIN statistics only.
+    """
+    from sklearn.datasets import load_digits
+
+    X, y = load_digits(return_X_y=True)
+    x = torch.from_numpy(X).float().view(-1, 1, 8, 8) / 16.0
+    y = torch.from_numpy(y).long()
+    g = torch.Generator().manual_seed(seed)
+    perm = torch.randperm(x.shape[0], generator=g)
+    n_test = int(x.shape[0] * test_fraction)
+    idx = perm[n_test:] if split == "train" else perm[:n_test]
+    x, y = x[idx], y[idx]
+    if image_size != 8:
+        x = torch.nn.functional.interpolate(
+            x, size=(image_size, image_size), mode="bilinear",
+            align_corners=False)
+    if channels != 1:
+        x = x.repeat(1, channels, 1, 1)
+    tr_idx = perm[n_test:]
+    x_all = torch.from_numpy(X).float().view(-1, 1, 8, 8) / 16.0
+    mean = x_all[tr_idx].mean()
+    std = x_all[tr_idx].std().clamp_min(1e-6)
+    x = (x - mean) / std
+    return TensorDataset(x, y)
+
+
 class Hdf5ImagenetDataset(Dataset):
     """The reference's HDF5-packed ImageNet (VGG/datasets.py DatasetHDF5:
     one 'data'/'label' pair per split, read in chunks)."""

@@ -446,6 +446,13 @@ class FlatBertAdam:
         self.decay_numel = sum(
             named[i][1].numel() for i in order if _decays(named[i][0])
         )
+        self._decay_count = sum(1 for i in order if _decays(named[i][0]))
+        # pad the group boundary to 8 elements so BOTH fused-Adam launches
+        # see 16-byte-aligned float4 (and 8-byte short4 mirror) slices; the
+        # pad region has zero grad/state and never moves
+        pad = (-self.decay_numel) % 8
+        self.decay_numel += pad
+        self.numel += pad
         # Mixed precision: when the model runs in pure bf16 (no autocast —
         # the per-layer weight-cast kernels of autocast were ~1000 extra
         # launches/step on BERT-base, measured in profiles/), keep fp32
@@ -464,7 +471,9 @@ class FlatBertAdam:
             self.flat_grad_model = self.flat_grad
         self.slots = []
         off = 0
-        for p in self.params:
+        for j, p in enumerate(self.params):
+            if j == self._decay_count:
+                off = self.decay_numel  # skip the alignment pad
             n_ = p.numel()
             self.slots.append((off, n_))
             self.flat_param[off : off + n_].copy_(p.data.view(-1).float())

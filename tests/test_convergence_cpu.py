@@ -58,3 +58,24 @@ def test_flat_adam_engine_descends_like_dense():
     assert dense[-1] < 0.4 * dense[0]
     assert ok[-1] < 0.4 * ok[0]
     assert ok[-1] <= max(4 * dense[-1], dense[-1] + 0.3), (ok[-5:], dense[-5:])
+
+
+def test_real_digits_oktopk_tracks_dense_world2():
+    """REAL-dataset end-to-end regression (VERDICT r01 what's-missing 4):
+    resnet20 on the bundled UCI digits images through the full stack
+    (hooks -> bucketed sparse engine -> SGD) at gloo world 2 — oktopk@1%
+    must reach held-out test top-1 within a few points of dense and far
+    above chance.  Reference bar: the CIFAR accuracy loop,
+    VGG/dl_trainer.py:709-784."""
+    import sys as _sys
+    import os as _os
+    _sys.path.insert(0, _os.path.join(
+        _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))), "tools"))
+    from convergence_real import run
+
+    epochs = 8
+    dense = run("dense", 0.01, epochs, world=2, model="resnet20")
+    sparse = run("oktopk", 0.01, epochs, world=2, model="resnet20")
+    assert dense[-1] > 90.0, dense
+    assert sparse[-1] > 85.0, sparse
+    assert sparse[-1] >= dense[-1] - 8.0, (sparse, dense)
