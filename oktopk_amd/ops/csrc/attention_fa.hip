@@ -1,0 +1,326 @@
+// Flash (online-softmax) self-attention forward — bf16, head_dim 64, any
+// seq multiple of 128.  Generalizes attention.hip's seq=128 single-pass
+// kernel: KV is processed in 128-key tiles with the
+// online max/sum rescale (cdna_hip_programming.md T13 discussion — the
+// textbook order: a tile's P is exponentiated only after the rescale
+// decision that covers it), so nothing O(S^2) is materialised.  For the
+// backward only the per-row LSE (log-sum-exp) is saved; P is recomputed
+// tile-free in torch and the dropout mask is REGENERATED from the same
+// philox counters (dropout_mask_mul below), so the forward writes only
+// ctx + lse.
+//
+// Structure per 256-thread block (4 waves): 64 query rows (16 per wave),
+// one (batch, head) per blockIdx.x, query block per blockIdx.y — grid
+// B*NH*(S/64) blocks, 2-3 blocks/CU by LDS.  QK^T and PV are
+// v_mfma_f32_16x16x32_bf16 tiles; softmax runs on the accumulator layout
+// (row-quarter shuffle reduction, same idiom as attention.hip); P stages
+// through padded LDS for the PV A-fragments; V transposes into LDS at
+// stage time.
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8_f;
+typedef __attribute__((ext_vector_type(4))) float f32x4_f;
+
+struct PhiloxArgsFA {
+    unsigned long long seed;
+    unsigned long long offset;
+    const unsigned long long* seed_ptr;    // captured (hipGraph) variant
+    const unsigned long long* offset_ptr;
+    unsigned int intragraph;
+    int captured;
+};
+
+__device__ __forceinline__ float fa_b2f(short u) {
+    union { float f; uint32_t i; } c;
+    c.i = ((uint32_t)(uint16_t)u) << 16;
+    return c.f;
+}
+
+__device__ __forceinline__ short fa_f2b(float f) {
+    union { float f; uint32_t i; } c;
+    c.f = f;
+    uint32_t lsb = (c.i >> 16) & 1;
+    c.i += 0x7fff + lsb;
+    return (short)(c.i >> 16);
+}
+
+__device__ __forceinline__ void fa_philox4(unsigned long long seed,
+                                           unsigned long long ctr,
+                                           uint32_t out[4]) {
+    uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+    uint32_t c0 = (uint32_t)ctr, c1 = (uint32_t)(ctr >> 32), c2 = 0, c3 = 0;
+    #pragma unroll
+    for (int i = 0; i < 10; ++i) {
+        uint32_t hi0 = __umulhi(0xD2511F53u, c0), lo0 = 0xD2511F53u * c0;
+        uint32_t hi1 = __umulhi(0xCD9E8D57u, c2), lo1 = 0xCD9E8D57u * c2;
+        uint32_t n0 = hi1 ^ c1 ^ k0, n1 = lo1, n2 = hi0 ^ c3 ^ k1, n3 = lo0;
+        c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+        k0 += 0x9E3779B9u; k1 += 0xBB67AE85u;
+    }
+    out[0] = c0; out[1] = c1; out[2] = c2; out[3] = c3;
+}
+
+#define FA_D 64
+#define FA_BM 64            // query rows per block (16 per wave)
+#define FA_BT 128           // keys per KV tile
+#define FA_LDK (FA_D + 8)   // K rows padded (bf16 elems)
+#define FA_LDV (FA_BT + 8)  // Vt rows padded
+#define FA_LDP (FA_BT + 8)  // P rows padded
+
+__global__ void __launch_bounds__(256)
+attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
+                   const short* __restrict__ mask, // [b, s] additive bf16
+                   short* __restrict__ out,        // [b, s, nh*hd]
+                   float* __restrict__ lse,        // [b*nh, s] (or null)
+                   int B, int NH, int S, float scale, float keep_prob,
+                   PhiloxArgsFA rng, int apply_dropout) {
+    const int bh = blockIdx.x;
+    const int b = bh / NH, h = bh % NH;
+    const int qblk = blockIdx.y;            // query block of FA_BM rows
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int fr = lane & 15;               // fragment row/col
+    const int fk = (lane >> 4) * 8;         // k-offset within K=32 fragment
+
+    __shared__ short k_lds[FA_BT * FA_LDK];
+    __shared__ short vt_lds[FA_D * FA_LDV];
+    __shared__ short p_lds[4 * 16 * FA_LDP];
+
+    unsigned long long seed = rng.seed, offset = rng.offset;
+    if (rng.captured) {
+        seed = *rng.seed_ptr;
+        offset = *rng.offset_ptr + rng.intragraph;
+    }
+    const float inv_keep = 1.f / keep_prob;
+    const uint32_t thresh = (uint32_t)(keep_prob * 4294967296.0);
+
+    const int64_t qkv_row = (int64_t)3 * NH * FA_D;
+    const int64_t base_b = (int64_t)b * S * qkv_row + (int64_t)h * FA_D;
+    const int q_base = qblk * FA_BM + wave * 16;  // this wave's 16 rows
+
+    // Q fragment is tile-invariant: load once (rows q_base+fr, k=fk..fk+8
+    // per 32-chunk)
+    bf16x8_f qfr[2];
+    #pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+        qfr[ks] = *reinterpret_cast<const bf16x8_f*>(
+            qkv + base_b + (int64_t)(q_base + fr) * qkv_row + ks * 32 + fk);
+
+    // online-softmax state: lane covers rows q_base + (lane>>4)*4 + r
+    float m_run[4], l_run[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+    f32x4_f oacc[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int ntiles = S / FA_BT;
+    for (int t = 0; t < ntiles; ++t) {
+        const int key0 = t * FA_BT;
+        // ---- stage K tile rows + V tile transposed ------------------
+        {
+            int tid = threadIdx.x;
+            #pragma unroll
+            for (int pass = 0; pass < 4; ++pass) {
+                int idx = pass * 256 + tid;    // 0..1023 = 128 rows x 8 chunks
+                int row = idx >> 3, d0 = (idx & 7) * 8;
+                const short* src = qkv + base_b +
+                    (int64_t)(key0 + row) * qkv_row + NH * FA_D + d0;
+                bf16x8_f kv = *reinterpret_cast<const bf16x8_f*>(src);
+                *reinterpret_cast<bf16x8_f*>(&k_lds[row * FA_LDK + d0]) = kv;
+                const short* vsrc = src + NH * FA_D;  // c=2
+                bf16x8_f vv = *reinterpret_cast<const bf16x8_f*>(vsrc);
+                #pragma unroll
+                for (int q = 0; q < 8; ++q)
+                    vt_lds[(d0 + q) * FA_LDV + row] = vv[q];
+            }
+        }
+        __syncthreads();
+
+        // ---- QK^T for this tile: M=16, N=128, K=64 ------------------
+        f32x4_f acc[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] = {0.f, 0.f, 0.f, 0.f};
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            bf16x8_f bfr[8];
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+                bfr[j] = *reinterpret_cast<const bf16x8_f*>(
+                    &k_lds[(j * 16 + fr) * FA_LDK + ks * 32 + fk]);
+            #pragma unroll
+            for (int j = 0; j < 8; ++j)
+                acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    qfr[ks], bfr[j], acc[j], 0, 0, 0);
+        }
+
+        // ---- scale + mask, online rescale, exponentiate -------------
+        float mvals[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+            mvals[j] = mask ? fa_b2f(mask[(int64_t)b * S + key0 + j * 16 + fr])
+                            : 0.f;
+        float alpha[4];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            float tm = -1e30f;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                acc[j][r] = acc[j][r] * scale + mvals[j];
+                tm = fmaxf(tm, acc[j][r]);
+            }
+            #pragma unroll
+            for (int off = 1; off < 16; off <<= 1)
+                tm = fmaxf(tm, __shfl_xor(tm, off, 64));
+            float mnew = fmaxf(m_run[r], tm);
+            alpha[r] = __expf(m_run[r] - mnew);
+            float tsum = 0.f;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                float e = __expf(acc[j][r] - mnew);
+                acc[j][r] = e;
+                tsum += e;
+            }
+            #pragma unroll
+            for (int off = 1; off < 16; off <<= 1)
+                tsum += __shfl_xor(tsum, off, 64);
+            l_run[r] = l_run[r] * alpha[r] + tsum;
+            m_run[r] = mnew;
+            // rescale O before this tile's PV lands (textbook order)
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) oacc[j][r] *= alpha[r];
+        }
+
+        // ---- dropout (on unnormalised P — the l sum above is pre-drop,
+        //      as softmax's denominator must be) + stage P -------------
+        short* my_p = &p_lds[wave * 16 * FA_LDP];
+        const int64_t drop_base = (int64_t)bh * S * S;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int prow = (lane >> 4) * 4 + r;
+            int qrow = q_base + prow;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int key = j * 16 + fr;
+                float a = acc[j][r];
+                if (apply_dropout) {
+                    int64_t elem = drop_base + (int64_t)qrow * S + key0 + key;
+                    uint32_t rnd[4];
+                    fa_philox4(seed, offset + (unsigned long long)(elem >> 2),
+                               rnd);
+                    a = (rnd[elem & 3] < thresh) ? a * inv_keep : 0.f;
+                }
+                my_p[prow * FA_LDP + key] = fa_f2b(a);
+            }
+        }
+        __syncthreads();
+
+        // ---- PV accumulate: M=16, N=64, K=128 -----------------------
+        #pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+            bf16x8_f afr = *reinterpret_cast<const bf16x8_f*>(
+                &my_p[fr * FA_LDP + ks * 32 + fk]);
+            bf16x8_f bfr[4];
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                bfr[j] = *reinterpret_cast<const bf16x8_f*>(
+                    &vt_lds[(j * 16 + fr) * FA_LDV + ks * 32 + fk]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                oacc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afr, bfr[j], oacc[j], 0, 0, 0);
+        }
+        __syncthreads();  // k_lds/vt_lds/p_lds free for the next tile
+    }
+
+    // ---- epilogue: normalise, write ctx + lse -----------------------
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int qrow = q_base + (lane >> 4) * 4 + r;
+        float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+        int64_t row = ((int64_t)b * S + qrow) * NH * FA_D + h * FA_D;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j)
+            out[row + j * 16 + fr] = fa_f2b(oacc[j][r] * inv_l);
+        if (lse && fr == 0)
+            lse[(int64_t)bh * S + qrow] =
+                m_run[r] + (l_run[r] > 0.f ? __logf(l_run[r]) : 0.f);
+    }
+}
+
+extern "C" void launch_attn_fwd_fa(const void* qkv, const void* mask, void* out,
+                                   void* lse, int B, int NH, int S, float scale,
+                                   float keep_prob, unsigned long long seed,
+                                   unsigned long long offset,
+                                   const void* seed_ptr, const void* offset_ptr,
+                                   unsigned int intragraph, int captured,
+                                   int apply_dropout, hipStream_t stream) {
+    PhiloxArgsFA rng;
+    rng.seed = seed;
+    rng.offset = offset;
+    rng.seed_ptr = (const unsigned long long*)seed_ptr;
+    rng.offset_ptr = (const unsigned long long*)offset_ptr;
+    rng.intragraph = intragraph;
+    rng.captured = captured;
+    hipLaunchKernelGGL(attn_fwd_fa_kernel, dim3(B * NH, S / FA_BM), dim3(256),
+                       0, stream, (const short*)qkv, (const short*)mask,
+                       (short*)out, (float*)lse, B, NH, S, scale, keep_prob,
+                       rng, apply_dropout);
+}
+
+// ---------------------------------------------------------------------------
+// dropout mask regeneration for the recompute backward: a[i] *= inv_keep or
+// 0 with the SAME philox counters the forward used (elem>>2 block, elem&3
+// word).  One thread per aligned 4-element philox block.
+// ---------------------------------------------------------------------------
+__global__ void dropout_mask_mul_kernel(short* __restrict__ a, int64_t n,
+                                        int64_t base_elem, PhiloxArgsFA rng,
+                                        uint32_t thresh, float inv_keep) {
+    unsigned long long seed = rng.seed, offset = rng.offset;
+    if (rng.captured) {  // hipGraph: same device-side state the fwd read
+        seed = *rng.seed_ptr;
+        offset = *rng.offset_ptr + rng.intragraph;
+    }
+    int64_t blk = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t nblk = (n + 3) >> 2;
+    for (; blk < nblk; blk += stride) {
+        int64_t e0 = blk * 4;
+        uint32_t rnd[4];
+        fa_philox4(seed, offset + (unsigned long long)((base_elem + e0) >> 2),
+                   rnd);
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            int64_t e = e0 + q;
+            if (e < n) {
+                float v = fa_b2f(a[e]);
+                v = (rnd[q] < thresh) ? v * inv_keep : 0.f;
+                a[e] = fa_f2b(v);
+            }
+        }
+    }
+}
+
+extern "C" void launch_dropout_mask_mul(void* a, int64_t n, int64_t base_elem,
+                                        unsigned long long seed,
+                                        unsigned long long offset,
+                                        const void* seed_ptr,
+                                        const void* offset_ptr,
+                                        unsigned int intragraph, int captured,
+                                        float keep_prob, hipStream_t stream) {
+    PhiloxArgsFA rng;
+    rng.seed = seed;
+    rng.offset = offset;
+    rng.seed_ptr = (const unsigned long long*)seed_ptr;
+    rng.offset_ptr = (const unsigned long long*)offset_ptr;
+    rng.intragraph = intragraph;
+    rng.captured = captured;
+    uint32_t thresh = (uint32_t)(keep_prob * 4294967296.0);
+    int64_t nblk = (n + 3) >> 2;
+    int blocks = (int)((nblk + 255) / 256);
+    if (blocks > 65535) blocks = 65535;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(dropout_mask_mul_kernel, dim3(blocks), dim3(256), 0,
+                       stream, (short*)a, n, base_elem, rng, thresh,
+                       1.f / keep_prob);
+}

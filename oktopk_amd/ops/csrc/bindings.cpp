@@ -50,6 +50,13 @@ void launch_attn_fwd(const void*, const void*, void*, void*, void*, int, int,
                      float, float, unsigned long long, unsigned long long,
                      const void*, const void*, unsigned int, int, int,
                      hipStream_t);
+void launch_attn_fwd_fa(const void*, const void*, void*, void*, int, int, int,
+                        float, float, unsigned long long, unsigned long long,
+                        const void*, const void*, unsigned int, int, int,
+                        hipStream_t);
+void launch_dropout_mask_mul(void*, int64_t, int64_t, unsigned long long,
+                             unsigned long long, const void*, const void*,
+                             unsigned int, int, float, hipStream_t);
 }
 
 namespace {
@@ -531,6 +538,95 @@ static std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, torch::Tensor mask
     return {out};
 }
 
+static std::vector<torch::Tensor> attn_fwd_fa(torch::Tensor qkv,
+                                              torch::Tensor mask,
+                                              int64_t num_heads,
+                                              double dropout_p, bool training,
+                                              bool want_lse) {
+    TORCH_CHECK(qkv.is_cuda() && qkv.scalar_type() == torch::kBFloat16 &&
+                qkv.is_contiguous());
+    TORCH_CHECK(qkv.dim() == 3, "qkv must be [b, s, 3*h]");
+    int64_t B = qkv.size(0), S = qkv.size(1);
+    int64_t H3 = qkv.size(2);
+    int64_t H = H3 / 3;
+    int64_t HD = H / num_heads;
+    TORCH_CHECK(S % 128 == 0 && HD == 64,
+                "flash attention needs seq % 128 == 0 and hd == 64");
+    const at::cuda::CUDAGuard guard(qkv.device());
+    auto out = torch::empty({B, S, H}, qkv.options());
+    torch::Tensor lse;
+    void* lse_ptr = nullptr;
+    if (want_lse) {
+        lse = torch::empty({B * num_heads, S},
+                           qkv.options().dtype(torch::kFloat32));
+        lse_ptr = lse.data_ptr();
+    }
+    const void* mptr = nullptr;
+    torch::Tensor mask_c;
+    if (mask.defined() && mask.numel()) {
+        mask_c = mask.reshape({B, S}).to(torch::kBFloat16).contiguous();
+        mptr = mask_c.data_ptr();
+    }
+    int apply_dropout = (training && dropout_p > 0.0) ? 1 : 0;
+    unsigned long long seed = 0, offset = 0;
+    const void* seed_ptr = nullptr;
+    const void* offset_ptr = nullptr;
+    unsigned int intragraph = 0;
+    int captured = 0;
+    if (apply_dropout) {
+        auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+            c10::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+        at::PhiloxCudaState state;
+        {
+            std::lock_guard<std::mutex> lock(gen->mutex_);
+            state = gen->philox_cuda_state((B * num_heads * S * S + 3) / 4 + 1);
+        }
+        if (state.captured_) {
+            captured = 1;
+            seed_ptr = state.seed_.ptr;
+            offset_ptr = state.offset_.ptr;
+            intragraph = state.offset_intragraph_;
+        } else {
+            seed = state.seed_.val;
+            offset = state.offset_.val;
+        }
+    }
+    float scale = 1.0f / std::sqrt((float)HD);
+    launch_attn_fwd_fa(qkv.data_ptr(), mptr, out.data_ptr(), lse_ptr, (int)B,
+                       (int)num_heads, (int)S, scale,
+                       (float)(1.0 - dropout_p), seed, offset, seed_ptr,
+                       offset_ptr, intragraph, captured, apply_dropout,
+                       cur_stream());
+    // philox state rides back as a CPU int64 tensor so the recompute
+    // backward can regenerate the identical dropout mask (pointers are
+    // round-tripped as ints; valid while the capture/graph lives)
+    auto st = torch::tensor(
+        {(int64_t)seed, (int64_t)offset, (int64_t)(uintptr_t)seed_ptr,
+         (int64_t)(uintptr_t)offset_ptr, (int64_t)intragraph,
+         (int64_t)captured},
+        torch::TensorOptions().dtype(torch::kInt64));
+    if (want_lse) return {out, lse, st};
+    return {out, st};
+}
+
+static void dropout_mask_mul_(torch::Tensor a, int64_t base_elem,
+                              torch::Tensor philox_state, double dropout_p) {
+    TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 &&
+                a.is_contiguous());
+    TORCH_CHECK(base_elem % 4 == 0, "base_elem must be philox-block aligned");
+    TORCH_CHECK(philox_state.numel() == 6);
+    auto st = philox_state.to(torch::kCPU);
+    auto* p = st.data_ptr<int64_t>();
+    const at::cuda::CUDAGuard guard(a.device());
+    // captured mode: the kernel reads the device-side seed/offset (same
+    // graph/replay as the fwd), so the mask is identical per replay
+    launch_dropout_mask_mul(
+        a.data_ptr(), a.numel(), base_elem, (unsigned long long)p[0],
+        (unsigned long long)p[1], (const void*)(uintptr_t)p[2],
+        (const void*)(uintptr_t)p[3], (unsigned int)p[4], (int)p[5],
+        (float)(1.0 - dropout_p), cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "oktopk_amd CDNA4 HIP kernels (gfx950)";
     m.def("count_gt", &count_gt, "count |t| > tau");
@@ -560,6 +656,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_fwd", &attn_fwd,
           "fused self-attention forward: softmax(QK^T*scale+mask) dropout @ V "
           "(bf16 MFMA, seq=128/hd=64; returns ctx [+P, A for backward])");
+    m.def("attn_fwd_fa", &attn_fwd_fa,
+          "flash (online-softmax) attention forward: any seq % 128 == 0, "
+          "hd=64 bf16; returns (ctx, lse, philox_state) — O(S) memory");
+    m.def("dropout_mask_mul_", &dropout_mask_mul_,
+          "regenerate the forward's philox dropout mask in-place: "
+          "a[i] = keep ? a[i]/keep_prob : 0");
     m.def("add_ln_fwd", &add_ln_fwd, "fused y=LN(x+r) forward (bf16)");
     m.def("add_ln_bwd", &add_ln_bwd, "fused add+LN backward (bf16, fp32 col sums)");
     m.def("linear_gelu", &linear_gelu,

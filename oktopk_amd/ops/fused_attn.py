@@ -21,16 +21,17 @@ import torch
 
 
 def fused_attn_available(x: torch.Tensor, num_heads: int, seq: int, dropout_p: float) -> bool:
-    # opt-in: consistently +0.3 ms/step vs the eager chain under hipGraph
-    # replay at the reference shape (6-run interleaved A/B, profiles/
-    # README.md r01-f) — 96-block grid underfill + P/A save traffic; enable
-    # with OKTOPK_FUSED_ATTN=1
+    # opt-in (OKTOPK_FUSED_ATTN=1): the production default is torch SDPA
+    # (AOTriton flash), which won the round-1 A/B at every reference shape
+    # (profiles/README.md r01-n).  The flash kernel (attention_fa.hip)
+    # covers any seq % 128 == 0 at hd 64; the legacy single-pass kernel
+    # (OKTOPK_ATTN_LEGACY=1) only seq=128.
     if os.environ.get("OKTOPK_FUSED_ATTN", "0") != "1":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
     hd = x.shape[-1] // (3 * num_heads)
-    if seq != 128 or hd != 64:
+    if seq % 128 != 0 or hd != 64:
         return False
     from . import hip_available
 
@@ -94,7 +95,95 @@ class _FusedAttention(torch.autograd.Function):
         return dqkv.view(b, s, h3), None, None, None, None
 
 
+class _FlashAttention(torch.autograd.Function):
+    """Online-softmax forward (any seq % 128 == 0): the kernel materialises
+    nothing O(S^2) — only ctx and the per-row LSE come back.  Backward
+    recomputes P = exp(QK^T*scale + mask - lse) (already normalised, no
+    softmax pass) and REGENERATES the dropout mask from the same philox
+    counters the forward consumed (dropout_mask_mul_), then runs the bmm
+    chain of _FusedAttention.backward."""
+
+    @staticmethod
+    def forward(ctx, qkv, mask, num_heads, dropout_p, training):
+        from oktopk_amd import _hip_ops
+
+        need_grad = qkv.requires_grad
+        outs = _hip_ops.attn_fwd_fa(
+            qkv,
+            mask if mask is not None else torch.empty(0, device=qkv.device),
+            num_heads,
+            float(dropout_p),
+            bool(training),
+            need_grad,
+        )
+        if need_grad:
+            out, lse, philox = outs
+            ctx.save_for_backward(qkv, lse, philox)
+            ctx.mask = mask
+            ctx.meta = (num_heads, dropout_p, training)
+        else:
+            out = outs[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, go):
+        from oktopk_amd import _hip_ops
+
+        qkv, lse, philox = ctx.saved_tensors
+        mask = ctx.mask
+        num_heads, dropout_p, training = ctx.meta
+        b, s, h3 = qkv.shape
+        h = h3 // 3
+        hd = h // num_heads
+        scale = 1.0 / math.sqrt(hd)
+        qkv5 = qkv.view(b, s, 3, num_heads, hd)
+        q = qkv5[:, :, 0].permute(0, 2, 1, 3).reshape(b * num_heads, s, hd)
+        k = qkv5[:, :, 1].permute(0, 2, 1, 3).reshape(b * num_heads, s, hd)
+        v = qkv5[:, :, 2].permute(0, 2, 1, 3).reshape(b * num_heads, s, hd)
+        go_h = go.view(b, s, num_heads, hd).permute(0, 2, 1, 3).reshape(
+            b * num_heads, s, hd
+        )
+        # recompute normalised P from the saved LSE (one bmm + exp; the
+        # softmax reduction never re-runs)
+        sc = torch.baddbmm(
+            torch.zeros(1, dtype=torch.float32, device=qkv.device),
+            q.float(), k.transpose(1, 2).float(), alpha=scale,
+        )
+        if mask is not None:
+            sc = sc + mask.reshape(b, 1, 1, s).float().expand(
+                b, num_heads, 1, s).reshape(b * num_heads, 1, s)
+        p = torch.exp(sc - lse.view(b * num_heads, s, 1)).to(qkv.dtype)
+        del sc
+        if training and dropout_p > 0:
+            a = p.clone()
+            _hip_ops.dropout_mask_mul_(a, 0, philox, float(dropout_p))
+        else:
+            a = p
+        dv = torch.bmm(a.transpose(1, 2), go_h)
+        da = torch.bmm(go_h, v.transpose(1, 2))
+        if training and dropout_p > 0:
+            keep = 1.0 - dropout_p
+            dp = da * (a != 0).to(da.dtype) / keep
+        else:
+            dp = da
+        pf = p.float()
+        dpf = dp.float()
+        ds = (pf * (dpf - (dpf * pf).sum(dim=-1, keepdim=True))).to(qkv.dtype)
+        dq = torch.bmm(ds, k) * scale
+        dk = torch.bmm(ds.transpose(1, 2), q) * scale
+        dqkv = torch.empty_like(qkv).view(b, s, 3, num_heads, hd)
+        dqkv[:, :, 0] = dq.view(b, num_heads, s, hd).permute(0, 2, 1, 3)
+        dqkv[:, :, 1] = dk.view(b, num_heads, s, hd).permute(0, 2, 1, 3)
+        dqkv[:, :, 2] = dv.view(b, num_heads, s, hd).permute(0, 2, 1, 3)
+        return dqkv.view(b, s, h3), None, None, None, None
+
+
 def fused_attention(qkv, mask, num_heads, dropout_p, training):
     """ctx = dropout(softmax(QK^T*scale + mask)) @ V from the fused qkv
-    projection buffer [b, s, 3h]; returns [b, s, h]."""
-    return _FusedAttention.apply(qkv, mask, num_heads, dropout_p, training)
+    projection buffer [b, s, 3h]; returns [b, s, h].  Dispatch: the flash
+    (online-softmax) kernel for any seq % 128 == 0 unless
+    OKTOPK_ATTN_LEGACY=1 forces the seq-128 single-pass kernel."""
+    s = qkv.shape[1]
+    if os.environ.get("OKTOPK_ATTN_LEGACY", "0") == "1" and s == 128:
+        return _FusedAttention.apply(qkv, mask, num_heads, dropout_p, training)
+    return _FlashAttention.apply(qkv, mask, num_heads, dropout_p, training)

@@ -22,6 +22,7 @@ setup(
                 "oktopk_amd/ops/csrc/linear_gelu.hip",
                 "oktopk_amd/ops/csrc/add_layernorm.hip",
                 "oktopk_amd/ops/csrc/attention.hip",
+                "oktopk_amd/ops/csrc/attention_fa.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
