@@ -324,3 +324,375 @@ extern "C" void launch_dropout_mask_mul(void* a, int64_t n, int64_t base_elem,
                        stream, (short*)a, n, base_elem, rng, thresh,
                        1.f / keep_prob);
 }
+
+// ---------------------------------------------------------------------------
+// Flash attention BACKWARD — two kernels, no O(S^2) materialisation.
+//
+//   dP_ij = gO_i . V_j          dS_ij = P_ij * (dP_ij - D_i)
+//   P_ij  = exp(scale*Q_i.K_j + mask_j - lse_i)    D_i = gO_i . O_i
+//   dQ_i  = scale * sum_j dS_ij K_j          (kernel A: query-parallel)
+//   dK_j  = scale * sum_i dS_ij Q_i          (kernel B: key-parallel)
+//   dV_j  = sum_i A_ij gO_i,  A = dropout(P)/keep  (kernel B)
+//
+// Both recompute P from the saved LSE (no softmax reduction) and
+// regenerate the dropout mask from the forward's philox counters.
+// Same MFMA fragment idioms as the forward; transposed LDS images where
+// a B-operand's K dimension runs along rows (K^T for dQ, Q^T/gO^T for
+// dK/dV), built scalar at stage time like the forward's V transpose.
+// ---------------------------------------------------------------------------
+
+#define FB_LDR (FA_D + 8)     // row-major row pitch (bf16)
+#define FB_LDT (FA_BT + 8)    // kernel-A transposed / P pitch (128 keys)
+#define FB_QT 64              // kernel-B query tile
+#define FB_LDQ (FB_QT + 8)    // kernel-B transposed / P pitch (64 q)
+
+__global__ void __launch_bounds__(256)
+attn_bwd_dq_kernel(const short* __restrict__ qkv,
+                   const short* __restrict__ go,   // [b, s, nh*hd]
+                   const float* __restrict__ lse,  // [b*nh, s]
+                   const float* __restrict__ dvec, // [b*nh, s]  D_i
+                   const short* __restrict__ mask, // [b, s] or null
+                   short* __restrict__ dqkv,       // [b, s, 3, nh, hd]
+                   int B, int NH, int S, float scale, float keep_prob,
+                   PhiloxArgsFA rng, int apply_dropout) {
+    const int bh = blockIdx.x;
+    const int b = bh / NH, h = bh % NH;
+    const int qblk = blockIdx.y;
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int fr = lane & 15, fk = (lane >> 4) * 8;
+
+    __shared__ short k_lds[FA_BT * FB_LDR];
+    __shared__ short kt_lds[FA_D * FB_LDT];
+    __shared__ short v_lds[FA_BT * FB_LDR];
+    __shared__ short p_lds[4 * 16 * FB_LDT];
+
+    unsigned long long seed = rng.seed, offset = rng.offset;
+    if (rng.captured) {
+        seed = *rng.seed_ptr;
+        offset = *rng.offset_ptr + rng.intragraph;
+    }
+    const float inv_keep = 1.f / keep_prob;
+    const uint32_t thresh = (uint32_t)(keep_prob * 4294967296.0);
+
+    const int64_t qkv_row = (int64_t)3 * NH * FA_D;
+    const int64_t base_b = (int64_t)b * S * qkv_row + (int64_t)h * FA_D;
+    const int64_t go_row = (int64_t)NH * FA_D;
+    const int64_t go_base = (int64_t)b * S * go_row + (int64_t)h * FA_D;
+    const int q_base = qblk * FA_BM + wave * 16;
+
+    bf16x8_f qfr[2], gofr[2];
+    #pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+        qfr[ks] = *reinterpret_cast<const bf16x8_f*>(
+            qkv + base_b + (int64_t)(q_base + fr) * qkv_row + ks * 32 + fk);
+        gofr[ks] = *reinterpret_cast<const bf16x8_f*>(
+            go + go_base + (int64_t)(q_base + fr) * go_row + ks * 32 + fk);
+    }
+    float lse_r[4], d_r[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int qrow = q_base + (lane >> 4) * 4 + r;
+        lse_r[r] = lse[(int64_t)bh * S + qrow];
+        d_r[r] = dvec[(int64_t)bh * S + qrow];
+    }
+    f32x4_f dq_acc[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) dq_acc[j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int ntiles = S / FA_BT;
+    for (int t = 0; t < ntiles; ++t) {
+        const int key0 = t * FA_BT;
+        {
+            int tid = threadIdx.x;
+            #pragma unroll
+            for (int pass = 0; pass < 4; ++pass) {
+                int idx = pass * 256 + tid;
+                int row = idx >> 3, d0 = (idx & 7) * 8;
+                const short* src = qkv + base_b +
+                    (int64_t)(key0 + row) * qkv_row + NH * FA_D + d0;
+                bf16x8_f kv = *reinterpret_cast<const bf16x8_f*>(src);
+                *reinterpret_cast<bf16x8_f*>(&k_lds[row * FB_LDR + d0]) = kv;
+                #pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    kt_lds[(d0 + e) * FB_LDT + row] = kv[e];
+                bf16x8_f vv = *reinterpret_cast<const bf16x8_f*>(src + NH * FA_D);
+                *reinterpret_cast<bf16x8_f*>(&v_lds[row * FB_LDR + d0]) = vv;
+            }
+        }
+        __syncthreads();
+
+        // S and dP tiles: M=16 q, N=128 key, K=64 d
+        f32x4_f accs[8], accdp[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            accs[j] = {0.f, 0.f, 0.f, 0.f};
+            accdp[j] = {0.f, 0.f, 0.f, 0.f};
+        }
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                bf16x8_f bk = *reinterpret_cast<const bf16x8_f*>(
+                    &k_lds[(j * 16 + fr) * FB_LDR + ks * 32 + fk]);
+                accs[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    qfr[ks], bk, accs[j], 0, 0, 0);
+                bf16x8_f bv = *reinterpret_cast<const bf16x8_f*>(
+                    &v_lds[(j * 16 + fr) * FB_LDR + ks * 32 + fk]);
+                accdp[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    gofr[ks], bv, accdp[j], 0, 0, 0);
+            }
+        }
+
+        // dS = P * (dP - D) * scale, staged bf16
+        short* my_p = &p_lds[wave * 16 * FB_LDT];
+        const int64_t drop_base = (int64_t)bh * S * S;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            int key = j * 16 + fr;
+            float mv = mask ? fa_b2f(mask[(int64_t)b * S + key0 + key]) : 0.f;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int prow = (lane >> 4) * 4 + r;
+                float p = __expf(accs[j][r] * scale + mv - lse_r[r]);
+                float dp = accdp[j][r];
+                if (apply_dropout) {
+                    int64_t elem = drop_base +
+                        (int64_t)(q_base + prow) * S + key0 + key;
+                    uint32_t rnd[4];
+                    fa_philox4(seed,
+                               offset + (unsigned long long)(elem >> 2), rnd);
+                    dp = (rnd[elem & 3] < thresh) ? dp * inv_keep : 0.f;
+                }
+                my_p[prow * FB_LDT + key] =
+                    fa_f2b(p * (dp - d_r[r]) * scale);
+            }
+        }
+        __syncthreads();
+
+        // dQ += dS @ K: M=16 q, N=64 d, K=128 key
+        #pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+            bf16x8_f afr = *reinterpret_cast<const bf16x8_f*>(
+                &my_p[fr * FB_LDT + ks * 32 + fk]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                bf16x8_f bfr = *reinterpret_cast<const bf16x8_f*>(
+                    &kt_lds[(j * 16 + fr) * FB_LDT + ks * 32 + fk]);
+                dq_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afr, bfr, dq_acc[j], 0, 0, 0);
+            }
+        }
+        __syncthreads();
+    }
+
+    // dqkv[b, q, 0, h, d]
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int qrow = q_base + (lane >> 4) * 4 + r;
+        int64_t row = (int64_t)b * S * qkv_row + (int64_t)qrow * qkv_row +
+                      (int64_t)h * FA_D;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j)
+            dqkv[row + j * 16 + fr] = fa_f2b(dq_acc[j][r]);
+    }
+}
+
+__global__ void __launch_bounds__(256)
+attn_bwd_dkv_kernel(const short* __restrict__ qkv,
+                    const short* __restrict__ go,
+                    const float* __restrict__ lse,
+                    const float* __restrict__ dvec,
+                    const short* __restrict__ mask,
+                    short* __restrict__ dqkv,
+                    int B, int NH, int S, float scale, float keep_prob,
+                    PhiloxArgsFA rng, int apply_dropout) {
+    const int bh = blockIdx.x;
+    const int b = bh / NH, h = bh % NH;
+    const int kblk = blockIdx.y;            // 64 key rows per block
+    const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int fr = lane & 15, fk = (lane >> 4) * 8;
+
+    __shared__ short q_lds[FB_QT * FB_LDR];
+    __shared__ short qt_lds[FA_D * FB_LDQ];
+    __shared__ short go_lds[FB_QT * FB_LDR];
+    __shared__ short got_lds[FA_D * FB_LDQ];
+    __shared__ short pa_lds[4 * 16 * FB_LDQ];   // A^T tile (for dV)
+    __shared__ short pb_lds[4 * 16 * FB_LDQ];   // dS^T tile (for dK)
+
+    unsigned long long seed = rng.seed, offset = rng.offset;
+    if (rng.captured) {
+        seed = *rng.seed_ptr;
+        offset = *rng.offset_ptr + rng.intragraph;
+    }
+    const float inv_keep = 1.f / keep_prob;
+    const uint32_t thresh = (uint32_t)(keep_prob * 4294967296.0);
+
+    const int64_t qkv_row = (int64_t)3 * NH * FA_D;
+    const int64_t base_b = (int64_t)b * S * qkv_row + (int64_t)h * FA_D;
+    const int64_t go_row = (int64_t)NH * FA_D;
+    const int64_t go_base = (int64_t)b * S * go_row + (int64_t)h * FA_D;
+    const int key_base = kblk * 64 + wave * 16;
+
+    bf16x8_f kfr[2], vfr[2];
+    #pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+        const short* kr = qkv + base_b +
+            (int64_t)(key_base + fr) * qkv_row + NH * FA_D;
+        kfr[ks] = *reinterpret_cast<const bf16x8_f*>(kr + ks * 32 + fk);
+        vfr[ks] = *reinterpret_cast<const bf16x8_f*>(kr + NH * FA_D + ks * 32 + fk);
+    }
+    float mask_r[4];
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int krow = key_base + (lane >> 4) * 4 + r;
+        mask_r[r] = mask ? fa_b2f(mask[(int64_t)b * S + krow]) : 0.f;
+    }
+    f32x4_f dk_acc[4], dv_acc[4];
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+        dk_acc[j] = {0.f, 0.f, 0.f, 0.f};
+        dv_acc[j] = {0.f, 0.f, 0.f, 0.f};
+    }
+
+    const int ntiles = S / FB_QT;
+    for (int t = 0; t < ntiles; ++t) {
+        const int q0 = t * FB_QT;
+        {
+            int tid = threadIdx.x;
+            #pragma unroll
+            for (int pass = 0; pass < 2; ++pass) {
+                int idx = pass * 256 + tid;   // 0..511 = 64 rows x 8 chunks
+                int row = idx >> 3, d0 = (idx & 7) * 8;
+                const short* qs = qkv + base_b +
+                    (int64_t)(q0 + row) * qkv_row + d0;
+                bf16x8_f qv = *reinterpret_cast<const bf16x8_f*>(qs);
+                *reinterpret_cast<bf16x8_f*>(&q_lds[row * FB_LDR + d0]) = qv;
+                #pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    qt_lds[(d0 + e) * FB_LDQ + row] = qv[e];
+                const short* gs = go + go_base + (int64_t)(q0 + row) * go_row + d0;
+                bf16x8_f gv = *reinterpret_cast<const bf16x8_f*>(gs);
+                *reinterpret_cast<bf16x8_f*>(&go_lds[row * FB_LDR + d0]) = gv;
+                #pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    got_lds[(d0 + e) * FB_LDQ + row] = gv[e];
+            }
+        }
+        __syncthreads();
+
+        // S^T and dP^T tiles: M=16 key, N=64 q, K=64 d
+        f32x4_f accst[4], accdpt[4];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            accst[j] = {0.f, 0.f, 0.f, 0.f};
+            accdpt[j] = {0.f, 0.f, 0.f, 0.f};
+        }
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                bf16x8_f bq = *reinterpret_cast<const bf16x8_f*>(
+                    &q_lds[(j * 16 + fr) * FB_LDR + ks * 32 + fk]);
+                accst[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    kfr[ks], bq, accst[j], 0, 0, 0);
+                bf16x8_f bg = *reinterpret_cast<const bf16x8_f*>(
+                    &go_lds[(j * 16 + fr) * FB_LDR + ks * 32 + fk]);
+                accdpt[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    vfr[ks], bg, accdpt[j], 0, 0, 0);
+            }
+        }
+
+        // per-column (query) stats + dS^T / A^T staging
+        short* my_pa = &pa_lds[wave * 16 * FB_LDQ];
+        short* my_pb = &pb_lds[wave * 16 * FB_LDQ];
+        const int64_t drop_base = (int64_t)bh * S * S;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            int qcol = j * 16 + fr;
+            float lse_c = lse[(int64_t)bh * S + q0 + qcol];
+            float d_c = dvec[(int64_t)bh * S + q0 + qcol];
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int prow = (lane >> 4) * 4 + r;
+                int krow = key_base + prow;
+                float p = __expf(accst[j][r] * scale + mask_r[r] - lse_c);
+                float dp = accdpt[j][r];
+                float a = p;
+                if (apply_dropout) {
+                    int64_t elem = drop_base + (int64_t)(q0 + qcol) * S + krow;
+                    uint32_t rnd[4];
+                    fa_philox4(seed,
+                               offset + (unsigned long long)(elem >> 2), rnd);
+                    int keep = rnd[elem & 3] < thresh;
+                    a = keep ? p * inv_keep : 0.f;
+                    dp = keep ? dp * inv_keep : 0.f;
+                }
+                my_pa[prow * FB_LDQ + qcol] = fa_f2b(a);
+                my_pb[prow * FB_LDQ + qcol] = fa_f2b(p * (dp - d_c) * scale);
+            }
+        }
+        __syncthreads();
+
+        // dV += A^T @ gO ; dK += dS^T @ Q   (M=16 key, N=64 d, K=64 q)
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+            bf16x8_f apa = *reinterpret_cast<const bf16x8_f*>(
+                &my_pa[fr * FB_LDQ + ks * 32 + fk]);
+            bf16x8_f apb = *reinterpret_cast<const bf16x8_f*>(
+                &my_pb[fr * FB_LDQ + ks * 32 + fk]);
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                bf16x8_f bg = *reinterpret_cast<const bf16x8_f*>(
+                    &got_lds[(j * 16 + fr) * FB_LDQ + ks * 32 + fk]);
+                dv_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    apa, bg, dv_acc[j], 0, 0, 0);
+                bf16x8_f bq = *reinterpret_cast<const bf16x8_f*>(
+                    &qt_lds[(j * 16 + fr) * FB_LDQ + ks * 32 + fk]);
+                dk_acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    apb, bq, dk_acc[j], 0, 0, 0);
+            }
+        }
+        __syncthreads();
+    }
+
+    // dqkv[b, key, 1, h, d] = dK ; [b, key, 2, h, d] = dV
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        int krow = key_base + (lane >> 4) * 4 + r;
+        int64_t row = (int64_t)b * S * qkv_row + (int64_t)krow * qkv_row +
+                      (int64_t)h * FA_D;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            dqkv[row + NH * FA_D + j * 16 + fr] = fa_f2b(dk_acc[j][r]);
+            dqkv[row + 2 * NH * FA_D + j * 16 + fr] = fa_f2b(dv_acc[j][r]);
+        }
+    }
+}
+
+extern "C" void launch_attn_bwd_fa(const void* qkv, const void* go,
+                                   const void* lse, const void* dvec,
+                                   const void* mask, void* dqkv, int B, int NH,
+                                   int S, float scale, float keep_prob,
+                                   unsigned long long seed,
+                                   unsigned long long offset,
+                                   const void* seed_ptr, const void* offset_ptr,
+                                   unsigned int intragraph, int captured,
+                                   int apply_dropout, hipStream_t stream) {
+    PhiloxArgsFA rng;
+    rng.seed = seed;
+    rng.offset = offset;
+    rng.seed_ptr = (const unsigned long long*)seed_ptr;
+    rng.offset_ptr = (const unsigned long long*)offset_ptr;
+    rng.intragraph = intragraph;
+    rng.captured = captured;
+    hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(B * NH, S / FA_BM), dim3(256),
+                       0, stream, (const short*)qkv, (const short*)go,
+                       (const float*)lse, (const float*)dvec,
+                       (const short*)mask, (short*)dqkv, B, NH, S, scale,
+                       keep_prob, rng, apply_dropout);
+    hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(B * NH, S / 64), dim3(256),
+                       0, stream, (const short*)qkv, (const short*)go,
+                       (const float*)lse, (const float*)dvec,
+                       (const short*)mask, (short*)dqkv, B, NH, S, scale,
+                       keep_prob, rng, apply_dropout);
+}

@@ -57,6 +57,10 @@ void launch_attn_fwd_fa(const void*, const void*, void*, void*, int, int, int,
 void launch_dropout_mask_mul(void*, int64_t, int64_t, unsigned long long,
                              unsigned long long, const void*, const void*,
                              unsigned int, int, float, hipStream_t);
+void launch_attn_bwd_fa(const void*, const void*, const void*, const void*,
+                        const void*, void*, int, int, int, float, float,
+                        unsigned long long, unsigned long long, const void*,
+                        const void*, unsigned int, int, int, hipStream_t);
 }
 
 namespace {
@@ -609,6 +613,44 @@ static std::vector<torch::Tensor> attn_fwd_fa(torch::Tensor qkv,
     return {out, st};
 }
 
+static torch::Tensor attn_bwd_fa(torch::Tensor qkv, torch::Tensor go,
+                                 torch::Tensor lse, torch::Tensor dvec,
+                                 torch::Tensor mask, torch::Tensor philox_state,
+                                 int64_t num_heads, double dropout_p,
+                                 bool training) {
+    TORCH_CHECK(qkv.is_cuda() && qkv.scalar_type() == torch::kBFloat16 &&
+                qkv.is_contiguous());
+    TORCH_CHECK(go.scalar_type() == torch::kBFloat16 && go.is_contiguous());
+    TORCH_CHECK(lse.scalar_type() == torch::kFloat32 && lse.is_contiguous());
+    TORCH_CHECK(dvec.scalar_type() == torch::kFloat32 && dvec.is_contiguous());
+    int64_t B = qkv.size(0), S = qkv.size(1);
+    int64_t H = qkv.size(2) / 3;
+    int64_t HD = H / num_heads;
+    TORCH_CHECK(S % 128 == 0 && HD == 64,
+                "flash attention bwd needs seq % 128 == 0 and hd == 64");
+    const at::cuda::CUDAGuard guard(qkv.device());
+    auto dqkv = torch::empty_like(qkv);
+    const void* mptr = nullptr;
+    torch::Tensor mask_c;
+    if (mask.defined() && mask.numel()) {
+        mask_c = mask.reshape({B, S}).to(torch::kBFloat16).contiguous();
+        mptr = mask_c.data_ptr();
+    }
+    int apply_dropout = (training && dropout_p > 0.0) ? 1 : 0;
+    TORCH_CHECK(philox_state.numel() == 6);
+    auto st = philox_state.to(torch::kCPU);
+    auto* p = st.data_ptr<int64_t>();
+    float scale = 1.0f / std::sqrt((float)HD);
+    launch_attn_bwd_fa(qkv.data_ptr(), go.data_ptr(), lse.data_ptr(),
+                       dvec.data_ptr(), mptr, dqkv.data_ptr(), (int)B,
+                       (int)num_heads, (int)S, scale,
+                       (float)(1.0 - dropout_p), (unsigned long long)p[0],
+                       (unsigned long long)p[1], (const void*)(uintptr_t)p[2],
+                       (const void*)(uintptr_t)p[3], (unsigned int)p[4],
+                       (int)p[5], apply_dropout, cur_stream());
+    return dqkv;
+}
+
 static void dropout_mask_mul_(torch::Tensor a, int64_t base_elem,
                               torch::Tensor philox_state, double dropout_p) {
     TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 &&
@@ -659,6 +701,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_fwd_fa", &attn_fwd_fa,
           "flash (online-softmax) attention forward: any seq % 128 == 0, "
           "hd=64 bf16; returns (ctx, lse, philox_state) — O(S) memory");
+    m.def("attn_bwd_fa", &attn_bwd_fa,
+          "flash attention backward: dQ/dK/dV from (qkv, gO, lse, D) with "
+          "philox dropout-mask regeneration — nothing O(S^2) materialised");
     m.def("dropout_mask_mul_", &dropout_mask_mul_,
           "regenerate the forward's philox dropout mask in-place: "
           "a[i] = keep ? a[i]/keep_prob : 0");

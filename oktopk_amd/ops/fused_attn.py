@@ -118,7 +118,7 @@ class _FlashAttention(torch.autograd.Function):
         )
         if need_grad:
             out, lse, philox = outs
-            ctx.save_for_backward(qkv, lse, philox)
+            ctx.save_for_backward(qkv, lse, philox, out)
             ctx.mask = mask
             ctx.meta = (num_heads, dropout_p, training)
         else:
@@ -129,9 +129,22 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, go):
         from oktopk_amd import _hip_ops
 
-        qkv, lse, philox = ctx.saved_tensors
+        qkv, lse, philox, out = ctx.saved_tensors
         mask = ctx.mask
         num_heads, dropout_p, training = ctx.meta
+        if os.environ.get("OKTOPK_ATTN_BWD_TORCH", "0") != "1":
+            # hand-written flash backward (attn_bwd_dq/_dkv kernels):
+            # D_i = gO_i . O_i, then dQ/dK/dV with P recomputed from lse
+            b, s, h3 = qkv.shape
+            hd = (h3 // 3) // num_heads
+            go_c = go.contiguous()
+            d = (go_c.float() * out.float()).view(b, s, num_heads, hd) \
+                .sum(-1).permute(0, 2, 1).reshape(b * num_heads, s).contiguous()
+            dqkv = _hip_ops.attn_bwd_fa(
+                qkv, go_c, lse, d,
+                mask if mask is not None else torch.empty(0, device=qkv.device),
+                philox, num_heads, float(dropout_p), bool(training))
+            return dqkv, None, None, None, None
         b, s, h3 = qkv.shape
         h = h3 // 3
         hd = h // num_heads
