@@ -21,17 +21,23 @@ import torch
 
 
 def fused_attn_available(x: torch.Tensor, num_heads: int, seq: int, dropout_p: float) -> bool:
-    # opt-in (OKTOPK_FUSED_ATTN=1): the production default is torch SDPA
-    # (AOTriton flash), which won the round-1 A/B at every reference shape
-    # (profiles/README.md r01-n).  The flash kernel (attention_fa.hip)
-    # covers any seq % 128 == 0 at hd 64; the legacy single-pass kernel
-    # (OKTOPK_ATTN_LEGACY=1) only seq=128.
-    if os.environ.get("OKTOPK_FUSED_ATTN", "0") != "1":
+    # DEFAULT ON at seq 128/256 since round 2: the flash kernel pair
+    # (attention_fa.hip fwd + hand-written dQ/dKdV backward) beats torch
+    # SDPA (AOTriton) isolated (70.6 vs 121.0 us fwd+bwd at s128,
+    # profiles/attn_ab_r02*.txt) AND end-to-end under hipGraph (bert_base
+    # 10.30 -> 9.99 ms/step at s128, 12.77 -> 11.87 at s256).  At s512 the
+    # isolated A/B is a tie (174 vs 168) — SDPA stays default there;
+    # OKTOPK_FUSED_ATTN=1 forces ours at any seq % 128 == 0,
+    # OKTOPK_FUSED_ATTN=0 forces SDPA everywhere.
+    env = os.environ.get("OKTOPK_FUSED_ATTN", "")
+    if env == "0":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
     hd = x.shape[-1] // (3 * num_heads)
     if seq % 128 != 0 or hd != 64:
+        return False
+    if env != "1" and seq > 256:
         return False
     from . import hip_available
 
