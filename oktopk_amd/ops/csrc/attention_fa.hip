@@ -193,23 +193,27 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
 
         // ---- dropout (on unnormalised P — the l sum above is pre-drop,
         //      as softmax's denominator must be) + stage P -------------
+        // Dropout counter mapping (shared by fwd, both bwd kernels and the
+        // regen kernel): counter = (bh*(S/4) + qrow/4)*S + key, word =
+        // qrow&3 — ONE philox call covers a lane's 4 query rows of one key
+        // (the accumulator layout), 4x fewer philox rounds than per-element.
         short* my_p = &p_lds[wave * 16 * FA_LDP];
-        const int64_t drop_base = (int64_t)bh * S * S;
+        const int64_t ctr_row =
+            (int64_t)bh * (S >> 2) + (q_base >> 2) + (lane >> 4);
         #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            int prow = (lane >> 4) * 4 + r;
-            int qrow = q_base + prow;
+        for (int j = 0; j < 8; ++j) {
+            int key = j * 16 + fr;
+            uint32_t rnd[4];
+            if (apply_dropout)
+                fa_philox4(seed,
+                           offset + (unsigned long long)(ctr_row * S + key0 + key),
+                           rnd);
             #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                int key = j * 16 + fr;
+            for (int r = 0; r < 4; ++r) {
+                int prow = (lane >> 4) * 4 + r;
                 float a = acc[j][r];
-                if (apply_dropout) {
-                    int64_t elem = drop_base + (int64_t)qrow * S + key0 + key;
-                    uint32_t rnd[4];
-                    fa_philox4(seed, offset + (unsigned long long)(elem >> 2),
-                               rnd);
-                    a = (rnd[elem & 3] < thresh) ? a * inv_keep : 0.f;
-                }
+                if (apply_dropout)
+                    a = (rnd[r] < thresh) ? a * inv_keep : 0.f;
                 my_p[prow * FA_LDP + key] = fa_f2b(a);
             }
         }
@@ -273,35 +277,39 @@ extern "C" void launch_attn_fwd_fa(const void* qkv, const void* mask, void* out,
 // 0 with the SAME philox counters the forward used (elem>>2 block, elem&3
 // word).  One thread per aligned 4-element philox block.
 // ---------------------------------------------------------------------------
-__global__ void dropout_mask_mul_kernel(short* __restrict__ a, int64_t n,
-                                        int64_t base_elem, PhiloxArgsFA rng,
+__global__ void dropout_mask_mul_kernel(short* __restrict__ a, int64_t bh_n,
+                                        int64_t S, PhiloxArgsFA rng,
                                         uint32_t thresh, float inv_keep) {
     unsigned long long seed = rng.seed, offset = rng.offset;
     if (rng.captured) {  // hipGraph: same device-side state the fwd read
         seed = *rng.seed_ptr;
         offset = *rng.offset_ptr + rng.intragraph;
     }
-    int64_t blk = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    // a is [BH, S, S]; grouped counter mapping (see the fwd kernel):
+    // counter = (bh*(S/4) + q/4)*S + key, word = q&3 — one work item per
+    // (bh, q-group-of-4, key), applying 4 row-strided updates.
+    int64_t item = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    int64_t nblk = (n + 3) >> 2;
-    for (; blk < nblk; blk += stride) {
-        int64_t e0 = blk * 4;
+    int64_t nitems = bh_n * (S >> 2) * S;
+    for (; item < nitems; item += stride) {
+        int64_t key = item % S;
+        int64_t qg = (item / S) % (S >> 2);
+        int64_t bh = item / (S * (S >> 2));
         uint32_t rnd[4];
-        fa_philox4(seed, offset + (unsigned long long)((base_elem + e0) >> 2),
+        fa_philox4(seed,
+                   offset + (unsigned long long)((bh * (S >> 2) + qg) * S + key),
                    rnd);
+        int64_t base = bh * S * S + qg * 4 * S + key;
         #pragma unroll
-        for (int q = 0; q < 4; ++q) {
-            int64_t e = e0 + q;
-            if (e < n) {
-                float v = fa_b2f(a[e]);
-                v = (rnd[q] < thresh) ? v * inv_keep : 0.f;
-                a[e] = fa_f2b(v);
-            }
+        for (int w = 0; w < 4; ++w) {
+            float v = fa_b2f(a[base + w * S]);
+            v = (rnd[w] < thresh) ? v * inv_keep : 0.f;
+            a[base + w * S] = fa_f2b(v);
         }
     }
 }
 
-extern "C" void launch_dropout_mask_mul(void* a, int64_t n, int64_t base_elem,
+extern "C" void launch_dropout_mask_mul(void* a, int64_t bh_n, int64_t S,
                                         unsigned long long seed,
                                         unsigned long long offset,
                                         const void* seed_ptr,
@@ -316,12 +324,12 @@ extern "C" void launch_dropout_mask_mul(void* a, int64_t n, int64_t base_elem,
     rng.intragraph = intragraph;
     rng.captured = captured;
     uint32_t thresh = (uint32_t)(keep_prob * 4294967296.0);
-    int64_t nblk = (n + 3) >> 2;
-    int blocks = (int)((nblk + 255) / 256);
+    int64_t nitems = bh_n * (S >> 2) * S;
+    int blocks = (int)((nitems + 255) / 256);
     if (blocks > 65535) blocks = 65535;
     if (blocks < 1) blocks = 1;
     hipLaunchKernelGGL(dropout_mask_mul_kernel, dim3(blocks), dim3(256), 0,
-                       stream, (short*)a, n, base_elem, rng, thresh,
+                       stream, (short*)a, bh_n, S, rng, thresh,
                        1.f / keep_prob);
 }
 
@@ -445,24 +453,24 @@ attn_bwd_dq_kernel(const short* __restrict__ qkv,
 
         // dS = P * (dP - D) * scale, staged bf16
         short* my_p = &p_lds[wave * 16 * FB_LDT];
-        const int64_t drop_base = (int64_t)bh * S * S;
+        const int64_t ctr_row =
+            (int64_t)bh * (S >> 2) + (q_base >> 2) + (lane >> 4);
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
             int key = j * 16 + fr;
             float mv = mask ? fa_b2f(mask[(int64_t)b * S + key0 + key]) : 0.f;
+            uint32_t rnd[4];
+            if (apply_dropout)
+                fa_philox4(seed,
+                           offset + (unsigned long long)(ctr_row * S + key0 + key),
+                           rnd);
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int prow = (lane >> 4) * 4 + r;
                 float p = __expf(accs[j][r] * scale + mv - lse_r[r]);
                 float dp = accdp[j][r];
-                if (apply_dropout) {
-                    int64_t elem = drop_base +
-                        (int64_t)(q_base + prow) * S + key0 + key;
-                    uint32_t rnd[4];
-                    fa_philox4(seed,
-                               offset + (unsigned long long)(elem >> 2), rnd);
-                    dp = (rnd[elem & 3] < thresh) ? dp * inv_keep : 0.f;
-                }
+                if (apply_dropout)
+                    dp = (rnd[r] < thresh) ? dp * inv_keep : 0.f;
                 my_p[prow * FB_LDT + key] =
                     fa_f2b(p * (dp - d_r[r]) * scale);
             }
@@ -602,15 +610,20 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
             }
         }
 
-        // per-column (query) stats + dS^T / A^T staging
+        // per-column (query) stats + dS^T / A^T staging.  The dropout
+        // counter is grouped by QUERY rows (the fwd's layout), so here —
+        // key-parallel — each element needs its own philox call; the
+        // word index is the query row's low bits.
         short* my_pa = &pa_lds[wave * 16 * FB_LDQ];
         short* my_pb = &pb_lds[wave * 16 * FB_LDQ];
-        const int64_t drop_base = (int64_t)bh * S * S;
         #pragma unroll
         for (int j = 0; j < 4; ++j) {
             int qcol = j * 16 + fr;
             float lse_c = lse[(int64_t)bh * S + q0 + qcol];
             float d_c = dvec[(int64_t)bh * S + q0 + qcol];
+            const int64_t ctr_base =
+                ((int64_t)bh * (S >> 2) + ((q0 + qcol) >> 2)) * S;
+            const int word = qcol & 3;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int prow = (lane >> 4) * 4 + r;
@@ -619,11 +632,11 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
                 float dp = accdpt[j][r];
                 float a = p;
                 if (apply_dropout) {
-                    int64_t elem = drop_base + (int64_t)(q0 + qcol) * S + krow;
                     uint32_t rnd[4];
                     fa_philox4(seed,
-                               offset + (unsigned long long)(elem >> 2), rnd);
-                    int keep = rnd[elem & 3] < thresh;
+                               offset + (unsigned long long)(ctr_base + krow),
+                               rnd);
+                    int keep = rnd[word] < thresh;
                     a = keep ? p * inv_keep : 0.f;
                     dp = keep ? dp * inv_keep : 0.f;
                 }

@@ -651,11 +651,12 @@ static torch::Tensor attn_bwd_fa(torch::Tensor qkv, torch::Tensor go,
     return dqkv;
 }
 
-static void dropout_mask_mul_(torch::Tensor a, int64_t base_elem,
-                              torch::Tensor philox_state, double dropout_p) {
+static void dropout_mask_mul_(torch::Tensor a, torch::Tensor philox_state,
+                              double dropout_p) {
     TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 &&
                 a.is_contiguous());
-    TORCH_CHECK(base_elem % 4 == 0, "base_elem must be philox-block aligned");
+    TORCH_CHECK(a.dim() == 3 && a.size(1) == a.size(2) &&
+                a.size(1) % 4 == 0, "a must be [BH, S, S]");
     TORCH_CHECK(philox_state.numel() == 6);
     auto st = philox_state.to(torch::kCPU);
     auto* p = st.data_ptr<int64_t>();
@@ -663,7 +664,7 @@ static void dropout_mask_mul_(torch::Tensor a, int64_t base_elem,
     // captured mode: the kernel reads the device-side seed/offset (same
     // graph/replay as the fwd), so the mask is identical per replay
     launch_dropout_mask_mul(
-        a.data_ptr(), a.numel(), base_elem, (unsigned long long)p[0],
+        a.data_ptr(), a.size(0), a.size(1), (unsigned long long)p[0],
         (unsigned long long)p[1], (const void*)(uintptr_t)p[2],
         (const void*)(uintptr_t)p[3], (unsigned int)p[4], (int)p[5],
         (float)(1.0 - dropout_p), cur_stream());

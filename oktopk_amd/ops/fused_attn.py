@@ -175,7 +175,7 @@ class _FlashAttention(torch.autograd.Function):
         del sc
         if training and dropout_p > 0:
             a = p.clone()
-            _hip_ops.dropout_mask_mul_(a, 0, philox, float(dropout_p))
+            _hip_ops.dropout_mask_mul_(a, philox, float(dropout_p))
         else:
             a = p
         dv = torch.bmm(a.transpose(1, 2), go_h)

@@ -600,7 +600,7 @@ def test_flash_attention_dropout_mask_regen():
     # reference A from regenerated mask
     ref, p_ref = _attn_ref(qkv, torch.zeros(b, s).bfloat16().cuda(), nh)
     a = p_ref.reshape(b * nh, s, s).bfloat16().contiguous()
-    _hip_ops.dropout_mask_mul_(a, 0, philox, drop_p)
+    _hip_ops.dropout_mask_mul_(a, philox, drop_p)
     dropped = ((a == 0) & (p_ref.reshape(b * nh, s, s).bfloat16() != 0))
     frac = dropped.float().mean().item()
     assert abs(frac - drop_p) < 0.01, frac
@@ -713,7 +713,7 @@ def test_flash_attention_bwd_dropout_autograd():
     fn = out.grad_fn
     qkv_s, lse_s, philox_s, out_s = fn.saved_tensors
     ones = torch.ones(b * nh, s, s, dtype=torch.bfloat16, device="cuda")
-    _hip_ops.dropout_mask_mul_(ones, 0, philox_s, drop_p)
+    _hip_ops.dropout_mask_mul_(ones, philox_s, drop_p)
     keep_mask = ones.float()  # inv_keep where kept, 0 where dropped
 
     gy = torch.randn_like(out)
