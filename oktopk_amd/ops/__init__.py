@@ -132,25 +132,34 @@ def fused_sgd_(param, grad, momentum_buf, lr, momentum, weight_decay, nesterov):
     )
 
 
-def fused_adam_(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps, weight_decay):
+def fused_adam_(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps,
+                weight_decay, gscale=None):
     return _backend(param).fused_adam_(
-        param, grad, exp_avg, exp_avg_sq, float(lr), float(beta1), float(beta2), float(eps), float(weight_decay)
+        param, grad, exp_avg, exp_avg_sq, float(lr), float(beta1), float(beta2),
+        float(eps), float(weight_decay), gscale
     )
 
 
 def fused_adam_mirror_(param, grad, exp_avg, exp_avg_sq, param_bf16, lr, beta1,
-                       beta2, eps, weight_decay):
+                       beta2, eps, weight_decay, gscale=None):
     """Adam step + bf16 weight-mirror write in one pass (GPU); CPU reference
-    steps then casts."""
+    steps then casts.  `gscale` (optional 1-elem tensor) pre-scales the
+    gradient read — the device-resident grad clip."""
     b = _backend(param)
     if hasattr(b, "fused_adam_mirror_"):
         return b.fused_adam_mirror_(
             param, grad, exp_avg, exp_avg_sq, param_bf16, float(lr), float(beta1),
-            float(beta2), float(eps), float(weight_decay)
+            float(beta2), float(eps), float(weight_decay), gscale
         )
     b.fused_adam_(param, grad, exp_avg, exp_avg_sq, float(lr), float(beta1),
-                  float(beta2), float(eps), float(weight_decay))
+                  float(beta2), float(eps), float(weight_decay), gscale)
     param_bf16.copy_(param)
+
+
+def grad_clip_scale(t: torch.Tensor, max_norm: float) -> torch.Tensor:
+    """Device-resident clip factor: min(1, max/(||t||+1e-6)) as a 1-elem
+    tensor, no host sync (feeds fused_adam_'s gscale)."""
+    return _backend(t).grad_clip_scale(t, float(max_norm))
 
 
 def l2norm(t: torch.Tensor) -> float:

@@ -36,7 +36,8 @@ void launch_count_totals(const int*, int, int, int64_t*, hipStream_t);
 void launch_scan_offsets(const int*, int, int*, hipStream_t);
 void launch_sgd(float*, const float*, float*, int64_t, float, float, float, int,
                 hipStream_t);
-void launch_adam(float*, const float*, float*, float*, void*, int64_t, float, float,
+void launch_clip_scale(const double*, float, float*, hipStream_t);
+void launch_adam(float*, const float*, float*, float*, void*, const float*, int64_t, float, float,
                  float, float, float, hipStream_t);
 void launch_sumsq(const float*, int64_t, double*, hipStream_t);
 void launch_linear_gelu(const void*, const void*, const float*, void*, void*, int,
@@ -375,22 +376,32 @@ static void fused_sgd_(torch::Tensor p, torch::Tensor g, torch::Tensor buf, doub
                cur_stream());
 }
 
+static const float* gscale_ptr(const c10::optional<torch::Tensor>& gs) {
+    if (!gs.has_value() || !gs->defined() || gs->numel() == 0) return nullptr;
+    TORCH_CHECK(gs->scalar_type() == torch::kFloat32 && gs->is_cuda() &&
+                gs->numel() == 1, "gscale must be a 1-elem cuda float tensor");
+    return gs->data_ptr<float>();
+}
+
 static void fused_adam_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-                        torch::Tensor v, double lr, double b1, double b2, double eps,
-                        double wd) {
+                        torch::Tensor v, double lr, double b1, double b2,
+                        double eps, double wd,
+                        c10::optional<torch::Tensor> gscale) {
     check_f32_1d(p, "p");
     check_f32_1d(g, "g");
     check_f32_1d(m, "m");
     check_f32_1d(v, "v");
     const at::cuda::CUDAGuard guard(p.device());
     launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-                v.data_ptr<float>(), nullptr, p.numel(), (float)lr, (float)b1,
-                (float)b2, (float)eps, (float)wd, cur_stream());
+                v.data_ptr<float>(), nullptr, gscale_ptr(gscale), p.numel(),
+                (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                cur_stream());
 }
 
 static void fused_adam_mirror_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                                torch::Tensor v, torch::Tensor p_bf16, double lr,
-                               double b1, double b2, double eps, double wd) {
+                               double b1, double b2, double eps, double wd,
+                               c10::optional<torch::Tensor> gscale) {
     check_f32_1d(p, "p");
     check_f32_1d(g, "g");
     check_f32_1d(m, "m");
@@ -399,8 +410,24 @@ static void fused_adam_mirror_(torch::Tensor p, torch::Tensor g, torch::Tensor m
                 p_bf16.numel() == p.numel());
     const at::cuda::CUDAGuard guard(p.device());
     launch_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-                v.data_ptr<float>(), p_bf16.data_ptr(), p.numel(), (float)lr,
-                (float)b1, (float)b2, (float)eps, (float)wd, cur_stream());
+                v.data_ptr<float>(), p_bf16.data_ptr(), gscale_ptr(gscale),
+                p.numel(), (float)lr, (float)b1, (float)b2, (float)eps,
+                (float)wd, cur_stream());
+}
+
+static torch::Tensor grad_clip_scale(torch::Tensor t, double max_norm) {
+    // device-resident clip factor: scale = max/(||t||+1e-6) if ||t||>max
+    // else 1 — no host sync (reference clips via a host norm check,
+    // BERT/.../optimization.py:197)
+    check_f32_1d(t, "t");
+    const at::cuda::CUDAGuard guard(t.device());
+    auto ss = torch::zeros({1}, t.options().dtype(torch::kFloat64));
+    launch_sumsq(t.data_ptr<float>(), t.numel(), ss.data_ptr<double>(),
+                 cur_stream());
+    auto out = torch::empty({1}, t.options());
+    launch_clip_scale(ss.data_ptr<double>(), (float)max_norm,
+                      out.data_ptr<float>(), cur_stream());
+    return out;
 }
 
 static double l2norm(torch::Tensor t) {
@@ -692,9 +719,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ef_restore_upcast_", &ef_restore_upcast_,
           "t = float(g_bf16) + r; r = t (fused upcast + EF restore)");
     m.def("fused_sgd_", &fused_sgd_, "fused SGD step");
-    m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step");
+    m.def("fused_adam_", &fused_adam_, "fused (Bert)Adam step",
+          py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
+          py::arg("lr"), py::arg("b1"), py::arg("b2"), py::arg("eps"),
+          py::arg("wd"), py::arg("gscale") = py::none());
     m.def("fused_adam_mirror_", &fused_adam_mirror_,
-          "fused Adam step + bf16 weight-mirror write");
+          "fused Adam step + bf16 weight-mirror write",
+          py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
+          py::arg("p_bf16"), py::arg("lr"), py::arg("b1"), py::arg("b2"),
+          py::arg("eps"), py::arg("wd"), py::arg("gscale") = py::none());
+    m.def("grad_clip_scale", &grad_clip_scale,
+          "device-resident clip factor min(1, max/||t||) (no host sync)");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
     m.def("attn_fwd", &attn_fwd,
           "fused self-attention forward: softmax(QK^T*scale+mask) dropout @ V "

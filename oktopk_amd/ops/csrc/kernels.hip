@@ -740,8 +740,14 @@ __device__ __forceinline__ float adam_one(float pi, float gi, float& mi,
 // optimizer pads group boundaries) unaligned-base fallback is n4 = 0.
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
-                            short* __restrict__ p_bf16, int64_t n, int64_t n4,
+                            short* __restrict__ p_bf16,
+                            const float* __restrict__ gscale_ptr,
+                            int64_t n, int64_t n4,
                             float lr, float b1, float b2, float eps, float wd) {
+    // device-side gradient-clip scale (grad_clip_scale kernel) — folding
+    // the clip into the g read removes the separate full-tensor mul pass
+    // AND the host .item() sync of the norm check
+    const float gs = gscale_ptr ? *gscale_ptr : 1.f;
     float4* p4 = reinterpret_cast<float4*>(p);
     const float4* g4 = reinterpret_cast<const float4*>(g);
     float4* m4 = reinterpret_cast<float4*>(m);
@@ -751,10 +757,10 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
     int64_t stride = (int64_t)gridDim.x * BLOCK;
     for (; i < n4; i += stride) {
         float4 pi = p4[i], gi = g4[i], mi = m4[i], vi = v4[i];
-        pi.x = adam_one(pi.x, gi.x, mi.x, vi.x, lr, b1, b2, eps, wd);
-        pi.y = adam_one(pi.y, gi.y, mi.y, vi.y, lr, b1, b2, eps, wd);
-        pi.z = adam_one(pi.z, gi.z, mi.z, vi.z, lr, b1, b2, eps, wd);
-        pi.w = adam_one(pi.w, gi.w, mi.w, vi.w, lr, b1, b2, eps, wd);
+        pi.x = adam_one(pi.x, gi.x * gs, mi.x, vi.x, lr, b1, b2, eps, wd);
+        pi.y = adam_one(pi.y, gi.y * gs, mi.y, vi.y, lr, b1, b2, eps, wd);
+        pi.z = adam_one(pi.z, gi.z * gs, mi.z, vi.z, lr, b1, b2, eps, wd);
+        pi.w = adam_one(pi.w, gi.w * gs, mi.w, vi.w, lr, b1, b2, eps, wd);
         m4[i] = mi; v4[i] = vi; p4[i] = pi;
         if (p_bf16)
             b4[i] = make_short4(adam_f2b(pi.x), adam_f2b(pi.y),
@@ -763,7 +769,7 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
     for (i = n4 * 4 + (int64_t)blockIdx.x * BLOCK + threadIdx.x; i < n;
          i += stride) {
         float mi = m[i], vi = v[i];
-        float pn = adam_one(p[i], g[i], mi, vi, lr, b1, b2, eps, wd);
+        float pn = adam_one(p[i], g[i] * gs, mi, vi, lr, b1, b2, eps, wd);
         m[i] = mi; v[i] = vi; p[i] = pn;
         // fused bf16 weight-mirror write (saves the separate cast pass of
         // the pure-bf16 model path)
@@ -771,8 +777,23 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
     }
 }
 
+__global__ void clip_scale_kernel(const double* __restrict__ ss,
+                                  float max_norm, float* __restrict__ out) {
+    if (threadIdx.x == 0) {
+        double gn = sqrt(*ss);
+        out[0] = gn > (double)max_norm
+                     ? (float)((double)max_norm / (gn + 1e-6)) : 1.f;
+    }
+}
+
+extern "C" void launch_clip_scale(const double* ss, float max_norm, float* out,
+                                  hipStream_t stream) {
+    hipLaunchKernelGGL(clip_scale_kernel, dim3(1), dim3(64), 0, stream,
+                       ss, max_norm, out);
+}
+
 extern "C" void launch_adam(float* p, const float* g, float* m, float* v,
-                            void* p_bf16, int64_t n,
+                            void* p_bf16, const float* gscale, int64_t n,
                             float lr, float b1, float b2, float eps, float wd,
                             hipStream_t stream) {
     bool aligned = (((uintptr_t)p | (uintptr_t)g | (uintptr_t)m |
@@ -780,7 +801,8 @@ extern "C" void launch_adam(float* p, const float* g, float* m, float* v,
                    (((uintptr_t)p_bf16) & 7) == 0;
     int64_t n4 = aligned ? n >> 2 : 0;
     hipLaunchKernelGGL(adam_kernel, dim3(n_blocks(n, 8)), dim3(BLOCK), 0, stream,
-                       p, g, m, v, (short*)p_bf16, n, n4, lr, b1, b2, eps, wd);
+                       p, g, m, v, (short*)p_bf16, gscale, n, n4, lr, b1, b2,
+                       eps, wd);
 }
 
 // ---------------------------------------------------------------------------

@@ -186,16 +186,27 @@ def fused_adam_(
     beta2: float,
     eps: float,
     weight_decay: float,
+    gscale=None,
 ) -> None:
     """BertAdam-style update: no bias correction, decoupled weight decay.
+    `gscale` (1-elem tensor) pre-scales the gradient read (device clip).
 
     Reference: BertAdam.step, BERT/bert/transformers/optimization.py:183-224."""
+    if gscale is not None:
+        grad = grad * gscale.to(grad.device, grad.dtype)
     exp_avg.mul_(beta1).add_(grad, alpha=1 - beta1)
     exp_avg_sq.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
     update = exp_avg / (exp_avg_sq.sqrt() + eps)
     if weight_decay != 0:
         update = update + weight_decay * param
     param.add_(update, alpha=-lr)
+
+
+def grad_clip_scale(t: torch.Tensor, max_norm: float) -> torch.Tensor:
+    gn = t.reshape(-1).float().norm(p=2)
+    scale = torch.where(gn > max_norm, max_norm / (gn + 1e-6),
+                        torch.ones_like(gn))
+    return scale.reshape(1)
 
 
 def l2norm(t: torch.Tensor) -> float:

@@ -521,10 +521,12 @@ class FlatBertAdam:
         else:
             self.reducer.run("flat", self.flat_grad)
         # 2. grad clip on the reduced gradient (reference optimization.py:197)
+        # — device-resident: the clip factor stays on GPU and is applied by
+        # the Adam kernel's gradient read (no host sync, no extra full-
+        # tensor mul pass)
+        gscale = None
         if self.max_grad_norm and self.max_grad_norm > 0:
-            gn = ops.l2norm(self.flat_grad)
-            if gn > self.max_grad_norm:
-                self.flat_grad.mul_(self.max_grad_norm / (gn + 1e-6))
+            gscale = ops.grad_clip_scale(self.flat_grad, self.max_grad_norm)
         # 3. fused Adam: one launch per weight-decay group (2 total); on
         # the bf16 path the model-weight mirror is written BY the Adam
         # kernel (saves a separate 660 MB cast pass)
@@ -537,12 +539,13 @@ class FlatBertAdam:
                 ops.fused_adam_mirror_(
                     self.flat_param[sl], self.flat_grad[sl], self.exp_avg[sl],
                     self.exp_avg_sq[sl], self.flat_param_model[sl], lr, b1, b2,
-                    self.eps, wd,
+                    self.eps, wd, gscale=gscale,
                 )
             else:
                 ops.fused_adam_(
                     self.flat_param[sl], self.flat_grad[sl], self.exp_avg[sl],
                     self.exp_avg_sq[sl], lr, b1, b2, self.eps, wd,
+                    gscale=gscale,
                 )
         if d > 0:
             _adam(slice(None, d), self.weight_decay)
