@@ -729,3 +729,35 @@ def test_flash_attention_bwd_dropout_autograd():
     ctx_ref.backward(gy.float())
     diff = (qkv.grad.float() - q2.grad).abs().max().item()
     assert diff < 0.15, diff
+
+
+def test_fused_adam_device_clip():
+    """gscale path == host-side clip + plain adam (reference semantics:
+    optimization.py:197 clips before the update)."""
+    from oktopk_amd import _hip_ops
+
+    torch.manual_seed(31)
+    n = 1_000_033
+    for max_norm in (0.5, 1e6):  # firing and non-firing clip
+        g = torch.randn(n, device="cuda") * 3.0
+        p0 = torch.randn(n, device="cuda")
+        # reference: host clip then adam
+        gn = g.norm(p=2).item()
+        g_ref = g * (max_norm / (gn + 1e-6)) if gn > max_norm else g.clone()
+        p_r, m_r, v_r = p0.clone(), torch.zeros_like(g), torch.zeros_like(g)
+        _hip_ops.fused_adam_(p_r, g_ref, m_r, v_r, 1e-3, 0.9, 0.999, 1e-6, 0.01)
+        # device path
+        scale = _hip_ops.grad_clip_scale(g, max_norm)
+        assert abs(scale.item() - (max_norm / (gn + 1e-6) if gn > max_norm else 1.0)) < 1e-5
+        p_d, m_d, v_d = p0.clone(), torch.zeros_like(g), torch.zeros_like(g)
+        _hip_ops.fused_adam_(p_d, g, m_d, v_d, 1e-3, 0.9, 0.999, 1e-6, 0.01,
+                             gscale=scale)
+        assert torch.allclose(p_d, p_r, atol=1e-6), (p_d - p_r).abs().max()
+        assert torch.allclose(m_d, m_r, atol=1e-6)
+        # mirror variant
+        pb = torch.zeros(n, device="cuda", dtype=torch.bfloat16)
+        p_m, m_m, v_m = p0.clone(), torch.zeros_like(g), torch.zeros_like(g)
+        _hip_ops.fused_adam_mirror_(p_m, g, m_m, v_m, pb, 1e-3, 0.9, 0.999,
+                                    1e-6, 0.01, gscale=scale)
+        assert torch.allclose(p_m, p_r, atol=1e-6)
+        assert torch.allclose(pb.float(), p_r, atol=0.05, rtol=0.01)
