@@ -411,13 +411,23 @@ class AllReducer:
             # n floats and re-compacting (saves two full-tensor passes)
             self._time(name, "reduce", time.perf_counter() - s2)
             s3 = time.perf_counter()
-            if exact:
-                gidx, gval = idx, val
-            else:
-                # one nonzero() for both gathers (boolean indexing would
-                # run it once per indexed tensor, two host round-trips)
-                sel = (val.abs() > st.tau_global).nonzero(as_tuple=False).squeeze(1)
-                gidx, gval = idx[sel], val[sel]
+            if not exact:
+                # fully fused tail: one pass over the ~k selection does the
+                # threshold filter, the densify scatter AND the residual
+                # credit — the only host sync is the 8-byte count for the
+                # feedback controller (replaces nonzero() + two gathers +
+                # fill + masked credit)
+                result = t
+                result.zero_()
+                gsz = ops.scatter_gt_credit_(result, st.residual, idx, val,
+                                             st.tau_global, 1.0)
+                if gsz < ok.global_lo_num * k // ok.global_lo_den:
+                    st.tau_global /= ok.scale_global_increase
+                elif gsz > ok.global_hi_num * k // ok.global_hi_den:
+                    st.tau_global *= ok.scale_global_decrease
+                self._time(name, "allgather", time.perf_counter() - s3)
+                return result
+            gidx, gval = idx, val
 
         if P == 1:
             # no communication: the packed wire round-trip is pure overhead
@@ -707,16 +717,18 @@ class AllReducer:
                 st.tau_global = float(top.values[-1].item())
                 g_sel_idx = idx[top.indices]
                 g_sel_val = val[top.indices]
-        else:
-            sel = (val.abs() > st.tau_global).nonzero(as_tuple=False).squeeze(1)
-            g_sel_idx, g_sel_val = idx[sel], val[sel]
-            gsz = g_sel_idx.numel()
-            if gsz < ok.global_lo_num * k // ok.global_lo_den:
-                st.tau_global /= ok.scale_global_increase
-            elif gsz > ok.global_hi_num * k // ok.global_hi_den:
-                st.tau_global *= ok.scale_global_decrease
-        ops.fill_sparse_scaled_(sl, g_sel_idx, g_sel_val, 1.0)
-        self._residual_credit(st, idx, g_sel_idx, sl.numel(), sl.device)
+            ops.fill_sparse_scaled_(sl, g_sel_idx, g_sel_val, 1.0)
+            self._residual_credit(st, idx, g_sel_idx, sl.numel(), sl.device)
+            return
+        # fused tail (see _oktopk's P==1 branch): filter + scatter + credit
+        # in one pass, 8-byte count readback only
+        sl.zero_()
+        gsz = ops.scatter_gt_credit_(sl, st.residual, idx, val,
+                                     st.tau_global, 1.0)
+        if gsz < ok.global_lo_num * k // ok.global_lo_den:
+            st.tau_global /= ok.scale_global_increase
+        elif gsz > ok.global_hi_num * k // ok.global_hi_den:
+            st.tau_global *= ok.scale_global_decrease
 
     def _repartition(self, st: TensorState, idx: torch.Tensor, n: int,
                      it: int) -> None:

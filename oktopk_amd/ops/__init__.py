@@ -156,6 +156,14 @@ def fused_adam_mirror_(param, grad, exp_avg, exp_avg_sq, param_bf16, lr, beta1,
     param_bf16.copy_(param)
 
 
+def scatter_gt_credit_(result, residual, idx, val, tau, scale=1.0) -> int:
+    """Fused world-1 round-2 tail: where |val|>tau, result[idx]=val*scale
+    and residual[idx]=0; returns the selected count.  `result` must be
+    zeroed by the caller."""
+    return int(_backend(result).scatter_gt_credit_(
+        result, residual, idx, val, float(tau), float(scale)))
+
+
 def grad_clip_scale(t: torch.Tensor, max_norm: float) -> torch.Tensor:
     """Device-resident clip factor: min(1, max/(||t||+1e-6)) as a 1-elem
     tensor, no host sync (feeds fused_adam_'s gscale)."""

@@ -24,6 +24,9 @@ void launch_hist(const float*, int64_t, uint32_t, uint32_t, int, int, unsigned i
 void launch_scatter_add(float*, const int32_t*, const float*, int64_t, hipStream_t);
 void launch_scatter_set_scaled(float*, const int32_t*, const float*, float, int64_t,
                                hipStream_t);
+void launch_scatter_gt_credit(const int32_t*, const float*, int64_t, float,
+                              float, float*, float*, unsigned long long*,
+                              hipStream_t);
 void launch_zero_at(float*, const int32_t*, int64_t, hipStream_t);
 void launch_zero_at_masked(float*, const int32_t*, const bool*, int64_t, hipStream_t);
 void launch_isin_sorted(const int32_t*, int64_t, const int32_t*, int64_t, bool*,
@@ -415,6 +418,26 @@ static void fused_adam_mirror_(torch::Tensor p, torch::Tensor g, torch::Tensor m
                 (float)wd, cur_stream());
 }
 
+static int64_t scatter_gt_credit_(torch::Tensor result,
+                                  torch::Tensor residual, torch::Tensor idx,
+                                  torch::Tensor val, double tau,
+                                  double scale) {
+    // result must arrive ZEROED; returns the number of scattered entries
+    check_f32_1d(result, "result");
+    check_f32_1d(residual, "residual");
+    check_f32_1d(val, "val");
+    TORCH_CHECK(idx.scalar_type() == torch::kInt32 && idx.is_contiguous());
+    const at::cuda::CUDAGuard guard(result.device());
+    auto cnt = torch::zeros({1}, result.options().dtype(torch::kInt64));
+    launch_scatter_gt_credit(idx.data_ptr<int32_t>(), val.data_ptr<float>(),
+                             idx.numel(), (float)tau, (float)scale,
+                             result.data_ptr<float>(),
+                             residual.data_ptr<float>(),
+                             (unsigned long long*)cnt.data_ptr<int64_t>(),
+                             cur_stream());
+    return cnt.cpu().item<int64_t>();
+}
+
 static torch::Tensor grad_clip_scale(torch::Tensor t, double max_norm) {
     // device-resident clip factor: scale = max/(||t||+1e-6) if ||t||>max
     // else 1 — no host sync (reference clips via a host norm check,
@@ -728,6 +751,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
           py::arg("p_bf16"), py::arg("lr"), py::arg("b1"), py::arg("b2"),
           py::arg("eps"), py::arg("wd"), py::arg("gscale") = py::none());
+    m.def("scatter_gt_credit_", &scatter_gt_credit_,
+          "fused world-1 round-2 tail: result[idx]=val*scale and "
+          "residual[idx]=0 where |val|>tau; returns the count");
     m.def("grad_clip_scale", &grad_clip_scale,
           "device-resident clip factor min(1, max/||t||) (no host sync)");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");

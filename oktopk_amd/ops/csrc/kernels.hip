@@ -410,6 +410,54 @@ extern "C" void launch_zero_at(float* dest, const int32_t* idx, int64_t m,
                        dest, idx, m);
 }
 
+// scatter_gt_credit: the fused world-1 round-2 tail.  For each selected
+// (idx, val) pair with |val| > tau: result[idx] = val*scale, residual[idx]
+// = 0 (the residual credit — at P==1 the global selection is a subset of
+// the local one), and count it.  Replaces the reference-shaped chain
+// boolean-mask -> nonzero (host sync) -> two gathers -> fill_sparse ->
+// masked credit (VGG/allreducer.py:853-1081's owner filter, degenerate
+// P==1 case) with one pass over the ~k selection.
+__global__ void scatter_gt_credit_kernel(const int32_t* __restrict__ idx,
+                                         const float* __restrict__ val,
+                                         int64_t m, uint32_t tau_bits,
+                                         float scale,
+                                         float* __restrict__ result,
+                                         float* __restrict__ residual,
+                                         unsigned long long* __restrict__ cnt) {
+    int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * BLOCK;
+    unsigned long long c = 0;
+    for (; i < m; i += stride) {
+        float v = val[i];
+        if (sel_gt(abs_bits(v), tau_bits)) {
+            int32_t j = idx[i];
+            result[j] = v * scale;
+            residual[j] = 0.f;
+            ++c;
+        }
+    }
+    for (int off = 32; off > 0; off >>= 1) c += __shfl_down(c, off, 64);
+    __shared__ unsigned long long ws[WAVES_PER_BLOCK];
+    int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+    if (lane == 0) ws[wave] = c;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long s = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; ++w) s += ws[w];
+        atomicAdd(cnt, s);
+    }
+}
+
+extern "C" void launch_scatter_gt_credit(const int32_t* idx, const float* val,
+                                         int64_t m, float tau, float scale,
+                                         float* result, float* residual,
+                                         unsigned long long* cnt,
+                                         hipStream_t stream) {
+    hipLaunchKernelGGL(scatter_gt_credit_kernel, dim3(n_blocks(m, 4)),
+                       dim3(BLOCK), 0, stream, idx, val, m, tau_to_bits(tau),
+                       scale, result, residual, cnt);
+}
+
 // ---------------------------------------------------------------------------
 // isin_sorted: binary search of each a[i] in ascending b
 // ---------------------------------------------------------------------------

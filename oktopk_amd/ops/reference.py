@@ -202,6 +202,14 @@ def fused_adam_(
     param.add_(update, alpha=-lr)
 
 
+def scatter_gt_credit_(result, residual, idx, val, tau, scale=1.0) -> int:
+    sel = val.abs() > tau
+    j = idx.long()[sel]
+    result[j] = val[sel] * scale
+    residual[j] = 0.0
+    return int(sel.sum())
+
+
 def grad_clip_scale(t: torch.Tensor, max_norm: float) -> torch.Tensor:
     gn = t.reshape(-1).float().norm(p=2)
     scale = torch.where(gn > max_norm, max_norm / (gn + 1e-6),
