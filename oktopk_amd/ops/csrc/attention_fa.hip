@@ -68,6 +68,11 @@ __device__ __forceinline__ void fa_philox4(unsigned long long seed,
 #define FA_LDV (FA_BT + 8)  // Vt rows padded
 #define FA_LDP (FA_BT + 8)  // P rows padded
 
+// IM = 16-row query fragments per wave (1 -> 64-row blocks, 2 -> 128-row
+// blocks).  IM=2 halves the per-block KV restaging traffic — the win at
+// large seq where the KV sweep dominates; IM=1 gives 2x the blocks — the
+// win at small seq where grid fill dominates (launcher picks by S).
+template <int IM>
 __global__ void __launch_bounds__(256)
 attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
                    const short* __restrict__ mask, // [b, s] additive bf16
@@ -77,14 +82,14 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
                    PhiloxArgsFA rng, int apply_dropout) {
     const int bh = blockIdx.x;
     const int b = bh / NH, h = bh % NH;
-    const int qblk = blockIdx.y;            // query block of FA_BM rows
+    const int qblk = blockIdx.y;            // query block of IM*64 rows
     const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
     const int fr = lane & 15;               // fragment row/col
     const int fk = (lane >> 4) * 8;         // k-offset within K=32 fragment
 
     __shared__ short k_lds[FA_BT * FA_LDK];
     __shared__ short vt_lds[FA_D * FA_LDV];
-    __shared__ short p_lds[4 * 16 * FA_LDP];
+    __shared__ short p_lds[4 * 16 * IM * FA_LDP];
 
     unsigned long long seed = rng.seed, offset = rng.offset;
     if (rng.captured) {
@@ -96,23 +101,29 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
 
     const int64_t qkv_row = (int64_t)3 * NH * FA_D;
     const int64_t base_b = (int64_t)b * S * qkv_row + (int64_t)h * FA_D;
-    const int q_base = qblk * FA_BM + wave * 16;  // this wave's 16 rows
+    const int q_base = qblk * FA_BM * IM + wave * 16 * IM;  // wave's rows
 
-    // Q fragment is tile-invariant: load once (rows q_base+fr, k=fk..fk+8
-    // per 32-chunk)
-    bf16x8_f qfr[2];
+    // Q fragments are tile-invariant: load once
+    bf16x8_f qfr[IM][2];
     #pragma unroll
-    for (int ks = 0; ks < 2; ++ks)
-        qfr[ks] = *reinterpret_cast<const bf16x8_f*>(
-            qkv + base_b + (int64_t)(q_base + fr) * qkv_row + ks * 32 + fk);
+    for (int i2 = 0; i2 < IM; ++i2)
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+            qfr[i2][ks] = *reinterpret_cast<const bf16x8_f*>(
+                qkv + base_b +
+                (int64_t)(q_base + i2 * 16 + fr) * qkv_row + ks * 32 + fk);
 
-    // online-softmax state: lane covers rows q_base + (lane>>4)*4 + r
-    float m_run[4], l_run[4];
+    // online-softmax state: lane covers rows q_base + i2*16 + (lane>>4)*4+r
+    float m_run[IM][4], l_run[IM][4];
     #pragma unroll
-    for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
-    f32x4_f oacc[4];
+    for (int i2 = 0; i2 < IM; ++i2)
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) { m_run[i2][r] = -1e30f; l_run[i2][r] = 0.f; }
+    f32x4_f oacc[IM][4];
     #pragma unroll
-    for (int j = 0; j < 4; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
+    for (int i2 = 0; i2 < IM; ++i2)
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) oacc[i2][j] = {0.f, 0.f, 0.f, 0.f};
 
     const int ntiles = S / FA_BT;
     for (int t = 0; t < ntiles; ++t) {
@@ -137,10 +148,12 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
         }
         __syncthreads();
 
-        // ---- QK^T for this tile: M=16, N=128, K=64 ------------------
-        f32x4_f acc[8];
+        // ---- QK^T for this tile: M=16*IM, N=128, K=64 ---------------
+        f32x4_f acc[IM][8];
         #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[j] = {0.f, 0.f, 0.f, 0.f};
+        for (int i2 = 0; i2 < IM; ++i2)
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) acc[i2][j] = {0.f, 0.f, 0.f, 0.f};
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
             bf16x8_f bfr[8];
@@ -149,9 +162,11 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
                 bfr[j] = *reinterpret_cast<const bf16x8_f*>(
                     &k_lds[(j * 16 + fr) * FA_LDK + ks * 32 + fk]);
             #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                acc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    qfr[ks], bfr[j], acc[j], 0, 0, 0);
+            for (int i2 = 0; i2 < IM; ++i2)
+                #pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    acc[i2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        qfr[i2][ks], bfr[j], acc[i2][j], 0, 0, 0);
         }
 
         // ---- scale + mask, online rescale, exponentiate -------------
@@ -160,35 +175,37 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
         for (int j = 0; j < 8; ++j)
             mvals[j] = mask ? fa_b2f(mask[(int64_t)b * S + key0 + j * 16 + fr])
                             : 0.f;
-        float alpha[4];
         #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-            float tm = -1e30f;
+        for (int i2 = 0; i2 < IM; ++i2) {
             #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                acc[j][r] = acc[j][r] * scale + mvals[j];
-                tm = fmaxf(tm, acc[j][r]);
+            for (int r = 0; r < 4; ++r) {
+                float tm = -1e30f;
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    acc[i2][j][r] = acc[i2][j][r] * scale + mvals[j];
+                    tm = fmaxf(tm, acc[i2][j][r]);
+                }
+                #pragma unroll
+                for (int off = 1; off < 16; off <<= 1)
+                    tm = fmaxf(tm, __shfl_xor(tm, off, 64));
+                float mnew = fmaxf(m_run[i2][r], tm);
+                float alpha = __expf(m_run[i2][r] - mnew);
+                float tsum = 0.f;
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    float e = __expf(acc[i2][j][r] - mnew);
+                    acc[i2][j][r] = e;
+                    tsum += e;
+                }
+                #pragma unroll
+                for (int off = 1; off < 16; off <<= 1)
+                    tsum += __shfl_xor(tsum, off, 64);
+                l_run[i2][r] = l_run[i2][r] * alpha + tsum;
+                m_run[i2][r] = mnew;
+                // rescale O before this tile's PV lands (textbook order)
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) oacc[i2][j][r] *= alpha;
             }
-            #pragma unroll
-            for (int off = 1; off < 16; off <<= 1)
-                tm = fmaxf(tm, __shfl_xor(tm, off, 64));
-            float mnew = fmaxf(m_run[r], tm);
-            alpha[r] = __expf(m_run[r] - mnew);
-            float tsum = 0.f;
-            #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                float e = __expf(acc[j][r] - mnew);
-                acc[j][r] = e;
-                tsum += e;
-            }
-            #pragma unroll
-            for (int off = 1; off < 16; off <<= 1)
-                tsum += __shfl_xor(tsum, off, 64);
-            l_run[r] = l_run[r] * alpha[r] + tsum;
-            m_run[r] = mnew;
-            // rescale O before this tile's PV lands (textbook order)
-            #pragma unroll
-            for (int j = 0; j < 4; ++j) oacc[j][r] *= alpha[r];
         }
 
         // ---- dropout (on unnormalised P — the l sum above is pre-drop,
@@ -197,58 +214,67 @@ attn_fwd_fa_kernel(const short* __restrict__ qkv,  // [b, s, 3, nh, hd]
         // regen kernel): counter = (bh*(S/4) + qrow/4)*S + key, word =
         // qrow&3 — ONE philox call covers a lane's 4 query rows of one key
         // (the accumulator layout), 4x fewer philox rounds than per-element.
-        short* my_p = &p_lds[wave * 16 * FA_LDP];
-        const int64_t ctr_row =
-            (int64_t)bh * (S >> 2) + (q_base >> 2) + (lane >> 4);
+        short* my_p = &p_lds[wave * 16 * IM * FA_LDP];
         #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            int key = j * 16 + fr;
-            uint32_t rnd[4];
-            if (apply_dropout)
-                fa_philox4(seed,
-                           offset + (unsigned long long)(ctr_row * S + key0 + key),
-                           rnd);
+        for (int i2 = 0; i2 < IM; ++i2) {
+            const int64_t ctr_row = (int64_t)bh * (S >> 2) +
+                ((q_base + i2 * 16) >> 2) + (lane >> 4);
             #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int prow = (lane >> 4) * 4 + r;
-                float a = acc[j][r];
+            for (int j = 0; j < 8; ++j) {
+                int key = j * 16 + fr;
+                uint32_t rnd[4];
                 if (apply_dropout)
-                    a = (rnd[r] < thresh) ? a * inv_keep : 0.f;
-                my_p[prow * FA_LDP + key] = fa_f2b(a);
+                    fa_philox4(seed,
+                               offset + (unsigned long long)(ctr_row * S + key0 + key),
+                               rnd);
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int prow = i2 * 16 + (lane >> 4) * 4 + r;
+                    float a = acc[i2][j][r];
+                    if (apply_dropout)
+                        a = (rnd[r] < thresh) ? a * inv_keep : 0.f;
+                    my_p[prow * FA_LDP + key] = fa_f2b(a);
+                }
             }
         }
         __syncthreads();
 
-        // ---- PV accumulate: M=16, N=64, K=128 -----------------------
+        // ---- PV accumulate: M=16*IM, N=64, K=128 --------------------
         #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
-            bf16x8_f afr = *reinterpret_cast<const bf16x8_f*>(
-                &my_p[fr * FA_LDP + ks * 32 + fk]);
             bf16x8_f bfr[4];
             #pragma unroll
             for (int j = 0; j < 4; ++j)
                 bfr[j] = *reinterpret_cast<const bf16x8_f*>(
                     &vt_lds[(j * 16 + fr) * FA_LDV + ks * 32 + fk]);
             #pragma unroll
-            for (int j = 0; j < 4; ++j)
-                oacc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    afr, bfr[j], oacc[j], 0, 0, 0);
+            for (int i2 = 0; i2 < IM; ++i2) {
+                bf16x8_f afr = *reinterpret_cast<const bf16x8_f*>(
+                    &my_p[(i2 * 16 + fr) * FA_LDP + ks * 32 + fk]);
+                #pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    oacc[i2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        afr, bfr[j], oacc[i2][j], 0, 0, 0);
+            }
         }
         __syncthreads();  // k_lds/vt_lds/p_lds free for the next tile
     }
 
     // ---- epilogue: normalise, write ctx + lse -----------------------
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        int qrow = q_base + (lane >> 4) * 4 + r;
-        float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
-        int64_t row = ((int64_t)b * S + qrow) * NH * FA_D + h * FA_D;
+    for (int i2 = 0; i2 < IM; ++i2) {
         #pragma unroll
-        for (int j = 0; j < 4; ++j)
-            out[row + j * 16 + fr] = fa_f2b(oacc[j][r] * inv_l);
-        if (lse && fr == 0)
-            lse[(int64_t)bh * S + qrow] =
-                m_run[r] + (l_run[r] > 0.f ? __logf(l_run[r]) : 0.f);
+        for (int r = 0; r < 4; ++r) {
+            int qrow = q_base + i2 * 16 + (lane >> 4) * 4 + r;
+            float inv_l = l_run[i2][r] > 0.f ? 1.f / l_run[i2][r] : 0.f;
+            int64_t row = ((int64_t)b * S + qrow) * NH * FA_D + h * FA_D;
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                out[row + j * 16 + fr] = fa_f2b(oacc[i2][j][r] * inv_l);
+            if (lse && fr == 0)
+                lse[(int64_t)bh * S + qrow] =
+                    m_run[i2][r] + (l_run[i2][r] > 0.f ? __logf(l_run[i2][r]) : 0.f);
+        }
     }
 }
 
@@ -266,10 +292,21 @@ extern "C" void launch_attn_fwd_fa(const void* qkv, const void* mask, void* out,
     rng.offset_ptr = (const unsigned long long*)offset_ptr;
     rng.intragraph = intragraph;
     rng.captured = captured;
-    hipLaunchKernelGGL(attn_fwd_fa_kernel, dim3(B * NH, S / FA_BM), dim3(256),
-                       0, stream, (const short*)qkv, (const short*)mask,
-                       (short*)out, (float*)lse, B, NH, S, scale, keep_prob,
-                       rng, apply_dropout);
+    if (S >= 512) {
+        // IM=2: 128-row blocks halve the per-block KV restaging sweep —
+        // the dominant cost once S/64 blocks already fill the grid
+        hipLaunchKernelGGL(attn_fwd_fa_kernel<2>,
+                           dim3(B * NH, S / (2 * FA_BM)), dim3(256), 0,
+                           stream, (const short*)qkv, (const short*)mask,
+                           (short*)out, (float*)lse, B, NH, S, scale,
+                           keep_prob, rng, apply_dropout);
+    } else {
+        hipLaunchKernelGGL(attn_fwd_fa_kernel<1>,
+                           dim3(B * NH, S / FA_BM), dim3(256), 0, stream,
+                           (const short*)qkv, (const short*)mask, (short*)out,
+                           (float*)lse, B, NH, S, scale, keep_prob, rng,
+                           apply_dropout);
+    }
 }
 
 // ---------------------------------------------------------------------------
