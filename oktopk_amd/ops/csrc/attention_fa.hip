@@ -18,6 +18,7 @@
 // stage time.
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_f;
 typedef __attribute__((ext_vector_type(4))) float f32x4_f;
@@ -292,7 +293,9 @@ extern "C" void launch_attn_fwd_fa(const void* qkv, const void* mask, void* out,
     rng.offset_ptr = (const unsigned long long*)offset_ptr;
     rng.intragraph = intragraph;
     rng.captured = captured;
-    if (S >= 512) {
+    const char* im_env = getenv("OKTOPK_FA_IM");
+    int im = im_env ? atoi(im_env) : (S >= 512 ? 2 : 1);
+    if (im == 2 && S % 128 == 0) {
         // IM=2: 128-row blocks halve the per-block KV restaging sweep —
         // the dominant cost once S/64 blocks already fill the grid
         hipLaunchKernelGGL(attn_fwd_fa_kernel<2>,
