@@ -294,7 +294,10 @@ extern "C" void launch_attn_fwd_fa(const void* qkv, const void* mask, void* out,
     rng.intragraph = intragraph;
     rng.captured = captured;
     const char* im_env = getenv("OKTOPK_FA_IM");
-    int im = im_env ? atoi(im_env) : (S >= 512 ? 2 : 1);
+    // measured (same-box A/B, s512 bs8h12): IM=2 LOSES — fwd 51.4 vs
+    // 39.4 us (acc[2][8] costs ~64 VGPRs -> fewer waves; the KV sweep
+    // was already latency-hidden).  Default stays IM=1 at every seq.
+    int im = im_env ? atoi(im_env) : 1;
     if (im == 2 && S % 128 == 0) {
         // IM=2: 128-row blocks halve the per-block KV restaging sweep —
         // the dominant cost once S/64 blocks already fill the grid
