@@ -139,6 +139,13 @@ class Trainer:
             else torch.device("cpu")
         )
         self.model_name = model_name
+        if not model_name.startswith("bert"):
+            # conv/RNN recipes dispatch through MIOpen: the full find mode
+            # picks measurably better solvers (vgg16 2.69 -> 2.51 ms/step,
+            # profiles/README.md r02-q); search cost amortizes in warmup
+            import os as _os
+
+            _os.environ.setdefault("MIOPEN_FIND_MODE", "1")
         self.cfg = cfg or EngineConfig.preset(
             "bert" if model_name.startswith("bert") else
             ("lstm" if model_name.startswith("lstm") else "vgg")
