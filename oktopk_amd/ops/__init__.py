@@ -170,5 +170,10 @@ def grad_clip_scale(t: torch.Tensor, max_norm: float) -> torch.Tensor:
     return _backend(t).grad_clip_scale(t, float(max_norm))
 
 
+def sumsq_into_(acc: torch.Tensor, t: torch.Tensor) -> None:
+    """acc (fp64[1], same device) += sum(t^2); no host sync on GPU."""
+    return _backend(t).sumsq_into_(acc, t)
+
+
 def l2norm(t: torch.Tensor) -> float:
     return float(_backend(t).l2norm(t))

@@ -453,6 +453,18 @@ static torch::Tensor grad_clip_scale(torch::Tensor t, double max_norm) {
     return out;
 }
 
+static void sumsq_into_(torch::Tensor acc, torch::Tensor t) {
+    // accumulate sum(t^2) into acc (fp64[1], device) — launch_sumsq's
+    // atomicAdd accumulates, so repeated calls over buckets build the
+    // global norm without any host sync
+    check_f32_1d(t, "t");
+    TORCH_CHECK(acc.scalar_type() == torch::kFloat64 && acc.is_cuda() &&
+                acc.numel() == 1);
+    const at::cuda::CUDAGuard guard(t.device());
+    launch_sumsq(t.data_ptr<float>(), t.numel(), acc.data_ptr<double>(),
+                 cur_stream());
+}
+
 static double l2norm(torch::Tensor t) {
     check_f32_1d(t, "t");
     const at::cuda::CUDAGuard guard(t.device());
@@ -757,6 +769,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("grad_clip_scale", &grad_clip_scale,
           "device-resident clip factor min(1, max/||t||) (no host sync)");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
+    m.def("sumsq_into_", &sumsq_into_,
+          "accumulate sum(t^2) into a device fp64[1] (no host sync)");
     m.def("attn_fwd", &attn_fwd,
           "fused self-attention forward: softmax(QK^T*scale+mask) dropout @ V "
           "(bf16 MFMA, seq=128/hd=64; returns ctx [+P, A for backward])");

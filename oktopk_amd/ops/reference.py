@@ -217,5 +217,9 @@ def grad_clip_scale(t: torch.Tensor, max_norm: float) -> torch.Tensor:
     return scale.reshape(1)
 
 
+def sumsq_into_(acc: torch.Tensor, t: torch.Tensor) -> None:
+    acc += (t.reshape(-1).double() ** 2).sum()
+
+
 def l2norm(t: torch.Tensor) -> float:
     return float(t.reshape(-1).norm(p=2).item())

@@ -326,15 +326,20 @@ class _DistributedOptimizer:
         if not self.local:
             self.synchronize()
         if self.max_grad_norm and self.max_grad_norm > 0:
-            sq = 0.0
+            # global-norm clip with NO host sync: fp64 sumsq accumulates
+            # across buckets on device, the scale is a device scalar and
+            # the bucket multiplies broadcast it (reference clips host-
+            # side per step, VGG/main_trainer.py:96-99)
+            dev = self.buckets[0].flat.device
+            acc = torch.zeros(1, dtype=torch.float64, device=dev)
             for b in self.buckets:
-                gn = ops.l2norm(b.flat)
-                sq += gn * gn
-            total = sq ** 0.5
-            if total > self.max_grad_norm:
-                scale = self.max_grad_norm / (total + 1e-6)
-                for b in self.buckets:
-                    b.flat.mul_(scale)
+                ops.sumsq_into_(acc, b.flat)
+            gn = acc.sqrt().to(torch.float32)
+            mx = self.max_grad_norm
+            scale = torch.where(gn > mx, mx / (gn + 1e-6),
+                                torch.ones_like(gn))
+            for b in self.buckets:
+                b.flat.mul_(scale)
         out = self.optimizer.step(closure)
         return out
 
