@@ -87,6 +87,17 @@ def main():
                   lambda: H.compact_adaptive_ef(t, r, gbf, taus6, N),
                   (2 + 4 * 3 + 4) * N)
     bench("zero_at 110k", lambda: H.zero_at_(r, idx), 4 * idx.numel())
+    # round-2 kernels
+    gsc = H.grad_clip_scale(t, 1.0)
+    bench("grad_clip_scale (sumsq+scale)", lambda: H.grad_clip_scale(t, 1.0), 4 * N)
+    bench("adam + device clip", lambda: H.fused_adam_(dest, t, m, v, 1e-3, 0.9,
+                                                      0.999, 1e-6, 0.01, gsc),
+          7 * 4 * N)
+    acc64 = torch.zeros(1, dtype=torch.float64, device="cuda")
+    bench("sumsq_into", lambda: H.sumsq_into_(acc64, t), 4 * N)
+    bench("scatter_gt_credit 110k",
+          lambda: H.scatter_gt_credit_(dest, r, idx, val, 0.0, 1.0),
+          3 * 8 * idx.numel())
 
     print(f"{'op':28s} {'ms':>9s} {'GB/s':>8s}")
     for name, ms, bw in rows:
