@@ -71,8 +71,10 @@ class BertSelfAttention(nn.Module):
         super().__init__()
         self.nh = cfg.num_attention_heads
         self.hd = cfg.hidden_size // cfg.num_attention_heads
-        self.qkv = nn.Linear(cfg.hidden_size, 3 * cfg.hidden_size)
-        self.out = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        from ..ops.fused_linear import ColsumLinear
+
+        self.qkv = ColsumLinear(cfg.hidden_size, 3 * cfg.hidden_size)
+        self.out = ColsumLinear(cfg.hidden_size, cfg.hidden_size)
         self.attn_drop = nn.Dropout(cfg.attention_probs_dropout_prob)
 
     def forward(self, x, attn_mask=None):
@@ -109,8 +111,10 @@ class BertLayer(nn.Module):
         super().__init__()
         self.attn = BertSelfAttention(cfg)
         self.ln1 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
-        self.fc1 = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
-        self.fc2 = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
+        from ..ops.fused_linear import ColsumLinear
+
+        self.fc1 = ColsumLinear(cfg.hidden_size, cfg.intermediate_size)
+        self.fc2 = ColsumLinear(cfg.intermediate_size, cfg.hidden_size)
         self.ln2 = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
         self.drop = nn.Dropout(cfg.hidden_dropout_prob)
 

@@ -43,6 +43,9 @@ void launch_clip_scale(const double*, float, float*, hipStream_t);
 void launch_adam(float*, const float*, float*, float*, void*, const float*, int64_t, float, float,
                  float, float, float, hipStream_t);
 void launch_sumsq(const float*, int64_t, double*, hipStream_t);
+void launch_colsum_bf16(const void*, int64_t, int64_t, float*, hipStream_t);
+void launch_attn_rowdot(const void*, const void*, int64_t, int64_t, int64_t,
+                        float*, hipStream_t);
 void launch_linear_gelu(const void*, const void*, const float*, void*, void*, int,
                         int, int, int, hipStream_t);
 void launch_add_ln_fwd(const void*, const void*, const void*, const void*, void*,
@@ -453,6 +456,37 @@ static torch::Tensor grad_clip_scale(torch::Tensor t, double max_norm) {
     return out;
 }
 
+static torch::Tensor colsum_bf16(torch::Tensor gy) {
+    // bias gradient: sum over all leading dims of a bf16 [..., C] tensor,
+    // returned bf16 [C] (fp32 accumulation)
+    TORCH_CHECK(gy.is_cuda() && gy.scalar_type() == torch::kBFloat16 &&
+                gy.is_contiguous());
+    int64_t C = gy.size(-1);
+    int64_t R = gy.numel() / C;
+    const at::cuda::CUDAGuard guard(gy.device());
+    auto acc = torch::zeros({C}, gy.options().dtype(torch::kFloat32));
+    launch_colsum_bf16(gy.data_ptr(), R, C, acc.data_ptr<float>(),
+                       cur_stream());
+    return acc.to(torch::kBFloat16);
+}
+
+static torch::Tensor attn_rowdot(torch::Tensor go, torch::Tensor out,
+                                 int64_t num_heads) {
+    // D[bh, s] = sum_d go[b,s,h,d]*out[b,s,h,d]  (head_dim 64)
+    TORCH_CHECK(go.is_cuda() && go.scalar_type() == torch::kBFloat16 &&
+                go.is_contiguous() && out.is_contiguous() &&
+                out.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(go.dim() == 3 && go.sizes() == out.sizes());
+    int64_t B = go.size(0), S = go.size(1);
+    TORCH_CHECK(go.size(2) == num_heads * 64, "head_dim must be 64");
+    const at::cuda::CUDAGuard guard(go.device());
+    auto d = torch::empty({B * num_heads, S},
+                          go.options().dtype(torch::kFloat32));
+    launch_attn_rowdot(go.data_ptr(), out.data_ptr(), B, S, num_heads,
+                       d.data_ptr<float>(), cur_stream());
+    return d;
+}
+
 static void sumsq_into_(torch::Tensor acc, torch::Tensor t) {
     // accumulate sum(t^2) into acc (fp64[1], device) — launch_sumsq's
     // atomicAdd accumulates, so repeated calls over buckets build the
@@ -769,6 +803,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("grad_clip_scale", &grad_clip_scale,
           "device-resident clip factor min(1, max/||t||) (no host sync)");
     m.def("l2norm", &l2norm, "L2 norm (fp64 accumulate)");
+    m.def("colsum_bf16", &colsum_bf16, "bias grad: column sum of bf16 [R, C]");
+    m.def("attn_rowdot", &attn_rowdot,
+          "flash-bwd D: per-(head,row) dot of gO and O (head_dim 64)");
     m.def("sumsq_into_", &sumsq_into_,
           "accumulate sum(t^2) into a device fp64[1] (no host sync)");
     m.def("attn_fwd", &attn_fwd,

@@ -142,10 +142,8 @@ class _FlashAttention(torch.autograd.Function):
             # hand-written flash backward (attn_bwd_dq/_dkv kernels):
             # D_i = gO_i . O_i, then dQ/dK/dV with P recomputed from lse
             b, s, h3 = qkv.shape
-            hd = (h3 // 3) // num_heads
             go_c = go.contiguous()
-            d = (go_c.float() * out.float()).view(b, s, num_heads, hd) \
-                .sum(-1).permute(0, 2, 1).reshape(b * num_heads, s).contiguous()
+            d = _hip_ops.attn_rowdot(go_c, out, num_heads)
             dqkv = _hip_ops.attn_bwd_fa(
                 qkv, go_c, lse, d,
                 mask if mask is not None else torch.empty(0, device=qkv.device),

@@ -64,3 +64,45 @@ class _FusedLinearGelu(torch.autograd.Function):
 def fused_linear_gelu(x: torch.Tensor, weight: torch.Tensor, bias=None) -> torch.Tensor:
     """y = gelu(x @ weight^T + bias), bf16, single MFMA kernel forward."""
     return _FusedLinearGelu.apply(x, weight, bias)
+
+
+# ---------------------------------------------------------------------------
+# ColsumLinear: nn.Linear whose BACKWARD computes the bias gradient with the
+# colsum_bf16 kernel instead of torch's reduce_kernel chain (the r02-i
+# profile showed ~60 such reductions/step at ~10 us each on BERT-base).
+# The input/weight gradients are the same two GEMMs torch runs.
+# ---------------------------------------------------------------------------
+
+class _LinearColsumBias(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        return torch.nn.functional.linear(x, w, b)
+
+    @staticmethod
+    def backward(ctx, gy):
+        from oktopk_amd import _hip_ops
+
+        x, w = ctx.saved_tensors
+        gy_c = gy.contiguous()
+        gx = gy_c @ w
+        gy2 = gy_c.reshape(-1, gy_c.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        gw = gy2.t() @ x2
+        gb = _hip_ops.colsum_bf16(gy_c)
+        return gx, gw, gb
+
+
+class ColsumLinear(torch.nn.Linear):
+    """Drop-in nn.Linear (same state_dict keys); kernel bias-grad path when
+    running bf16 on GPU with the HIP extension, plain F.linear otherwise."""
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and self.bias is not None
+                and self.weight.dtype == torch.bfloat16):
+            from . import hip_available
+
+            if hip_available():
+                return _LinearColsumBias.apply(x, self.weight, self.bias)
+        return torch.nn.functional.linear(x, self.weight, self.bias)

@@ -761,3 +761,43 @@ def test_fused_adam_device_clip():
                                     1e-6, 0.01, gscale=scale)
         assert torch.allclose(p_m, p_r, atol=1e-6)
         assert torch.allclose(pb.float(), p_r, atol=0.05, rtol=0.01)
+
+
+def test_colsum_linear_grads_match_torch():
+    """ColsumLinear (kernel bias grad) must match nn.functional.linear's
+    autograd on the same bf16 inputs, all three gradients."""
+    from oktopk_amd.ops.fused_linear import ColsumLinear
+
+    torch.manual_seed(41)
+    for r, c_in, c_out in [(1024, 768, 3072), (7, 768, 768), (64, 64, 2304)]:
+        lin = ColsumLinear(c_in, c_out).bfloat16().cuda()
+        x = (torch.randn(4, r // 4 if r % 4 == 0 else r, c_in)
+             if r % 4 == 0 else torch.randn(1, r, c_in))
+        x = x.bfloat16().cuda().requires_grad_(True)
+        y = lin(x)
+        gy = torch.randn_like(y)
+        y.backward(gy)
+
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = lin.weight.detach().clone().requires_grad_(True)
+        b2 = lin.bias.detach().clone().requires_grad_(True)
+        torch.nn.functional.linear(x2, w2, b2).backward(gy)
+        assert torch.allclose(x.grad, x2.grad, atol=1e-2, rtol=1e-2)
+        assert torch.allclose(lin.weight.grad, w2.grad, atol=1e-2, rtol=1e-2)
+        # bias: fp32-accumulated colsum vs torch's bf16-chain; compare in fp32
+        assert torch.allclose(lin.bias.grad.float(), b2.grad.float(),
+                              atol=0.05, rtol=0.02), (
+            (lin.bias.grad.float() - b2.grad.float()).abs().max())
+
+
+def test_attn_rowdot_matches_torch():
+    from oktopk_amd import _hip_ops
+
+    torch.manual_seed(42)
+    b, s, nh, hd = 3, 256, 4, 64
+    go = torch.randn(b, s, nh * hd).bfloat16().cuda()
+    out = torch.randn(b, s, nh * hd).bfloat16().cuda()
+    d = _hip_ops.attn_rowdot(go, out, nh)
+    ref = (go.float() * out.float()).view(b, s, nh, hd).sum(-1) \
+        .permute(0, 2, 1).reshape(b * nh, s)
+    assert torch.allclose(d, ref, atol=1e-2, rtol=1e-3), (d - ref).abs().max()
