@@ -100,7 +100,8 @@ class ColsumLinear(torch.nn.Linear):
     def forward(self, x):
         if (x.is_cuda and x.dtype == torch.bfloat16
                 and self.bias is not None
-                and self.weight.dtype == torch.bfloat16):
+                and self.weight.dtype == torch.bfloat16
+                and os.environ.get("OKTOPK_COLSUM_BIAS", "1") != "0"):
             from . import hip_available
 
             if hip_available():
