@@ -910,17 +910,28 @@ __global__ void colsum_bf16_kernel(const short* __restrict__ gy, int64_t R,
     int64_t r0 = (int64_t)blockIdx.y * slab;
     int64_t r1 = r0 + slab;
     if (r1 > R) r1 = R;
-    float acc = 0.f;
-    for (int64_t r = r0; r < r1; ++r) acc += ks_b2f(gy[r * C + col]);
-    atomicAdd(&out[col], acc);
+    // 4 independent partials keep 4 row loads in flight per thread
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int64_t r = r0;
+    for (; r + 4 <= r1; r += 4) {
+        a0 += ks_b2f(gy[r * C + col]);
+        a1 += ks_b2f(gy[(r + 1) * C + col]);
+        a2 += ks_b2f(gy[(r + 2) * C + col]);
+        a3 += ks_b2f(gy[(r + 3) * C + col]);
+    }
+    for (; r < r1; ++r) a0 += ks_b2f(gy[r * C + col]);
+    atomicAdd(&out[col], (a0 + a1) + (a2 + a3));
 }
 
 extern "C" void launch_colsum_bf16(const void* gy, int64_t R, int64_t C,
                                    float* out, hipStream_t stream) {
-    int64_t slab = 64;
+    // fill the chip: ~1024 blocks; column strips x row slabs
+    int64_t xblocks = (C + BLOCK - 1) / BLOCK;
+    int64_t target = 1024 / (xblocks ? xblocks : 1) + 1;
+    int64_t slab = (R + target - 1) / target;
+    if (slab < 8) slab = 8;
     int64_t nslab = (R + slab - 1) / slab;
-    if (nslab > 64) { nslab = 64; slab = (R + 63) / 64; }
-    dim3 grid((unsigned)((C + BLOCK - 1) / BLOCK), (unsigned)nslab);
+    dim3 grid((unsigned)xblocks, (unsigned)nslab);
     hipLaunchKernelGGL(colsum_bf16_kernel, grid, dim3(BLOCK), 0, stream,
                        (const short*)gy, R, C, slab, out);
 }
