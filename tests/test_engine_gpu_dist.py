@@ -39,7 +39,9 @@ def _mass_conservation_gpu(rank, comp):
         out_sum += out
         assert out.is_cuda and torch.isfinite(out).all()
     res = eng.states["w"].residual.clone()
-    for x in (in_sum, out_sum, res):
+    # out_sum is rank-identical (the allreduced result) — do NOT sum it
+    # over ranks; in_sum and the residuals are per-rank and must be
+    for x in (in_sum, res):
         xc = x.cpu()
         dist.all_reduce(xc)
         x.copy_(xc.cuda())
