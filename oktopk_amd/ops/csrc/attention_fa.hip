@@ -569,6 +569,11 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
     __shared__ short got_lds[FA_D * FB_LDQ];
     __shared__ short pa_lds[4 * 16 * FB_LDQ];   // A^T tile (for dV)
     __shared__ short pb_lds[4 * 16 * FB_LDQ];   // dS^T tile (for dK)
+    // cooperative dropout mask for this (64 q x 64 key) tile: one philox
+    // call covers a q-quad of one key (the shared counter mapping), built
+    // by all 256 threads in 4 items each — 1024 calls for 4096 elements
+    // instead of the per-element 4096 the first version paid
+    __shared__ unsigned char mask_lds[FB_QT * (FB_QT + 8)];
 
     unsigned long long seed = rng.seed, offset = rng.offset;
     if (rng.captured) {
@@ -583,6 +588,7 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
     const int64_t go_row = (int64_t)NH * FA_D;
     const int64_t go_base = (int64_t)b * S * go_row + (int64_t)h * FA_D;
     const int key_base = kblk * 64 + wave * 16;
+    const int tid2 = threadIdx.x;
 
     bf16x8_f kfr[2], vfr[2];
     #pragma unroll
@@ -631,6 +637,24 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
         }
         __syncthreads();
 
+        if (apply_dropout) {
+            #pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                int item = i * 256 + tid2;     // 0..1023 = 16 q-quads x 64 keys
+                int qg = item >> 6, key = item & 63;
+                uint32_t rnd[4];
+                fa_philox4(seed,
+                           offset + (unsigned long long)(
+                               ((int64_t)bh * (S >> 2) + ((q0 >> 2) + qg)) * S
+                               + kblk * 64 + key),
+                           rnd);
+                #pragma unroll
+                for (int w = 0; w < 4; ++w)
+                    mask_lds[(qg * 4 + w) * (FB_QT + 8) + key] =
+                        rnd[w] < thresh;
+            }
+        }
+
         // S^T and dP^T tiles: M=16 key, N=64 q, K=64 d
         f32x4_f accst[4], accdpt[4];
         #pragma unroll
@@ -653,10 +677,11 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
             }
         }
 
-        // per-column (query) stats + dS^T / A^T staging.  The dropout
-        // counter is grouped by QUERY rows (the fwd's layout), so here —
-        // key-parallel — each element needs its own philox call; the
-        // word index is the query row's low bits.
+        if (apply_dropout)
+            __syncthreads();  // mask_lds written by all threads above
+
+        // per-column (query) stats + dS^T / A^T staging; dropout keep
+        // bits come from the cooperative LDS mask
         short* my_pa = &pa_lds[wave * 16 * FB_LDQ];
         short* my_pb = &pb_lds[wave * 16 * FB_LDQ];
         #pragma unroll
@@ -664,22 +689,15 @@ attn_bwd_dkv_kernel(const short* __restrict__ qkv,
             int qcol = j * 16 + fr;
             float lse_c = lse[(int64_t)bh * S + q0 + qcol];
             float d_c = dvec[(int64_t)bh * S + q0 + qcol];
-            const int64_t ctr_base =
-                ((int64_t)bh * (S >> 2) + ((q0 + qcol) >> 2)) * S;
-            const int word = qcol & 3;
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int prow = (lane >> 4) * 4 + r;
-                int krow = key_base + prow;
                 float p = __expf(accst[j][r] * scale + mask_r[r] - lse_c);
                 float dp = accdpt[j][r];
                 float a = p;
                 if (apply_dropout) {
-                    uint32_t rnd[4];
-                    fa_philox4(seed,
-                               offset + (unsigned long long)(ctr_base + krow),
-                               rnd);
-                    int keep = rnd[word] < thresh;
+                    int key_local = wave * 16 + prow;
+                    int keep = mask_lds[qcol * (FB_QT + 8) + key_local];
                     a = keep ? p * inv_keep : 0.f;
                     dp = keep ? dp * inv_keep : 0.f;
                 }
