@@ -21,23 +21,20 @@ import torch
 
 
 def fused_attn_available(x: torch.Tensor, num_heads: int, seq: int, dropout_p: float) -> bool:
-    # DEFAULT ON at seq 128/256 since round 2: the flash kernel pair
-    # (attention_fa.hip fwd + hand-written dQ/dKdV backward) beats torch
-    # SDPA (AOTriton) isolated (70.6 vs 121.0 us fwd+bwd at s128,
-    # profiles/attn_ab_r02*.txt) AND end-to-end under hipGraph (bert_base
-    # 10.30 -> 9.99 ms/step at s128, 12.77 -> 11.87 at s256).  At s512 the
-    # isolated A/B is a tie (174 vs 168) — SDPA stays default there;
-    # OKTOPK_FUSED_ATTN=1 forces ours at any seq % 128 == 0,
-    # OKTOPK_FUSED_ATTN=0 forces SDPA everywhere.
-    env = os.environ.get("OKTOPK_FUSED_ATTN", "")
-    if env == "0":
+    # DEFAULT ON at every seq % 128 == 0 (hd 64) since round 2: the flash
+    # kernel pair (attention_fa.hip fwd + hand-written dQ/dKdV backward)
+    # beats torch SDPA (AOTriton) isolated (70.6 vs 121.0 us fwd+bwd at
+    # s128; 199 vs 213 at s512 with dropout after the cooperative LDS
+    # dropout mask) AND end-to-end under hipGraph on every measured
+    # shape: bert_base 10.30 -> 9.99 (s128), 12.77 -> 11.87 (s256),
+    # 17.0 -> 16.1 (s512); bert_large s512 35.7 -> 34.3 (2-rep A/Bs,
+    # profiles/README.md r02).  OKTOPK_FUSED_ATTN=0 forces SDPA.
+    if os.environ.get("OKTOPK_FUSED_ATTN", "") == "0":
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
     hd = x.shape[-1] // (3 * num_heads)
     if seq % 128 != 0 or hd != 64:
-        return False
-    if env != "1" and seq > 256:
         return False
     from . import hip_available
 
