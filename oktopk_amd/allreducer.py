@@ -68,8 +68,6 @@ class TensorState:
     # scratch dense byte-mask for residual credit (allocated lazily)
     mask: Optional[torch.Tensor] = None
     grad_src: Optional[torch.Tensor] = None  # transient per-run fused-EF source
-    values_snapshot: Optional[torch.Tensor] = None  # topkA family residual credit
-    indexes_snapshot: Optional[torch.Tensor] = None
 
     def state_dict(self) -> dict:
         return {
