@@ -219,3 +219,55 @@ def test_elastic_rank_failure_world4():
     re-rendezvous as world 3, restore the pre-step snapshot, retry the
     step, and train on to a lower loss with rank-identical parameters."""
     _spawn_elastic(_failure_worker, 4, die_rank=3, die_step=3)
+
+
+def _double_failure_worker(rank, world, port_store, die_rank, die_step):
+    """Two failures in sequence: rank 3 dies at step 2, rank 2 dies at step
+    5; generations advance 4 -> 3 -> 2 and the final pair still trains."""
+    import torch.distributed as dist
+
+    from oktopk_amd import Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    from oktopk_amd.elastic import ElasticAgent, apply_shrink
+    from oktopk_amd.trainer import Trainer
+
+    agent = ElasticAgent("127.0.0.1", port_store, rank, world,
+                         heartbeat_s=0.2, grace_s=1.5)
+    comm = Comm(dist.group.WORLD)
+    cfg = EngineConfig(compressor="oktopk", density=0.05,
+                       oktopk=OkTopkConfig(dense_warmup_iters=1,
+                                           region_repartition_interval=4))
+    tr = Trainer("mnistnet", batch_size=16, comm=comm, cfg=cfg, dtype="fp32")
+
+    deaths = {2: 3, 5: 2}  # step -> original rank that dies
+    for step in range(9):
+        if deaths.get(step) == rank:
+            os._exit(0)
+        dead = agent.check_alarm()
+        loss = None
+        if dead is None:
+            snap = _snapshot(tr)
+            try:
+                loss = tr.step()
+            except RuntimeError:
+                dead = agent.raise_alarm(agent.find_dead())
+        if dead is not None:
+            new_comm = agent.rebuild(dead)
+            _restore(tr, snap)
+            apply_shrink(tr.opt.reducer, tr, new_comm)
+            loss = tr.step()
+        assert loss is not None and loss == loss
+
+    assert agent.generation == 2 and len(agent.ranks) == 2, (
+        agent.generation, agent.ranks)
+    flat = torch.cat([p.detach().reshape(-1) for p in tr.model.parameters()])
+    ref = flat.clone()
+    dist.broadcast(ref, src=0)
+    same = torch.equal(flat, ref)
+    dist.barrier()
+    assert same
+    agent.stop()
+
+
+def test_elastic_double_failure_world4():
+    _spawn_elastic(_double_failure_worker, 4, die_rank=None, die_step=None)
