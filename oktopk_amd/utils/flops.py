@@ -95,6 +95,12 @@ def get_model_complexity_info(
 
     for m in model.modules():
         fn = _HOOKS.get(type(m))
+        if fn is None:
+            # subclasses (e.g. ColsumLinear) count as their base op
+            for klass, f in _HOOKS.items():
+                if isinstance(m, klass):
+                    fn = f
+                    break
         if fn is not None:
             handles.append(m.register_forward_hook(make_hook(fn)))
 
