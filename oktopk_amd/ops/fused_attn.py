@@ -1,16 +1,17 @@
-"""Fused self-attention with autograd.
+"""Fused self-attention with autograd — the production path since round 2.
 
-Forward: one CDNA4 kernel (ops/csrc/attention.hip) — QK^T, masked softmax,
-capture-safe philox dropout, PV — consuming the qkv projection buffer
-directly and emitting the context in (b, s, h) layout (the permute/copy
-chain of the eager path disappears).
+_FlashAttention (default, any seq % 128 == 0 / hd 64): online-softmax
+forward (ops/csrc/attention_fa.hip) saving only ctx + per-row LSE, and a
+hand-written two-kernel backward (query-parallel dQ; key-parallel dK/dV)
+that recomputes P from the LSE and REGENERATES the philox dropout mask —
+nothing O(S^2) is ever materialised.  Beats torch SDPA (AOTriton) at
+every measured shape (profiles/attn_ab_r02*).  The torch-recompute
+backward is kept behind OKTOPK_ATTN_BWD_TORCH=1 as the slow oracle.
 
-Backward (hand-written, torch bmms): with P (post-softmax) and A (post-
-dropout) saved by the forward,
+_FusedAttention (legacy, OKTOPK_ATTN_LEGACY=1, seq=128 only): the
+round-1 single-pass kernel saving P and A, with a torch-bmm backward,
     dV = A^T gO,   dA = gO V^T,   dP = dA * mask/keep   (mask = [A != 0]),
     dS = P * (dP - rowsum(dP * P)),  dQ = dS K * scale,  dK = dS^T Q * scale.
-The [A != 0] mask reconstruction is exact except at entries where P itself
-is bf16-zero — which carry no gradient anyway.
 """
 from __future__ import annotations
 
