@@ -242,3 +242,38 @@ def _hybrid_gpu_worker(rank):
 
 def test_gpu_world4_hybrid_dp_pp():
     run_dist(_hybrid_gpu_worker, 4)
+
+
+def _prod_combo_gpu(rank):
+    """The exact 8-GPU BERT production combination on GPU at world 2:
+    bf16 grad_src (fused EF upcast), bf16 wire, oktopk — vs the fp32-wire
+    engine on the same inputs (selected values within bf16 rounding)."""
+    import torch.distributed as dist
+
+    from oktopk_amd import AllReducer, Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+
+    torch.cuda.set_device(0)
+
+    def build(wire):
+        return AllReducer(
+            Comm(dist.group.WORLD),
+            EngineConfig(compressor="oktopk", density=DENSITY, wire_dtype=wire,
+                         oktopk=OkTopkConfig(dense_warmup_iters=0,
+                                             region_repartition_interval=2)))
+
+    e32, e16 = build("fp32"), build("bf16")
+    for it in range(4):
+        g_bf16 = _grad(rank, it).bfloat16().cuda()
+        t32 = torch.zeros(N, device="cuda")
+        t16 = torch.zeros(N, device="cuda")
+        o32 = e32.run("w", t32, grad_src=g_bf16)
+        o16 = e16.run("w", t16, grad_src=g_bf16)
+        nz = o32.nonzero().view(-1)
+        assert nz.numel() > 0
+        assert torch.allclose(o16[nz], o32[nz], rtol=0.02, atol=1e-3), (
+            (o16[nz] - o32[nz]).abs().max())
+
+
+def test_gpu_world2_bf16_gradsrc_bf16_wire():
+    run_dist(_prod_combo_gpu, 2)
