@@ -173,6 +173,64 @@ class ElasticAgent:
         self._thread.join(timeout=5)
 
 
+def snapshot_trainer(trainer) -> dict:
+    """In-memory recovery point (model + optimizer incl. EF residuals +
+    iteration) — what a failed step is retried from."""
+    import copy
+
+    return {
+        "model": copy.deepcopy(trainer.model.state_dict()),
+        "opt": copy.deepcopy(trainer.opt.state_dict()),
+        "iteration": trainer.iteration,
+    }
+
+
+def restore_trainer(trainer, snap: dict) -> None:
+    trainer.model.load_state_dict(snap["model"])
+    trainer.opt.load_state_dict(snap["opt"])
+    trainer.iteration = snap["iteration"]
+
+
+class ElasticRunner:
+    """Production step loop with rank-failure recovery (the shape proven in
+    tests/test_elastic_checkpoint.py): every step runs against the newest
+    snapshot; a collective error or heartbeat stall triggers alarm ->
+    generation re-rendezvous -> restore -> apply_shrink -> retry.  Restore
+    MUST precede apply_shrink so the snapshot's old-world region boundaries
+    are reset for the new world size.  `snapshot_interval` > 1 amortises
+    the snapshot cost for large models at the price of losing up to N-1
+    steps of progress on failure (checkpoint semantics)."""
+
+    def __init__(self, trainer, agent: ElasticAgent,
+                 snapshot_interval: int = 1):
+        self.trainer = trainer
+        self.agent = agent
+        self.snapshot_interval = max(1, snapshot_interval)
+        self._snap = None
+        self._since_snap = 0
+
+    def step(self) -> float:
+        tr = self.trainer
+        dead = self.agent.check_alarm()
+        loss = None
+        if dead is None:
+            if self._snap is None or self._since_snap >= self.snapshot_interval:
+                self._snap = snapshot_trainer(tr)
+                self._since_snap = 0
+            try:
+                loss = tr.step()
+                self._since_snap += 1
+            except RuntimeError:
+                dead = self.agent.raise_alarm(self.agent.find_dead())
+        if dead is not None:
+            new_comm = self.agent.rebuild(dead)
+            restore_trainer(tr, self._snap)
+            apply_shrink(tr.opt.reducer, tr, new_comm)
+            loss = tr.step()
+            self._since_snap = 1
+        return loss
+
+
 def install_preemption_handler(save_fn: Callable[[], None],
                                signals: Iterable[int] = (signal.SIGUSR1, signal.SIGTERM)):
     """Checkpoint on SLURM preemption signals (reference

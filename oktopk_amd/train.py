@@ -20,7 +20,7 @@ from .comm import init_from_env
 from .config import EngineConfig
 from .trainer import Trainer
 from .utils import MetricWriter, get_logger, save_checkpoint, load_checkpoint
-from .elastic import install_preemption_handler
+from .elastic import ElasticAgent, ElasticRunner, install_preemption_handler
 
 
 def build_argparser() -> argparse.ArgumentParser:
@@ -62,6 +62,14 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--profiling-norm", action="store_true",
                    help="run the dense oracle alongside and log EPS")
     p.add_argument("--logdir", type=str, default="logs")
+    # elastic failure recovery (reference MPI.ERRORS_RETURN + err_callback)
+    p.add_argument("--elastic", action="store_true",
+                   help="rank-failure detection + shrink-and-continue "
+                        "(heartbeats over a side TCPStore; survivors "
+                        "re-rendezvous and retry from an in-memory snapshot)")
+    p.add_argument("--elastic-port", type=int, default=0,
+                   help="side TCPStore port (default MASTER_PORT+17)")
+    p.add_argument("--elastic-snapshot-interval", type=int, default=1)
     return p
 
 
@@ -119,12 +127,24 @@ def main(argv=None) -> int:
             )
         )
 
+    runner = None
+    if args.elastic and comm.size > 1:
+        port = args.elastic_port or (
+            int(os.environ.get("MASTER_PORT", "29500")) + 17)
+        agent = ElasticAgent(os.environ.get("MASTER_ADDR", "127.0.0.1"),
+                             port, comm.rank, comm.size)
+        runner = ElasticRunner(trainer, agent,
+                               args.elastic_snapshot_interval)
+        logger.info("elastic recovery armed (store port %d)", port)
+
     samples_per_iter = args.batch_size * comm.size
     for epoch in range(start_epoch, args.max_epochs):
         trainer.set_epoch(epoch)
         t0 = time.time()
         for it in range(args.iters_per_epoch):
-            loss = trainer.step()
+            loss = runner.step() if runner is not None else trainer.step()
+            # comm may have shrunk mid-epoch: rank/size via the trainer
+            comm = trainer.comm
             if it % 10 == 0 and comm.rank == 0:
                 elapsed = time.time() - t0
                 ips = samples_per_iter * (it + 1) / max(elapsed, 1e-9)

@@ -271,3 +271,41 @@ def _double_failure_worker(rank, world, port_store, die_rank, die_step):
 
 def test_elastic_double_failure_world4():
     _spawn_elastic(_double_failure_worker, 4, die_rank=None, die_step=None)
+
+
+def _runner_failure_worker(rank, world, port_store, die_rank, die_step):
+    """ElasticRunner (the production --elastic wrapper) under a real rank
+    death: rank `die_rank` exits at step `die_step`; survivors' runner
+    recovers internally and the loop simply continues."""
+    import torch.distributed as dist
+
+    from oktopk_amd import Comm, EngineConfig
+    from oktopk_amd.config import OkTopkConfig
+    from oktopk_amd.elastic import ElasticAgent, ElasticRunner
+    from oktopk_amd.trainer import Trainer
+
+    agent = ElasticAgent("127.0.0.1", port_store, rank, world,
+                         heartbeat_s=0.2, grace_s=1.5)
+    comm = Comm(dist.group.WORLD)
+    cfg = EngineConfig(compressor="oktopk", density=0.05,
+                       oktopk=OkTopkConfig(dense_warmup_iters=1,
+                                           region_repartition_interval=4))
+    tr = Trainer("mnistnet", batch_size=16, comm=comm, cfg=cfg, dtype="fp32")
+    runner = ElasticRunner(tr, agent, snapshot_interval=2)
+    for step in range(10):
+        if rank == die_rank and step == die_step:
+            os._exit(0)
+        loss = runner.step()
+        assert loss == loss
+    assert tr.comm.size == world - 1
+    flat = torch.cat([p.detach().reshape(-1) for p in tr.model.parameters()])
+    ref = flat.clone()
+    dist.broadcast(ref, src=0)
+    same = torch.equal(flat, ref)
+    dist.barrier()
+    assert same
+    agent.stop()
+
+
+def test_elastic_runner_world3():
+    _spawn_elastic(_runner_failure_worker, 3, die_rank=2, die_step=4)
