@@ -65,3 +65,13 @@ def test_baseline_config1_vgg_cpu_world2(tmp_path):
     metrics = os.path.join(tmp, "logs", "metrics.jsonl")
     lines = [json.loads(l) for l in open(metrics)]
     assert any(d["tag"] == "train/loss" for d in lines)
+
+
+def test_cli_elastic_happy_path(tmp_path):
+    """--elastic arms the agent/runner without disturbing a failure-free
+    run (the failure path itself is covered by the ElasticRunner death
+    tests in test_elastic_checkpoint.py)."""
+    tmp = str(tmp_path)
+    out = _run_cli(tmp, ["--elastic"])
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert os.path.exists(os.path.join(tmp, "ck", "checkpoint.epoch.0.pth"))
