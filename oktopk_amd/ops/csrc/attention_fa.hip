@@ -317,8 +317,9 @@ extern "C" void launch_attn_fwd_fa(const void* qkv, const void* mask, void* out,
 
 // ---------------------------------------------------------------------------
 // dropout mask regeneration for the recompute backward: a[i] *= inv_keep or
-// 0 with the SAME philox counters the forward used (elem>>2 block, elem&3
-// word).  One thread per aligned 4-element philox block.
+// 0 with the SAME grouped philox counters the forward used (counter =
+// (bh*S/4 + q/4)*S + key, word = q&3) — one work item per (bh, q-quad,
+// key), four row-strided updates each.
 // ---------------------------------------------------------------------------
 __global__ void dropout_mask_mul_kernel(short* __restrict__ a, int64_t bh_n,
                                         int64_t S, PhiloxArgsFA rng,
