@@ -655,7 +655,7 @@ def test_flash_attention_spiked_key_rescale():
         (out.float() - ref).abs().max())
 
 
-@pytest.mark.parametrize("s", [128, 256])
+@pytest.mark.parametrize("s", [128, 256, 384])
 def test_flash_attention_bwd_kernel_vs_torch_recompute(s):
     """The hand-written flash backward (attn_bwd_dq/_dkv) must match the
     torch-recompute backward (OKTOPK_ATTN_BWD_TORCH=1, itself validated
