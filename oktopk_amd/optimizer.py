@@ -116,6 +116,10 @@ class _ReducerWorker:
         self._done: "queue.Queue" = queue.Queue()
         self._exc: Optional[BaseException] = None
         self._inflight = 0
+        # submit() runs on the autograd-hook thread, drain() on the main
+        # thread; backward()'s join orders them in practice, but the lock
+        # makes the counter safe under any interleaving
+        self._lock = threading.Lock()
         self._thread = threading.Thread(
             target=self._loop, daemon=True, name="oktopk-reducer")
         self._thread.start()
@@ -140,16 +144,21 @@ class _ReducerWorker:
                 self._done.put(None)
 
     def submit(self, item) -> None:
-        self._inflight += 1
+        with self._lock:
+            self._inflight += 1
         self._q.put(item)
 
     def drain(self) -> None:
         """Block until every submitted item finished; re-raise any engine
         exception on the caller (the reference's msg_queue2 'DONE' join,
         VGG/distributed_optimizer.py:96-105)."""
-        while self._inflight:
+        while True:
+            with self._lock:
+                if self._inflight == 0:
+                    break
             self._done.get()
-            self._inflight -= 1
+            with self._lock:
+                self._inflight -= 1
         if self._exc is not None:
             exc, self._exc = self._exc, None
             raise exc
