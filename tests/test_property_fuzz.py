@@ -89,3 +89,38 @@ def test_chunked_engine_invariants_world1(n, density, chunks, seed):
     assert res.numel() == n  # chunks tile the tensor exactly
     err = (total_out + res - total_in).abs().max().item()
     assert err <= 1e-4, (n, density, chunks, err)
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    sizes=st.lists(st.integers(min_value=8, max_value=3000), min_size=1,
+                   max_size=5),
+    density=st.floats(min_value=0.005, max_value=0.5),
+    warmup=st.integers(min_value=0, max_value=2),
+    iters=st.integers(min_value=1, max_value=5),
+    seed=st.integers(min_value=0, max_value=10_000),
+)
+def test_run_many_equals_serial_world1(sizes, density, warmup, iters, seed):
+    """The batched drain contract: run_many over ANY bucket list must be
+    bit-equal to serial run() calls, including mixed warmup/sparse items
+    and ragged sizes (the hook-muted hipGraph/grad-accum drain path)."""
+    def mkcfg():
+        return EngineConfig(
+            compressor="oktopk", density=density,
+            oktopk=OkTopkConfig(dense_warmup_iters=warmup,
+                                local_threshold_recompute_interval=3,
+                                global_threshold_recompute_interval=2,
+                                region_repartition_interval=2))
+
+    eng_a, eng_b = AllReducer(Comm(None), mkcfg()), AllReducer(Comm(None), mkcfg())
+    g = torch.Generator().manual_seed(seed)
+    for it in range(iters):
+        gs = [torch.randn(n, generator=g) for n in sizes]
+        outs = [eng_a.run(f"b{j}", x.clone()) for j, x in enumerate(gs)]
+        items = [(f"b{j}", x.clone(), None) for j, x in enumerate(gs)]
+        eng_b.run_many(items)
+        for j in range(len(sizes)):
+            assert torch.equal(outs[j], items[j][1]), (it, j)
+    for j in range(len(sizes)):
+        assert torch.equal(eng_a.states[f"b{j}"].residual,
+                           eng_b.states[f"b{j}"].residual)
