@@ -247,7 +247,10 @@ static std::vector<torch::Tensor> compact_adaptive_ef(
                     ntau, g.chunk, g.nblocks, counts.data_ptr<int>(),
                     cur_stream());
     auto b = bump_choose(counts, ntau, nw, hi_limit);
-    auto out = compact_finish(t, taus[b.chosen], g, counts, b.chosen, b.total);
+    // pass B compacts from the residual SNAPSHOT (= the restored values);
+    // t is left untouched — the callers densify into it afterwards anyway
+    auto out = compact_finish(residual, taus[b.chosen], g, counts, b.chosen,
+                              b.total);
     out.push_back(torch::tensor((int64_t)b.chosen));
     out.push_back(torch::tensor(b.total));
     return out;

@@ -548,7 +548,10 @@ __global__ void ef_upcast_vec_kernel(float* __restrict__ t, float* __restrict__ 
         tb.y = bf16bits_to_f32(gv[5]) + rb.y;
         tb.z = bf16bits_to_f32(gv[6]) + rb.z;
         tb.w = bf16bits_to_f32(gv[7]) + rb.w;
-        t4[2 * i] = ta; t4[2 * i + 1] = tb;
+        // only the residual snapshot is written: the engine's steady
+        // state never reads the restored t again (the result densify
+        // zeroes it), and pass B compacts from r — one full-tensor
+        // write saved per step
         r4[2 * i] = ta; r4[2 * i + 1] = tb;
     }
 }
@@ -694,7 +697,10 @@ __global__ void ef_count_kernel(float* __restrict__ t, float* __restrict__ r,
             tb.x = ub.x + rb.x; tb.y = ub.y + rb.y;
             tb.z = ub.z + rb.z; tb.w = ub.w + rb.w;
         }
-        t4[2 * i] = ta; t4[2 * i + 1] = tb;
+        // only the residual snapshot is written: the engine's steady
+        // state never reads the restored t again (the result densify
+        // zeroes it), and pass B compacts from r — one full-tensor
+        // write saved per step
         r4[2 * i] = ta; r4[2 * i + 1] = tb;
         uint32_t a[8] = {abs_bits(ta.x), abs_bits(ta.y), abs_bits(ta.z),
                          abs_bits(ta.w), abs_bits(tb.x), abs_bits(tb.y),
@@ -710,7 +716,6 @@ __global__ void ef_count_kernel(float* __restrict__ t, float* __restrict__ r,
     }
     for (int64_t i = vend + lane; i < end; i += 64) {
         float v = (HAS_G ? bf16bits_to_f32(g[i]) : t[i]) + r[i];
-        t[i] = v;
         r[i] = v;
         uint32_t a = abs_bits(v);
         #pragma unroll

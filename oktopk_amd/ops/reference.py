@@ -82,11 +82,14 @@ def compact_adaptive_ef(t: torch.Tensor, residual: torch.Tensor, grad,
     (with optional bf16 grad upcast), snapshot residual, then the bump-rule
     compaction of compact_adaptive."""
     if grad is not None:
-        t.copy_(grad.reshape(-1).to(t.dtype) + residual)
+        restored = grad.reshape(-1).to(t.dtype) + residual
     else:
-        t.add_(residual)
-    residual.copy_(t)
-    return compact_adaptive(t, taus, hi_limit)
+        restored = t + residual
+    residual.copy_(restored)
+    # t is deliberately NOT written (contract since round 2): the engine's
+    # steady state densifies the result into t without reading the
+    # restored values — saving a full-tensor write on the GPU path
+    return compact_adaptive(restored, taus, hi_limit)
 
 
 def scatter_add_(dest: torch.Tensor, idx: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
