@@ -548,10 +548,7 @@ __global__ void ef_upcast_vec_kernel(float* __restrict__ t, float* __restrict__ 
         tb.y = bf16bits_to_f32(gv[5]) + rb.y;
         tb.z = bf16bits_to_f32(gv[6]) + rb.z;
         tb.w = bf16bits_to_f32(gv[7]) + rb.w;
-        // only the residual snapshot is written: the engine's steady
-        // state never reads the restored t again (the result densify
-        // zeroes it), and pass B compacts from r — one full-tensor
-        // write saved per step
+        t4[2 * i] = ta; t4[2 * i + 1] = tb;
         r4[2 * i] = ta; r4[2 * i + 1] = tb;
     }
 }
